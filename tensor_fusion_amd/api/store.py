@@ -1,0 +1,252 @@
+"""Embedded object store — the stack's API server.
+
+The reference keeps all state in a kube-apiserver (CRDs) and consumes it via
+informers (list+watch). This store provides the same contract in-process:
+thread-safe CRUD with resource versions, label selection, watch queues and
+informer-style event handlers, plus optional JSON-lines persistence so a
+hypervisor restart can rebuild state (reference §3.5 recovery paths).
+"""
+from __future__ import annotations
+
+import copy
+import json
+import os
+import threading
+import time
+from dataclasses import asdict, is_dataclass
+from queue import Empty, Queue
+from typing import Callable, Dict, Iterable, List, Optional, Tuple
+
+from .types import TFObject
+
+
+class Conflict(Exception):
+    """Resource-version conflict (optimistic concurrency)."""
+
+
+class NotFound(Exception):
+    pass
+
+
+class AlreadyExists(Exception):
+    pass
+
+
+Event = Tuple[str, TFObject]  # ("ADDED"|"MODIFIED"|"DELETED", obj)
+
+
+class Watch:
+    def __init__(self, store: "Store", kind: str, q: Queue):
+        self._store = store
+        self._kind = kind
+        self._q = q
+        self.closed = False
+
+    def next(self, timeout: Optional[float] = None) -> Optional[Event]:
+        try:
+            return self._q.get(timeout=timeout)
+        except Empty:
+            return None
+
+    def __iter__(self):
+        while not self.closed:
+            ev = self.next(timeout=0.2)
+            if ev is not None:
+                yield ev
+
+    def stop(self):
+        self.closed = True
+        self._store._drop_watch(self._kind, self._q)
+
+
+class Store:
+    def __init__(self, persist_dir: Optional[str] = None):
+        self._lock = threading.RLock()
+        self._objs: Dict[str, Dict[str, TFObject]] = {}  # kind -> key -> obj
+        self._rv = 0
+        self._watches: Dict[str, List[Queue]] = {}
+        self._handlers: Dict[str, List[Callable[[str, TFObject], None]]] = {}
+        self._persist_dir = persist_dir
+        if persist_dir:
+            os.makedirs(persist_dir, exist_ok=True)
+            self._load()
+
+    # ------------------------------------------------------------- CRUD
+
+    def create(self, obj: TFObject) -> TFObject:
+        with self._lock:
+            kind = obj.kind
+            key = obj.meta.key
+            bucket = self._objs.setdefault(kind, {})
+            if key in bucket:
+                raise AlreadyExists(f"{kind} {key}")
+            self._rv += 1
+            obj.meta.resource_version = self._rv
+            bucket[key] = copy.deepcopy(obj)
+            self._persist(kind)
+            self._notify("ADDED", bucket[key])
+            return copy.deepcopy(bucket[key])
+
+    def get(self, kind: str, name: str, namespace: str = "") -> TFObject:
+        key = f"{namespace}/{name}" if namespace else name
+        with self._lock:
+            bucket = self._objs.get(kind, {})
+            if key not in bucket:
+                raise NotFound(f"{kind} {key}")
+            return copy.deepcopy(bucket[key])
+
+    def try_get(self, kind: str, name: str, namespace: str = "") -> Optional[TFObject]:
+        try:
+            return self.get(kind, name, namespace)
+        except NotFound:
+            return None
+
+    def update(self, obj: TFObject, check_rv: bool = True) -> TFObject:
+        with self._lock:
+            bucket = self._objs.setdefault(obj.kind, {})
+            key = obj.meta.key
+            cur = bucket.get(key)
+            if cur is None:
+                raise NotFound(f"{obj.kind} {key}")
+            if check_rv and obj.meta.resource_version != cur.meta.resource_version:
+                raise Conflict(f"{obj.kind} {key}: rv {obj.meta.resource_version} "
+                               f"!= {cur.meta.resource_version}")
+            self._rv += 1
+            obj = copy.deepcopy(obj)
+            obj.meta.resource_version = self._rv
+            bucket[key] = obj
+            self._persist(obj.kind)
+            self._notify("MODIFIED", obj)
+            return copy.deepcopy(obj)
+
+    def patch(self, kind: str, name: str, namespace: str,
+              fn: Callable[[TFObject], None], retries: int = 8) -> TFObject:
+        """Read-modify-write with conflict retry (the controllers' idiom)."""
+
+        for _ in range(retries):
+            obj = self.get(kind, name, namespace)
+            fn(obj)
+            try:
+                return self.update(obj)
+            except Conflict:
+                time.sleep(0.001)
+        raise Conflict(f"{kind} {namespace}/{name}: retries exhausted")
+
+    def delete(self, kind: str, name: str, namespace: str = "") -> None:
+        key = f"{namespace}/{name}" if namespace else name
+        with self._lock:
+            bucket = self._objs.get(kind, {})
+            obj = bucket.pop(key, None)
+            if obj is None:
+                raise NotFound(f"{kind} {key}")
+            self._persist(kind)
+            self._notify("DELETED", obj)
+
+    def list(self, kind: str, namespace: Optional[str] = None,
+             labels: Optional[Dict[str, str]] = None) -> List[TFObject]:
+        with self._lock:
+            out = []
+            for obj in self._objs.get(kind, {}).values():
+                if namespace is not None and obj.meta.namespace != namespace:
+                    continue
+                if labels and any(obj.meta.labels.get(k) != v for k, v in labels.items()):
+                    continue
+                out.append(copy.deepcopy(obj))
+            return out
+
+    # ------------------------------------------------------------ watch
+
+    def watch(self, kind: str) -> Watch:
+        q: Queue = Queue()
+        with self._lock:
+            self._watches.setdefault(kind, []).append(q)
+        return Watch(self, kind, q)
+
+    def on_change(self, kind: str, handler: Callable[[str, TFObject], None]):
+        """Informer-style synchronous handler (called under no lock)."""
+
+        with self._lock:
+            self._handlers.setdefault(kind, []).append(handler)
+
+    def _drop_watch(self, kind: str, q: Queue):
+        with self._lock:
+            try:
+                self._watches.get(kind, []).remove(q)
+            except ValueError:
+                pass
+
+    def _notify(self, event: str, obj: TFObject):
+        for q in self._watches.get(obj.kind, []):
+            q.put((event, copy.deepcopy(obj)))
+        for h in list(self._handlers.get(obj.kind, [])):
+            try:
+                h(event, copy.deepcopy(obj))
+            except Exception:  # handlers must not break the store
+                import traceback
+                traceback.print_exc()
+
+    # ---------------------------------------------------------- persist
+
+    def _persist(self, kind: str):
+        if not self._persist_dir:
+            return
+        path = os.path.join(self._persist_dir, f"{kind}.jsonl")
+        tmp = path + ".tmp"
+        with open(tmp, "w") as f:
+            for obj in self._objs.get(kind, {}).values():
+                f.write(json.dumps({"kind": kind, "obj": _to_dict(obj)}) + "\n")
+        os.replace(tmp, path)
+
+    def _load(self):
+        from . import types as T
+        for fn in os.listdir(self._persist_dir):
+            if not fn.endswith(".jsonl"):
+                continue
+            kind = fn[:-6]
+            cls = getattr(T, kind, None)
+            if cls is None:
+                continue
+            path = os.path.join(self._persist_dir, fn)
+            with open(path) as f:
+                for line in f:
+                    try:
+                        rec = json.loads(line)
+                        obj = _from_dict(cls, rec["obj"])
+                        self._objs.setdefault(kind, {})[obj.meta.key] = obj
+                        self._rv = max(self._rv, obj.meta.resource_version)
+                    except Exception:
+                        continue
+
+
+def _to_dict(obj):
+    return asdict(obj)
+
+
+def _from_dict(cls, d):
+    """Rebuild a dataclass tree from a dict (best-effort, tolerant)."""
+
+    import dataclasses
+    import typing
+
+    if not (is_dataclass(cls) and isinstance(d, dict)):
+        return d
+    kwargs = {}
+    hints = typing.get_type_hints(cls)
+    for f in dataclasses.fields(cls):
+        if f.name not in d:
+            continue
+        v = d[f.name]
+        t = hints.get(f.name, None)
+        origin = typing.get_origin(t)
+        if is_dataclass(t) and isinstance(v, dict):
+            v = _from_dict(t, v)
+        elif origin is list and v:
+            (et,) = typing.get_args(t)
+            if is_dataclass(et):
+                v = [_from_dict(et, x) for x in v]
+        elif origin is typing.Union and isinstance(v, dict):
+            args = [a for a in typing.get_args(t) if a is not type(None)]
+            if len(args) == 1 and is_dataclass(args[0]):
+                v = _from_dict(args[0], v)
+        kwargs[f.name] = v
+    return cls(**kwargs)
