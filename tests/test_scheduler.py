@@ -1,0 +1,192 @@
+"""Scheduler framework + GPUResourcesFit + topology plugin + gang."""
+import pytest
+
+from tensor_fusion_amd import constants as C
+from tensor_fusion_amd.allocator.gpuallocator import GpuAllocator
+from tensor_fusion_amd.api.store import Store
+from tensor_fusion_amd.api.types import Node, ObjectMeta, Pod
+from tensor_fusion_amd.gang.manager import GangManager
+from tensor_fusion_amd.scheduler.framework import Code, Scheduler
+from tensor_fusion_amd.scheduler.gpuresources import GPUResourcesFit
+from tensor_fusion_amd.scheduler.gputopo import GPUNetworkTopologyAware
+
+from helpers import make_node_gpus
+
+
+def build_cluster(store, nodes=2, per_node=8):
+    for i in range(nodes):
+        store.create(Node(meta=ObjectMeta(name=f"node-{i}")))
+    alloc = GpuAllocator(store=store)
+    for i in range(nodes):
+        for g in make_node_gpus(f"node-{i}", count=per_node):
+            store.create(g)
+    return alloc
+
+
+def gpu_pod(name, ns="default", tflops="200", vram="16Gi", extra=None,
+            workload=None):
+    annos = {
+        C.AnnoTflopsRequest: tflops,
+        C.AnnoVramRequest: vram,
+    }
+    annos.update(extra or {})
+    labels = {C.LabelComponent: C.ComponentWorker}
+    if workload:
+        labels[C.LabelWorkload] = workload
+    return Pod(meta=ObjectMeta(name=name, namespace=ns, annotations=annos,
+                               labels=labels),
+               scheduler_name="tensor-fusion-scheduler")
+
+
+def make_sched(store, alloc, gang=None, topo=False):
+    fit = GPUResourcesFit(store, alloc, gang=gang)
+    plugins = [fit]
+    if topo:
+        plugins.append(GPUNetworkTopologyAware(alloc))
+    return Scheduler(store, plugins)
+
+
+def test_schedule_simple_pod():
+    store = Store()
+    alloc = build_cluster(store)
+    sched = make_sched(store, alloc)
+    store.create(gpu_pod("p1"))
+    res = sched.schedule_pending()
+    assert len(res) == 1 and res[0].status == Code.Success
+    pod = store.get("Pod", "p1", "default")
+    assert pod.status.node in ("node-0", "node-1")
+    assert pod.meta.annotations[C.AnnoGpuIds].startswith("uuid-")
+    assert alloc.allocation("default/p1").bound
+
+
+def test_unschedulable_reports_reasons():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=1)
+    sched = make_sched(store, alloc)
+    store.create(gpu_pod("huge", vram="10000Gi"))
+    res = sched.schedule_pending()
+    assert res[0].status == Code.Unschedulable
+    assert res[0].reasons
+
+
+def test_binpack_fills_one_node_first():
+    store = Store()
+    alloc = build_cluster(store, nodes=2, per_node=2)
+    sched = make_sched(store, alloc)
+    for i in range(4):
+        store.create(gpu_pod(f"p{i}", tflops="100", vram="8Gi"))
+        sched.schedule_pending()
+    nodes = [store.get("Pod", f"p{i}", "default").status.node for i in range(4)]
+    # NodeCompactGPULowLoad: all four land on the same (fuller) node
+    assert len(set(nodes)) == 1
+
+
+def test_gang_all_or_nothing_success():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=8)
+    gang = GangManager(store)
+    sched = make_sched(store, alloc, gang=gang)
+    extra = {C.AnnoGangEnabled: "true", C.AnnoGangMinMembers: "4",
+             C.AnnoGangTimeout: "5", C.AnnoGangGroupKey: "tp4"}
+    for i in range(4):
+        store.create(gpu_pod(f"g{i}", tflops="300", vram="32Gi", extra=extra,
+                             workload="tp-job"))
+    res = sched.schedule_pending()
+    assert all(r.status == Code.Success for r in res), \
+        [(r.pod_key, r.status, r.reasons) for r in res]
+    nodes = {store.get("Pod", f"g{i}", "default").status.node for i in range(4)}
+    assert nodes == {"node-0"}
+
+
+def test_gang_times_out_without_quorum_and_rolls_back():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=8)
+    gang = GangManager(store)
+    gang_to = {C.AnnoGangEnabled: "true", C.AnnoGangMinMembers: "4",
+               C.AnnoGangTimeout: "0.3", C.AnnoGangGroupKey: "tp4"}
+    sched = make_sched(store, alloc, gang=gang)
+    for i in range(2):  # only 2 of 4 members exist
+        store.create(gpu_pod(f"g{i}", tflops="300", vram="32Gi", extra=gang_to))
+    res = sched.schedule_pending()
+    # PreEnqueue rejects: quorum unreachable with 2 known members
+    assert all(r.status == Code.Unschedulable for r in res)
+    assert all(alloc.allocation(f"default/g{i}") is None for i in range(2))
+
+    # now 4 members exist but 2 are unschedulably big: waiting members must
+    # time out and roll back
+    store.create(gpu_pod("g2", tflops="300", vram="32Gi", extra=gang_to))
+    store.create(gpu_pod("g3", tflops="300", vram="100000Gi", extra=gang_to))
+    sched._unsched_backoff.clear()
+    res = sched.schedule_pending()
+    by_key = {r.pod_key: r for r in res}
+    assert by_key["default/g3"].status == Code.Unschedulable
+    assert all(r.status == Code.Unschedulable for r in res)
+    for i in range(3):
+        assert alloc.allocation(f"default/g{i}") is None
+
+
+def test_topology_prefers_single_numa():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=8)
+    sched = make_sched(store, alloc, topo=True)
+    store.create(gpu_pod("tp", tflops="100", vram="8Gi",
+                         extra={C.AnnoGpuCount: "4"}))
+    res = sched.schedule_pending()
+    assert res[0].status == Code.Success
+    alloc_rec = alloc.allocation("default/tp")
+    assert len(alloc_rec.gpu_names) == 4
+    numas = {alloc.gpu(n).status.numa_node for n in alloc_rec.gpu_names}
+    assert len(numas) == 1  # whole set inside one NUMA domain
+
+
+def test_preemption_evicts_lower_qos():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=1)
+    sched = make_sched(store, alloc)
+    store.create(gpu_pod("low", vram="250Gi",
+                         extra={C.AnnoQos: C.QosLow}))
+    sched.schedule_pending()
+    assert alloc.allocation("default/low").bound
+
+    store.create(gpu_pod("high", vram="100Gi", extra={C.AnnoQos: C.QosHigh}))
+    res = sched.schedule_pending()
+    r = [x for x in res if x.pod_key == "default/high"][0]
+    assert r.status == Code.Unschedulable  # this cycle: eviction initiated
+    assert "preemption" in " ".join(r.reasons)
+    assert alloc.allocation("default/low") is None
+    assert store.get("Pod", "low", "default").status.phase == "Failed"
+
+    sched._unsched_backoff.clear()
+    res = sched.schedule_pending()
+    assert [x for x in res if x.pod_key == "default/high"][0].status == Code.Success
+
+
+def test_hard_isolation_cu_percent_annotation():
+    store = Store()
+    alloc = build_cluster(store, nodes=1, per_node=1)
+    sched = make_sched(store, alloc)
+    store.create(gpu_pod("hardpod", tflops="625", vram="16Gi",
+                         extra={C.AnnoIsolation: C.IsolationHard,
+                                C.AnnoTflopsLimit: "625"}))
+    res = sched.schedule_pending()
+    assert res[0].status == Code.Success
+    pod = store.get("Pod", "hardpod", "default")
+    pct = float(pod.meta.annotations[C.AnnoEffectiveHardCuPercent])
+    assert pct == pytest.approx(25.0, abs=0.5)  # 625/2500 → 64 CUs
+
+
+def test_partitioned_pod_gets_partition_id():
+    store = Store()
+    from tensor_fusion_amd.api.types import default_mi355x_partition_templates
+    alloc = GpuAllocator(
+        store=store, partition_templates=default_mi355x_partition_templates())
+    store.create(Node(meta=ObjectMeta(name="node-0")))
+    for g in make_node_gpus("node-0", count=1):
+        store.create(g)
+    sched = make_sched(store, alloc)
+    store.create(gpu_pod("part", tflops="300", vram="30Gi",
+                         extra={C.AnnoPartition: "true"}))
+    res = sched.schedule_pending()
+    assert res[0].status == Code.Success
+    pod = store.get("Pod", "part", "default")
+    assert pod.meta.annotations[C.AnnoPartitionId].endswith("xcd1-0")
