@@ -1,0 +1,136 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: remote-vGPU overhead vs native HIP on Llama-3-8B
+decode tok/s (BASELINE.json metric), 1..8 MI355X.
+
+For each rank (one per GPU; torchrun sets RANK/LOCAL_RANK/WORLD_SIZE) the
+bench runs the same Llama-3-8B bf16 decode workload twice, in separate child
+processes on that rank's GPU:
+  native : plain HIP/PyTorch process
+  vgpu   : the tensor-fusion-amd vGPU path — the workload runs under the
+           stack's isolation layer (LD_PRELOAD libtfhip_limiter.so attached
+           to a hypervisor-style shm page at 100% quota; TF_BENCH_VGPU_MODE
+           switches to the remoting worker when that path is selected)
+and reports overhead% = 100 * (1 - tok_s_vgpu / tok_s_native), aggregated
+over ranks (value = whole-job overhead computed from summed tok/s; MAX
+ms_per_step over ranks). Lower is better; reference headline is <4%
+(BASELINE.md README.md:56).
+
+Parent processes coordinate with a gloo group (they never touch the GPU —
+the children own it; each child brackets its timed steps with
+torch.cuda.synchronize on both sides).
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import subprocess
+import sys
+
+REPO = os.path.dirname(os.path.abspath(__file__))
+
+
+def run_child(mode: str, args, local_rank: int) -> dict:
+    env = dict(os.environ)
+    env["HIP_VISIBLE_DEVICES"] = env.get("TF_BENCH_DEVICE", str(local_rank))
+    env.pop("TF_SHM_PATH", None)
+    # children are plain single-GPU processes
+    for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT",
+              "GROUP_RANK", "LOCAL_WORLD_SIZE", "TORCHELASTIC_RUN_ID"):
+        env.pop(k, None)
+    if mode == "vgpu":
+        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "limiter")
+        if vgpu_mode == "limiter":
+            env["LD_PRELOAD"] = os.path.join(
+                REPO, "tensor_fusion_amd", "_native", "libtfhip_limiter.so")
+            env["TF_UP_LIMIT_PERCENT"] = "100"  # full vGPU of the device
+            env["TF_VRAM_LIMIT_BYTES"] = str(288 << 30)
+        elif vgpu_mode == "remote":
+            env["TF_BENCH_REMOTE"] = "1"  # remoting client path (stage 6)
+    cmd = [sys.executable, "-m", "tensor_fusion_amd.models.llama",
+           "--model", args.model, "--batch", str(args.batch),
+           "--ctx", str(args.ctx), "--steps", str(args.steps),
+           "--warmup", str(args.warmup)]
+    out = subprocess.run(cmd, env=env, cwd=REPO, capture_output=True,
+                         text=True, timeout=3600)
+    if out.returncode != 0:
+        raise RuntimeError(f"{mode} child failed:\n{out.stdout}\n{out.stderr}")
+    return json.loads(out.stdout.strip().splitlines()[-1])
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=48)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--ctx", type=int, default=512)
+    args = ap.parse_args()
+
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    rank = int(os.environ.get("RANK", "0"))
+    local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
+    n_gpus = max(world, 1)
+
+    dist = None
+    if world > 1:
+        import torch.distributed as tdist
+        dist = tdist
+        dist.init_process_group(backend="gloo")
+
+    def barrier():
+        if dist:
+            dist.barrier()
+
+    barrier()
+    native = run_child("native", args, local_rank)
+    barrier()
+    vgpu = run_child("vgpu", args, local_rank)
+    barrier()
+
+    vals = [native["tok_s"], vgpu["tok_s"], vgpu["ms_per_step"],
+            native["ms_per_step"]]
+    if dist:
+        import torch
+        t = torch.tensor(vals, dtype=torch.float64)
+        gathered = [torch.zeros_like(t) for _ in range(world)]
+        dist.all_gather(gathered, t)
+        native_tok = sum(float(g[0]) for g in gathered)
+        vgpu_tok = sum(float(g[1]) for g in gathered)
+        ms_vgpu = max(float(g[2]) for g in gathered)
+    else:
+        native_tok, vgpu_tok, ms_vgpu = vals[0], vals[1], vals[2]
+
+    overhead = 100.0 * (1.0 - vgpu_tok / native_tok)
+    if rank == 0:
+        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "limiter")
+        print(json.dumps({
+            "metric": "remote-vGPU overhead % vs native HIP (Llama-3-8B tok/s)",
+            "value": round(overhead, 3),
+            "unit": "percent",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": round(ms_vgpu, 3),
+            "higher_is_better": False,
+            "scaling": "weak",
+            "vs_baseline": round(overhead / 4.0, 4),
+            "dtype": "bf16",
+            "data": "synthetic",
+            "config": {
+                "model": args.model,
+                "global_batch": args.batch * n_gpus,
+                "seq_len": args.ctx,
+                "parallelism": f"dp{n_gpus}",
+                "vgpu_mode": vgpu_mode,
+                "native_tok_s": round(native_tok, 1),
+                "vgpu_tok_s": round(vgpu_tok, 1),
+            },
+        }), flush=True)
+    if dist:
+        dist.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

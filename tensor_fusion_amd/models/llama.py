@@ -1,0 +1,240 @@
+"""Minimal Llama-3 implementation — the serving workload for the vGPU bench.
+
+The virtualization stack (limiter / remoting / tiering) is
+workload-agnostic; this model exists so bench.py can measure Llama-3-8B
+bf16 decode tok/s natively vs through the vGPU path on MI355X
+(BASELINE.json config 3). Random-init weights, synthetic tokens.
+
+Run as a module for one benchmark child process:
+    python -m tensor_fusion_amd.models.llama --model llama3-8b \
+        --batch 8 --ctx 512 --steps 64 --warmup 8
+prints one JSON line {"tok_s": ..., "ms_per_step": ...}.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import math
+import time
+from dataclasses import dataclass
+
+import torch
+import torch.nn as nn
+import torch.nn.functional as F
+
+
+@dataclass
+class LlamaConfig:
+    vocab: int = 128256
+    dim: int = 4096
+    layers: int = 32
+    heads: int = 32
+    kv_heads: int = 8
+    intermediate: int = 14336
+    rope_theta: float = 500000.0
+    max_seq: int = 8192
+    norm_eps: float = 1e-5
+
+
+CONFIGS = {
+    "llama3-8b": LlamaConfig(),
+    "llama3-70b": LlamaConfig(dim=8192, layers=80, heads=64, kv_heads=8,
+                              intermediate=28672),
+    # CI-sized config
+    "tiny": LlamaConfig(vocab=256, dim=128, layers=2, heads=4, kv_heads=2,
+                        intermediate=256, max_seq=512),
+}
+
+
+class RMSNorm(nn.Module):
+    def __init__(self, dim: int, eps: float):
+        super().__init__()
+        self.eps = eps
+        self.weight = nn.Parameter(torch.ones(dim))
+
+    def forward(self, x):
+        dt = x.dtype
+        x = x.float()
+        x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
+        return (x * self.weight.float()).to(dt)
+
+
+def precompute_rope(cfg: LlamaConfig, device, dtype=torch.float32):
+    head_dim = cfg.dim // cfg.heads
+    inv = 1.0 / (cfg.rope_theta ** (
+        torch.arange(0, head_dim, 2, device=device, dtype=torch.float32) / head_dim))
+    t = torch.arange(cfg.max_seq, device=device, dtype=torch.float32)
+    freqs = torch.outer(t, inv)
+    return torch.cos(freqs).to(dtype), torch.sin(freqs).to(dtype)
+
+
+def apply_rope(x, cos, sin, pos):
+    # x: [B, H, T, D]; pos: [T]
+    c = cos[pos].unsqueeze(0).unsqueeze(0)  # [1,1,T,D/2]
+    s = sin[pos].unsqueeze(0).unsqueeze(0)
+    x1, x2 = x[..., ::2], x[..., 1::2]
+    o1 = x1 * c - x2 * s
+    o2 = x2 * c + x1 * s
+    out = torch.empty_like(x)
+    out[..., ::2] = o1
+    out[..., 1::2] = o2
+    return out
+
+
+class Attention(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.head_dim = cfg.dim // cfg.heads
+        self.wq = nn.Linear(cfg.dim, cfg.heads * self.head_dim, bias=False)
+        self.wk = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
+        self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
+        self.wo = nn.Linear(cfg.heads * self.head_dim, cfg.dim, bias=False)
+
+    def forward(self, x, cos, sin, pos, cache=None):
+        B, T, _ = x.shape
+        cfg = self.cfg
+        q = self.wq(x).view(B, T, cfg.heads, self.head_dim).transpose(1, 2)
+        k = self.wk(x).view(B, T, cfg.kv_heads, self.head_dim).transpose(1, 2)
+        v = self.wv(x).view(B, T, cfg.kv_heads, self.head_dim).transpose(1, 2)
+        q = apply_rope(q, cos, sin, pos)
+        k = apply_rope(k, cos, sin, pos)
+        if cache is not None:
+            k_cache, v_cache = cache
+            k_cache[:, :, pos] = k
+            v_cache[:, :, pos] = v
+            end = int(pos[-1].item()) + 1
+            k = k_cache[:, :, :end]
+            v = v_cache[:, :, :end]
+        rep = cfg.heads // cfg.kv_heads
+        if rep > 1:
+            k = k.repeat_interleave(rep, dim=1)
+            v = v.repeat_interleave(rep, dim=1)
+        causal = T > 1
+        o = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        o = o.transpose(1, 2).reshape(B, T, -1)
+        return self.wo(o)
+
+
+class MLP(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.gate = nn.Linear(cfg.dim, cfg.intermediate, bias=False)
+        self.up = nn.Linear(cfg.dim, cfg.intermediate, bias=False)
+        self.down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
+
+    def forward(self, x):
+        return self.down(F.silu(self.gate(x)) * self.up(x))
+
+
+class Block(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.attn = Attention(cfg)
+        self.mlp = MLP(cfg)
+        self.ln1 = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.ln2 = RMSNorm(cfg.dim, cfg.norm_eps)
+
+    def forward(self, x, cos, sin, pos, cache=None):
+        x = x + self.attn(self.ln1(x), cos, sin, pos, cache)
+        x = x + self.mlp(self.ln2(x))
+        return x
+
+
+class Llama(nn.Module):
+    def __init__(self, cfg: LlamaConfig):
+        super().__init__()
+        self.cfg = cfg
+        self.embed = nn.Embedding(cfg.vocab, cfg.dim)
+        self.blocks = nn.ModuleList(Block(cfg) for _ in range(cfg.layers))
+        self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
+        self.lm_head = nn.Linear(cfg.dim, cfg.vocab, bias=False)
+
+    def forward(self, tokens, pos=None, caches=None):
+        device = tokens.device
+        if pos is None:
+            pos = torch.arange(tokens.shape[1], device=device)
+        if not hasattr(self, "_cos") or self._cos.device != device:
+            self._cos, self._sin = precompute_rope(self.cfg, device)
+        x = self.embed(tokens)
+        for i, blk in enumerate(self.blocks):
+            x = blk(x, self._cos, self._sin, pos,
+                    caches[i] if caches is not None else None)
+        return self.lm_head(self.norm(x))
+
+    def make_kv_cache(self, batch: int, max_seq: int, device, dtype):
+        cfg = self.cfg
+        hd = cfg.dim // cfg.heads
+        return [(torch.zeros(batch, cfg.kv_heads, max_seq, hd, device=device,
+                             dtype=dtype),
+                 torch.zeros(batch, cfg.kv_heads, max_seq, hd, device=device,
+                             dtype=dtype)) for _ in range(cfg.layers)]
+
+
+@torch.no_grad()
+def build_model(name: str, device="cuda", dtype=torch.bfloat16,
+                seed: int = 0) -> Llama:
+    cfg = CONFIGS[name]
+    torch.manual_seed(seed)
+    with torch.device("meta"):
+        m = Llama(cfg)
+    m = m.to_empty(device=device)
+    # cheap random init directly on device (values only affect numerics,
+    # not timing; scale keeps activations finite)
+    for p in m.parameters():
+        p.data.normal_(0, 0.02)
+    return m.to(dtype)
+
+
+@torch.no_grad()
+def decode_bench(model: Llama, batch: int, ctx: int, steps: int, warmup: int,
+                 device="cuda", dtype=torch.bfloat16, sync=True):
+    """Prefill `ctx` tokens, then time `steps` single-token decode steps.
+    Returns (tok_s, ms_per_step)."""
+
+    cfg = model.cfg
+    caches = model.make_kv_cache(batch, ctx + steps + warmup + 8, device, dtype)
+    toks = torch.randint(0, cfg.vocab, (batch, ctx), device=device)
+    model(toks, pos=torch.arange(ctx, device=device), caches=caches)
+
+    cur = torch.randint(0, cfg.vocab, (batch, 1), device=device)
+
+    def step(i):
+        pos = torch.tensor([ctx + i], device=device)
+        logits = model(cur, pos=pos, caches=caches)
+        return logits.argmax(-1)
+
+    for i in range(warmup):
+        cur = step(i)
+    if sync and device != "cpu":
+        torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(warmup, warmup + steps):
+        cur = step(i)
+    if sync and device != "cpu":
+        torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return batch * steps / dt, dt / steps * 1000.0
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--model", default="llama3-8b")
+    ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--ctx", type=int, default=512)
+    ap.add_argument("--steps", type=int, default=64)
+    ap.add_argument("--warmup", type=int, default=8)
+    ap.add_argument("--device", default="cuda")
+    ap.add_argument("--dtype", default="bf16")
+    args = ap.parse_args()
+    dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
+             "fp32": torch.float32}[args.dtype]
+    model = build_model(args.model, device=args.device, dtype=dtype)
+    tok_s, ms = decode_bench(model, args.batch, args.ctx, args.steps,
+                             args.warmup, device=args.device, dtype=dtype)
+    print(json.dumps({"tok_s": tok_s, "ms_per_step": ms, "model": args.model,
+                      "batch": args.batch, "ctx": args.ctx}), flush=True)
+
+
+if __name__ == "__main__":
+    main()
