@@ -322,11 +322,40 @@ __attribute__((constructor)) void tf_limiter_init() {
 
 // --------------------------------------------------------------- dispatch
 
+// Resolve the real HIP entry point. RTLD_NEXT covers the LD_PRELOAD case
+// where the app links libamdhip64 directly; PyTorch instead dlopens its
+// bundled libamdhip64.so into a local namespace, so fall back to grabbing
+// the already-loaded instance by soname (RTLD_NOLOAD first — never load a
+// second HIP runtime beside the app's).
+void* resolve_real(const char* name) {
+  void* p = dlsym(RTLD_NEXT, name);
+  if (p) return p;
+  static std::atomic<void*> hip_handle{nullptr};
+  void* h = hip_handle.load(std::memory_order_acquire);
+  if (!h) {
+    static const char* sonames[] = {"libamdhip64.so", "libamdhip64.so.7",
+                                    "libamdhip64.so.6"};
+    for (const char* so : sonames) {
+      h = dlopen(so, RTLD_LAZY | RTLD_NOLOAD);
+      if (h) break;
+    }
+    if (!h) {
+      for (const char* so : sonames) {
+        h = dlopen(so, RTLD_LAZY);
+        if (h) break;
+      }
+    }
+    if (h) hip_handle.store(h, std::memory_order_release);
+  }
+  if (h) p = dlsym(h, name);
+  return p;
+}
+
 template <typename Fn>
 Fn real(const char* name, std::atomic<void*>* cache) {
   void* p = cache->load(std::memory_order_acquire);
   if (!p) {
-    p = dlsym(RTLD_NEXT, name);
+    p = resolve_real(name);
     if (!p) {
       fprintf(stderr, "[tf-limiter] missing real symbol %s\n", name);
       abort();
