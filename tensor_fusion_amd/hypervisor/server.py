@@ -1,0 +1,200 @@
+"""Hypervisor HTTP API (:8001) — consumed by limiters, workers, the
+operator and the TUI.
+
+Reference: pkg/hypervisor/server/server.go:95-123 + handlers/ — devices,
+workers, pod info (limits/qos/auto-freeze), process registration (container
+pid → host pid → shm), limiter list, VRAM-pressure /trap with low-QoS
+victim selection, snapshot/resume.
+"""
+from __future__ import annotations
+
+import os
+from typing import Optional
+
+from fastapi import FastAPI, Query
+from fastapi.responses import JSONResponse
+from pydantic import BaseModel
+
+from .. import constants as C
+from .device import DeviceController
+from .worker import WorkerController
+
+
+class ProcessRegistration(BaseModel):
+    container_pid: int
+    container_name: str = ""
+    namespace: str = ""
+    pod_name: str = ""
+
+
+class TrapRequest(BaseModel):
+    device_uuid: str = ""
+    bytes_needed: int = 0
+
+
+def resolve_host_pid(container_pid: int, namespace: str, pod: str) -> int:
+    """Container-pid → host-pid via /proc scan (reference legacy.go:448-481
+    walks /proc matching the pod's cgroup). Single-node mode: identity."""
+
+    target = f"{namespace}/{pod}"
+    try:
+        for entry in os.listdir("/proc"):
+            if not entry.isdigit():
+                continue
+            try:
+                with open(f"/proc/{entry}/environ", "rb") as f:
+                    env = f.read().split(b"\x00")
+                envmap = dict(e.split(b"=", 1) for e in env if b"=" in e)
+                if (envmap.get(b"POD_NAMESPACE", b"").decode() == namespace and
+                        envmap.get(b"POD_NAME", b"").decode() == pod):
+                    # NSpid match: the inner pid of this host process
+                    with open(f"/proc/{entry}/status") as f:
+                        for line in f:
+                            if line.startswith("NSpid:"):
+                                inner = line.split()[-1]
+                                if int(inner) == container_pid:
+                                    return int(entry)
+            except (OSError, ValueError):
+                continue
+    except OSError:
+        pass
+    return container_pid  # same pid namespace
+
+
+def build_app(devices: DeviceController, workers: WorkerController,
+              auto_freeze_rules=None, snapshot_fn=None) -> FastAPI:
+    app = FastAPI(title="tensor-fusion-amd hypervisor")
+    auto_freeze_rules = auto_freeze_rules or {}
+
+    @app.get("/healthz")
+    def healthz():
+        return {"ok": True}
+
+    @app.get("/api/v1/devices")
+    def get_devices():
+        out = []
+        for d in devices.devices():
+            m = None
+            try:
+                m = devices.metrics(d.index)
+            except Exception:
+                pass
+            out.append({
+                "uuid": d.uuid, "index": d.index, "name": d.name,
+                "numa_node": d.numa_node, "vram_total": d.vram_total,
+                "compute_units": d.compute_units, "xcd_count": d.xcd_count,
+                "fp16_tflops": d.fp16_tflops, "is_mock": d.is_mock,
+                "metrics": None if m is None else {
+                    "gfx_activity": m.gfx_activity,
+                    "umc_activity": m.umc_activity,
+                    "vram_used": m.vram_used,
+                },
+            })
+        return {"success": True, "data": out}
+
+    @app.get("/api/v1/workers")
+    def get_workers():
+        return {"success": True, "data": workers.worker_metrics()}
+
+    @app.get("/api/v1/pod")
+    def get_pod(namespace: str = Query("default"), pod: str = Query(...)):
+        st = workers.get(f"{namespace}/{pod}")
+        if st is None:
+            return JSONResponse(status_code=404, content={
+                "success": False, "message": "worker not found"})
+        spec = st.allocation.spec
+        af = auto_freeze_rules.get(spec.qos)
+        return {"success": True, "data": {
+            "pod_name": spec.name, "namespace": spec.namespace,
+            "gpu_uuids": spec.gpu_uuids,
+            "tflops_limit": spec.tflops_limit,
+            "vram_limit": spec.vram_limit,
+            "qos_level": spec.qos,
+            "compute_shard": st.allocation.up_limit_percent,
+            "isolation": spec.isolation,
+            "shm_path": st.allocation.shm_path,
+            "auto_freeze": {
+                "enable": bool(af and af.enable),
+                "freeze_to_mem_ttl": af.freeze_to_mem_ttl_s if af else 0,
+                "freeze_to_disk_ttl": af.freeze_to_disk_ttl_s if af else 0,
+            },
+        }}
+
+    @app.post("/api/v1/process")
+    def register_process(reg: ProcessRegistration):
+        host_pid = resolve_host_pid(reg.container_pid, reg.namespace,
+                                    reg.pod_name)
+        ok = workers.register_pid(f"{reg.namespace}/{reg.pod_name}", host_pid)
+        if not ok:
+            return JSONResponse(status_code=404, content={
+                "success": False, "message": "worker not found"})
+        return {"success": True, "data": {
+            "host_pid": host_pid, "container_pid": reg.container_pid}}
+
+    @app.get("/api/v1/limiter")
+    def limiter_list():
+        out = []
+        for st in workers.list():
+            spec = st.allocation.spec
+            out.append({
+                "worker": st.key, "qos": spec.qos,
+                "vram_limit": spec.vram_limit,
+                "up_limit_percent": st.allocation.up_limit_percent,
+                "devices": [d.uuid for d in st.allocation.devices],
+            })
+        return {"success": True, "data": out}
+
+    @app.post("/api/v1/trap")
+    def trap(req: TrapRequest):
+        """VRAM pressure: pick low-QoS victims whose working set should be
+        tiered/frozen (reference legacy.go:124-150)."""
+
+        order = {q: i for i, q in enumerate(C.QosLevels)}
+        victims = sorted(
+            (st for st in workers.list()
+             if order.get(st.allocation.spec.qos, 1) <= order[C.QosMedium]
+             and (not req.device_uuid or
+                  req.device_uuid in st.allocation.spec.gpu_uuids)),
+            key=lambda st: (order.get(st.allocation.spec.qos, 1), -st.started))
+        picked = []
+        freed = 0
+        for st in victims:
+            entry = st.shm.device(0)
+            picked.append({"worker": st.key,
+                           "vram_used": entry.pod_memory_used,
+                           "qos": st.allocation.spec.qos})
+            st.shm.set_flag(2, True)  # FLAG_VRAM_PRESSURE
+            freed += entry.pod_memory_used
+            if req.bytes_needed and freed >= req.bytes_needed:
+                break
+        return {"success": True, "data": {"victims": picked}}
+
+    @app.post("/api/v1/workers/{namespace}/{pod}/freeze")
+    def freeze(namespace: str, pod: str):
+        st = workers.get(f"{namespace}/{pod}")
+        if st is None:
+            return JSONResponse(status_code=404, content={"success": False})
+        st.shm.freeze(True)
+        return {"success": True}
+
+    @app.post("/api/v1/workers/{namespace}/{pod}/resume")
+    def resume(namespace: str, pod: str):
+        st = workers.get(f"{namespace}/{pod}")
+        if st is None:
+            return JSONResponse(status_code=404, content={"success": False})
+        st.shm.freeze(False)
+        return {"success": True}
+
+    @app.post("/api/v1/workers/{namespace}/{pod}/snapshot")
+    def snapshot(namespace: str, pod: str):
+        st = workers.get(f"{namespace}/{pod}")
+        if st is None:
+            return JSONResponse(status_code=404, content={"success": False})
+        if snapshot_fn is None:
+            return JSONResponse(status_code=501, content={
+                "success": False,
+                "message": "process snapshot requires CRIU+ROCm host support"})
+        res = snapshot_fn(st)
+        return {"success": True, "data": res}
+
+    return app

@@ -1,0 +1,222 @@
+"""Hypervisor suite against the mock accelerator backend (CPU CI) —
+the reference's hypervisor_suite_test.go pattern (mock driver)."""
+import os
+import sys
+import time
+
+import pytest
+from fastapi.testclient import TestClient
+
+from tensor_fusion_amd import constants as C
+from tensor_fusion_amd.api.store import Store
+from tensor_fusion_amd.api.types import ElasticRateLimitParams, ObjectMeta, Pod
+from tensor_fusion_amd.hypervisor.allocation import (AllocationController,
+                                                     WorkerSpec)
+from tensor_fusion_amd.hypervisor.backend.single_node import (ProcessSpec,
+                                                              SingleNodeBackend)
+from tensor_fusion_amd.hypervisor.backend.store_backend import StoreBackend
+from tensor_fusion_amd.hypervisor.device import Accelerator, DeviceController
+from tensor_fusion_amd.hypervisor.erl import ErlQuotaController, PidController
+from tensor_fusion_amd.hypervisor.server import build_app
+from tensor_fusion_amd.hypervisor.worker import WorkerController
+
+
+@pytest.fixture()
+def hyp(tmp_path, native_built):
+    accel = Accelerator(mock_devices=8)
+    devices = DeviceController(accel)
+    alloc = AllocationController(devices, shm_root=str(tmp_path / "shm"))
+    erl = ErlQuotaController(devices)
+    workers = WorkerController(devices, alloc, erl=erl,
+                               shm_root=str(tmp_path / "shm"))
+    yield devices, alloc, workers, erl
+    accel.shutdown()
+
+
+def make_spec(name="w1", uuids=None, **kw):
+    return WorkerSpec(namespace="default", name=name,
+                      gpu_uuids=uuids or ["GPU-mock-00"], **kw)
+
+
+def test_mock_discovery_and_topology(hyp):
+    devices, *_ = hyp
+    devs = devices.devices()
+    assert len(devs) == 8
+    assert devs[0].vram_total == 288 << 30
+    assert devs[0].compute_units == 256
+    tiers = devices.accel.topology(8)
+    assert all(tiers[i][j] == 0 for i in range(8) for j in range(8))
+    assert {d.numa_node for d in devs} == {0, 1}
+
+
+def test_allocation_env_soft(hyp):
+    devices, alloc, *_ = hyp
+    a = alloc.allocate(make_spec(tflops_limit=625.0, vram_limit=8 << 30))
+    assert a.env[C.EnvVisibleDevices] == "0"
+    assert a.env[C.EnvVramLimit] == str(8 << 30)
+    assert a.up_limit_percent == 25  # 625/2500
+    assert C.EnvCuMask not in a.env
+
+
+def test_allocation_env_hard_and_partitioned(hyp):
+    devices, alloc, *_ = hyp
+    a = alloc.allocate(make_spec(name="h", isolation=C.IsolationHard,
+                                 compute_percent_limit=25.0))
+    assert a.env[C.EnvCuMask] == "0:0-63"
+    b = alloc.allocate(make_spec(name="p", isolation=C.IsolationPartitioned,
+                                 partition_xcds=[2, 3]))
+    assert b.env[C.EnvCuMask] == "0:64-127"
+
+
+def test_worker_shm_lifecycle_and_orphan_sweep(hyp, tmp_path):
+    devices, alloc, workers, erl = hyp
+    st = workers.add_worker(make_spec(tflops_limit=1250.0, vram_limit=4 << 30))
+    d = st.shm.device(0)
+    assert d.uuid == "GPU-mock-00"
+    assert d.up_limit_percent == 50
+    assert d.mem_limit_bytes == 4 << 30
+    assert workers.register_pid("default/w1", os.getpid())
+    assert st.shm.pids() == [os.getpid()]
+
+    # orphan dir with a dead shm gets swept
+    orphan = tmp_path / "shm" / "default" / "ghost"
+    orphan.mkdir(parents=True)
+    (orphan / "shm").write_bytes(b"\x00" * 4096)
+    workers.sync_once()
+    assert not orphan.exists()
+
+    workers.remove_worker("default/w1")
+    assert workers.get("default/w1") is None
+    assert not (tmp_path / "shm" / "default" / "w1").exists()
+
+
+def test_pid_controller_converges():
+    """Plant: util = gain * rate. PID must settle near the setpoint."""
+
+    params = ElasticRateLimitParams()
+    pid = PidController(params)
+    gain = 0.01  # 1% util per 1 token/s
+    setpoint = 25.0
+    rate = pid.state.rate
+    for _ in range(200):
+        util = gain * rate
+        rate = pid.step(setpoint, util, dt=0.5)
+    assert abs(gain * rate - setpoint) < setpoint * 0.10, gain * rate
+
+    # regime change: kernels get 4x heavier → util jumps; PID recovers
+    gain = 0.04
+    for _ in range(200):
+        util = gain * rate
+        rate = pid.step(setpoint, util, dt=0.5)
+    assert abs(gain * rate - setpoint) < setpoint * 0.10, gain * rate
+
+
+def test_erl_controller_writes_rates(hyp, monkeypatch):
+    devices, alloc, workers, erl = hyp
+    monkeypatch.setenv("TF_ACCEL_MOCK_UTIL", "80")  # device reports 80% busy
+    st = workers.add_worker(make_spec(name="e", tflops_limit=625.0))
+    rate0 = st.shm.device(0).erl_refill_rate
+    for _ in range(10):
+        erl.tick(dt=0.5)
+    rate1 = st.shm.device(0).erl_refill_rate
+    assert rate1 < rate0  # 80% util over a 25% target → rate shrinks
+
+
+def test_store_backend_publishes_and_watches(hyp, tmp_path):
+    devices, alloc, workers, erl = hyp
+    store = Store()
+    be = StoreBackend(store, "node-0", devices, workers, pool="pool-a")
+    be.start()
+    gpus = store.list("GPU")
+    assert len(gpus) == 8
+    assert gpus[0].status.capacity.vram == 288 << 30
+    node = store.get("GPUNode", "node-0")
+    assert node.status.gpu_count == 8 and node.status.hypervisor_ready
+
+    pod = Pod(meta=ObjectMeta(
+        name="wk", namespace="default",
+        labels={C.LabelComponent: C.ComponentWorker},
+        annotations={C.AnnoGpuIds: "GPU-mock-03",
+                     C.AnnoVramLimit: "16Gi",
+                     C.AnnoTflopsLimit: "1250",
+                     C.AnnoIsolation: C.IsolationSoft}))
+    pod.status.node = "node-0"
+    store.create(pod)
+    st = workers.get("default/wk")
+    assert st is not None
+    assert st.allocation.devices[0].uuid == "GPU-mock-03"
+    assert st.allocation.up_limit_percent == 50
+
+    store.delete("Pod", "wk", "default")
+    assert workers.get("default/wk") is None
+
+
+def test_single_node_backend_spawn_restart(hyp, tmp_path):
+    devices, alloc, workers, erl = hyp
+    be = SingleNodeBackend(workers, state_file=str(tmp_path / "state.json"),
+                           limiter_path="/nonexistent")
+    marker = tmp_path / "ran"
+    be.add(ProcessSpec(
+        spec=make_spec(name="snw"),
+        command=[sys.executable, "-c",
+                 f"import pathlib; p = pathlib.Path(r'{marker}'); "
+                 "p.write_text(str(int(p.read_text())+1) if p.exists() else '1')"],
+        preload_limiter=False, restart=False))
+    be.reconcile_once()
+    st = be.procs["default/snw"]
+    st.proc.wait(timeout=10)
+    be.reconcile_once()
+    assert marker.read_text() == "1"
+    assert st.phase == "Exited"
+
+    # failing command: restart with backoff
+    be.add(ProcessSpec(spec=make_spec(name="bad"),
+                       command=[sys.executable, "-c", "raise SystemExit(3)"],
+                       preload_limiter=False))
+    for _ in range(3):
+        be.reconcile_once()
+        pst = be.procs["default/bad"]
+        if pst.proc:
+            pst.proc.wait(timeout=10)
+    be.reconcile_once()
+    pst = be.procs["default/bad"]
+    assert pst.restarts >= 1
+    assert pst.backoff_until > time.time() - 1
+
+    # state persists
+    be2 = SingleNodeBackend(workers, state_file=str(tmp_path / "state.json"))
+    assert "default/snw" in be2.procs
+    be.stop()
+
+
+def test_http_api(hyp):
+    devices, alloc, workers, erl = hyp
+    workers.add_worker(make_spec(name="api", tflops_limit=625.0,
+                                 vram_limit=2 << 30, qos=C.QosLow))
+    app = build_app(devices, workers)
+    c = TestClient(app)
+    assert c.get("/healthz").json() == {"ok": True}
+    devs = c.get("/api/v1/devices").json()["data"]
+    assert len(devs) == 8 and devs[0]["compute_units"] == 256
+
+    pod = c.get("/api/v1/pod", params={"namespace": "default",
+                                       "pod": "api"}).json()["data"]
+    assert pod["compute_shard"] == 25 and pod["qos_level"] == C.QosLow
+
+    r = c.post("/api/v1/process", json={
+        "container_pid": os.getpid(), "namespace": "default",
+        "pod_name": "api"}).json()
+    assert r["success"] and r["data"]["host_pid"] == os.getpid()
+
+    lim = c.get("/api/v1/limiter").json()["data"]
+    assert lim[0]["up_limit_percent"] == 25
+
+    trap = c.post("/api/v1/trap", json={"bytes_needed": 1}).json()
+    assert trap["data"]["victims"][0]["worker"] == "default/api"
+
+    assert c.post("/api/v1/workers/default/api/freeze").json()["success"]
+    st = workers.get("default/api")
+    assert st.shm.flags() & 1
+    assert c.post("/api/v1/workers/default/api/resume").json()["success"]
+    r = c.post("/api/v1/workers/default/api/snapshot")
+    assert r.status_code == 501  # honest: needs CRIU host support
