@@ -110,13 +110,23 @@ def build_remoting():
     worker = os.path.join(cdir, "worker_main.cpp")
     out = []
     if os.path.exists(client):
-        out.append(_cc("libtfhip_client.so",
-                       [client, os.path.join(cdir, "codeobj.cpp")],
+        out.append(_cc("libtfhip_client.so", [client],
                        extra=["-I", cdir, "-ldl", "-pthread"]))
     if os.path.exists(worker):
         out.append(_cc("tf_vgpu_worker",
                        [worker, os.path.join(cdir, "codeobj.cpp")],
                        extra=["-I", cdir, "-ldl", "-pthread"], shared=False))
+    ring_test = os.path.join(cdir, "ring_test.cpp")
+    if os.path.exists(ring_test):
+        out.append(_cc("tf_ring_test", [ring_test],
+                       extra=["-I", cdir, "-pthread"], shared=False))
+    testapp = os.path.join(cdir, "test_app.hip")
+    if os.path.exists(testapp):
+        target = os.path.join(OUT, "tf_remote_testapp")
+        if _newer(target, [testapp]):
+            _run([HIPCC, "--offload-arch=gfx950", "-O2", "-std=c++17",
+                  testapp, "-o", target])
+        out.append(target)
     return out
 
 

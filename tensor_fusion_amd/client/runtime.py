@@ -1,0 +1,99 @@
+"""Client-plane runtime: launch vGPU workers and compose client env.
+
+The remote flow (reference §3.1): client pod's stub resolves its
+TensorFusionConnection → connectionURL `native+<ip>+<port>+<worker>-<rev>`
+via the operator /connection endpoint, then attaches. Same-node transport
+is the shm ring (native/remoting/protocol.h); the socket path carries the
+bootstrap handshake.
+"""
+from __future__ import annotations
+
+import os
+import re
+import subprocess
+import time
+from dataclasses import dataclass
+from typing import Dict, Optional
+
+from .. import constants as C
+
+_NATIVE = os.path.join(os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+                       "_native")
+
+URL_RE = re.compile(r"^native\+(?P<ip>[^+]+)\+(?P<port>\d+)\+(?P<worker>.+)$")
+
+
+def parse_connection_url(url: str) -> Dict[str, str]:
+    m = URL_RE.match(url)
+    if not m:
+        raise ValueError(f"bad connection url {url!r}")
+    d = m.groupdict()
+    worker = d["worker"]
+    if "-" in worker:
+        name, _, rev = worker.rpartition("-")
+        if rev.isdigit():
+            d["worker"], d["rev"] = name, rev
+    return d
+
+
+@dataclass
+class WorkerHandle:
+    proc: subprocess.Popen
+    socket_path: str
+
+    def stop(self, timeout: float = 5.0):
+        if self.proc.poll() is None:
+            self.proc.terminate()
+            try:
+                self.proc.wait(timeout=timeout)
+            except subprocess.TimeoutExpired:
+                self.proc.kill()
+        try:
+            os.unlink(self.socket_path)
+        except OSError:
+            pass
+
+
+def start_worker(socket_path: str, device_index: int = 0,
+                 env: Optional[Dict[str, str]] = None,
+                 oneshot: bool = False, wait_s: float = 30.0) -> WorkerHandle:
+    """Spawn tf_vgpu_worker bound to one GPU, wait for its socket."""
+
+    exe = os.path.join(_NATIVE, "tf_vgpu_worker")
+    if not os.path.exists(exe):
+        raise FileNotFoundError(f"{exe} not built (python build_native.py)")
+    e = dict(os.environ)
+    e["HIP_VISIBLE_DEVICES"] = str(device_index)
+    e["ROCR_VISIBLE_DEVICES"] = str(device_index)
+    if oneshot:
+        e["TF_WORKER_ONESHOT"] = "1"
+    e.update(env or {})
+    os.makedirs(os.path.dirname(socket_path) or ".", exist_ok=True)
+    proc = subprocess.Popen([exe, socket_path], env=e)
+    deadline = time.time() + wait_s
+    while time.time() < deadline:
+        if os.path.exists(socket_path):
+            return WorkerHandle(proc=proc, socket_path=socket_path)
+        if proc.poll() is not None:
+            raise RuntimeError(
+                f"tf_vgpu_worker exited rc={proc.returncode} before listening")
+        time.sleep(0.05)
+    proc.kill()
+    raise TimeoutError(f"worker socket {socket_path} never appeared")
+
+
+def client_env(socket_path: str, base: Optional[Dict[str, str]] = None,
+               debug: bool = False) -> Dict[str, str]:
+    """Environment for a GPU-less client process driving a remote vGPU."""
+
+    e = dict(base if base is not None else os.environ)
+    client_lib = os.path.join(_NATIVE, C.ClientLibName)
+    prev = e.get("LD_PRELOAD", "")
+    e["LD_PRELOAD"] = client_lib + (" " + prev if prev else "")
+    e["TF_WORKER_SOCKET"] = socket_path
+    # the client process must NOT see a GPU: remoting is the only path
+    e["HIP_VISIBLE_DEVICES"] = ""
+    e["ROCR_VISIBLE_DEVICES"] = ""
+    if debug:
+        e["TF_CLIENT_DEBUG"] = "1"
+    return e

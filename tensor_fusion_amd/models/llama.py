@@ -91,7 +91,7 @@ class Attention(nn.Module):
         self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(cfg.heads * self.head_dim, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin, pos, cache=None):
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
         B, T, _ = x.shape
         cfg = self.cfg
         q = self.wq(x).view(B, T, cfg.heads, self.head_dim).transpose(1, 2)
@@ -103,7 +103,9 @@ class Attention(nn.Module):
             k_cache, v_cache = cache
             k_cache[:, :, pos] = k
             v_cache[:, :, pos] = v
-            end = int(pos[-1].item()) + 1
+            # pos_end is a host int: a device read here (`pos[-1].item()`)
+            # would force a D2H sync per layer per token.
+            end = pos_end if pos_end is not None else int(pos[-1].item()) + 1
             k = k_cache[:, :, :end]
             v = v_cache[:, :, :end]
         rep = cfg.heads // cfg.kv_heads
@@ -135,8 +137,8 @@ class Block(nn.Module):
         self.ln1 = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ln2 = RMSNorm(cfg.dim, cfg.norm_eps)
 
-    def forward(self, x, cos, sin, pos, cache=None):
-        x = x + self.attn(self.ln1(x), cos, sin, pos, cache)
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
+        x = x + self.attn(self.ln1(x), cos, sin, pos, cache, pos_end)
         x = x + self.mlp(self.ln2(x))
         return x
 
@@ -150,7 +152,7 @@ class Llama(nn.Module):
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.dim, cfg.vocab, bias=False)
 
-    def forward(self, tokens, pos=None, caches=None):
+    def forward(self, tokens, pos=None, caches=None, pos_end=None):
         device = tokens.device
         if pos is None:
             pos = torch.arange(tokens.shape[1], device=device)
@@ -159,7 +161,7 @@ class Llama(nn.Module):
         x = self.embed(tokens)
         for i, blk in enumerate(self.blocks):
             x = blk(x, self._cos, self._sin, pos,
-                    caches[i] if caches is not None else None)
+                    caches[i] if caches is not None else None, pos_end)
         return self.lm_head(self.norm(x))
 
     def make_kv_cache(self, batch: int, max_seq: int, device, dtype):
@@ -195,13 +197,17 @@ def decode_bench(model: Llama, batch: int, ctx: int, steps: int, warmup: int,
     cfg = model.cfg
     caches = model.make_kv_cache(batch, ctx + steps + warmup + 8, device, dtype)
     toks = torch.randint(0, cfg.vocab, (batch, ctx), device=device)
-    model(toks, pos=torch.arange(ctx, device=device), caches=caches)
+    model(toks, pos=torch.arange(ctx, device=device), caches=caches,
+          pos_end=ctx)
 
     cur = torch.randint(0, cfg.vocab, (batch, 1), device=device)
 
+    pos_buf = torch.empty(1, dtype=torch.long, device=device)
+
     def step(i):
-        pos = torch.tensor([ctx + i], device=device)
-        logits = model(cur, pos=pos, caches=caches)
+        # single small H2D per token (no device reads on the host path)
+        pos_buf.copy_(torch.tensor([ctx + i]), non_blocking=True)
+        logits = model(cur, pos=pos_buf, caches=caches, pos_end=ctx + i + 1)
         return logits.argmax(-1)
 
     for i in range(warmup):
