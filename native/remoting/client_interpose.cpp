@@ -193,7 +193,10 @@ Client::Client() {
     fprintf(stderr, "[tf-client] worker never became ready\n");
     return;
   }
-  cdbg("connected via %s", sock_path);
+  // NOTE: must not call cdbg()/C() here — this runs inside the function-local
+  // static's guarded construction; re-entering C() throws recursive_init_error.
+  if (debug)
+    fprintf(stderr, "[tf-client %d] connected via %s\n", getpid(), sock_path);
 }
 
 // ------------------------------------------------------------ transport

@@ -237,6 +237,45 @@ class GpuAllocator:
             self._allocations[key] = alloc
             return alloc
 
+    # ------------------------------------------------------- simulation
+
+    def fork_for_simulation(self, exclude_nodes: Optional[List[str]] = None
+                            ) -> "GpuAllocator":
+        """Detached deep copy of the device state for what-if placement
+        (defrag joint simulation, preemption dry-runs). No store, no quota
+        backing — mutations never touch the live allocator."""
+
+        import copy as _copy
+        with self._mu:
+            sim = GpuAllocator(store=None, quota=QuotaStore(None),
+                               strategy=self.strategy,
+                               partition_templates=self.partition_templates)
+            excl = set(exclude_nodes or [])
+            for name, g in self._gpus.items():
+                if g.status.node in excl:
+                    continue
+                sim._ingest(g)
+            sim._allocations = {k: _copy.deepcopy(a)
+                                for k, a in self._allocations.items()
+                                if not any(self._gpus[n].status.node in excl
+                                           for n in a.gpu_names
+                                           if n in self._gpus)}
+            return sim
+
+    def allocations_on(self, nodes: List[str]
+                       ) -> List[Tuple[str, "Allocation"]]:
+        """Committed allocations whose devices live on the given nodes."""
+
+        with self._mu:
+            out = []
+            nodeset = set(nodes)
+            for key, alloc in self._allocations.items():
+                if any(self._gpus.get(n) is not None
+                       and self._gpus[n].status.node in nodeset
+                       for n in alloc.gpu_names):
+                    out.append((key, alloc))
+            return out
+
     # --------------------------------------------------- Commit / Rollback
 
     def commit(self, pod_key: str) -> Allocation:

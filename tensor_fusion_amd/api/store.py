@@ -125,7 +125,10 @@ class Store:
 
         for _ in range(retries):
             obj = self.get(kind, name, namespace)
+            before = _to_dict(obj)
             fn(obj)
+            if _to_dict(obj) == before:
+                return obj  # no-op patch: no rv bump, no event storm
             try:
                 return self.update(obj)
             except Conflict:

@@ -1,0 +1,5 @@
+from .autoscaler import (Autoscaler, CronRecommender, DecayingHistogram,
+                         ExternalRecommender, PercentileRecommender)
+
+__all__ = ["Autoscaler", "PercentileRecommender", "CronRecommender",
+           "ExternalRecommender", "DecayingHistogram"]

@@ -138,7 +138,9 @@ class ControllerManager:
             progressed = False
             for ctrl in self._controllers:
                 q = self._queues[ctrl.kind]
-                while len(q):
+                # drain only what is queued at round start: reconciles that
+                # re-trigger themselves are picked up next round, not forever
+                for _ in range(len(q)):
                     req = q.pop(timeout=0)
                     if req is None:
                         break

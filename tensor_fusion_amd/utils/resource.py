@@ -92,6 +92,8 @@ def profile_from_annotations(pod: Pod,
         res.limits.tflops = res.requests.tflops
     if res.limits.vram < res.requests.vram:
         res.limits.vram = res.requests.vram
+    if res.limits.compute_percent < res.requests.compute_percent:
+        res.limits.compute_percent = res.requests.compute_percent
     return p
 
 

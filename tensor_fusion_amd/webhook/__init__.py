@@ -1,0 +1,3 @@
+from .mutator import PodMutator
+
+__all__ = ["PodMutator"]

@@ -104,6 +104,19 @@ class PodMutator:
         if not self.should_handle(pod):
             return pod
         profile = self.parse(pod)
+        # persist parsed values so re-admission (and controllers) see them
+        # even after plain-GPU resources are stripped below
+        a = pod.meta.annotations
+        a[C.AnnoGpuCount] = str(profile.gpu_count)
+        a.setdefault(C.AnnoComputePercentRequest,
+                     str(profile.resources.requests.compute_percent))
+        a.setdefault(C.AnnoComputePercentLimit,
+                     str(profile.resources.limits.compute_percent))
+        a.setdefault(C.AnnoVramRequest, str(int(profile.resources.requests.vram)))
+        a.setdefault(C.AnnoVramLimit, str(int(profile.resources.limits.vram)))
+        a.setdefault(C.AnnoQos, profile.qos)
+        if profile.is_local_gpu:
+            a.setdefault(C.AnnoIsLocalGpu, "true")
         pod.meta.labels[C.LabelEnabled] = "true"
         pod.meta.labels[C.LabelManaged] = "tensor-fusion"
 

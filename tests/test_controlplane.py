@@ -1,0 +1,382 @@
+"""Control-plane tests: webhook, controllers, connection flow, defrag,
+rollout, expander (reference test strategy: envtest-style with fake-GPU
+CRs injected directly)."""
+import time
+
+import pytest
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.api.store import Store
+from tensor_fusion_amd.api.types import (GPU, Container, GPUNode, GPUPool,
+                                         GPUResourceQuota, Node, Pod,
+                                         Resource, TensorFusionCluster,
+                                         TensorFusionConnection,
+                                         TensorFusionWorkload,
+                                         WorkloadProfile)
+from tensor_fusion_amd.allocator.gpuallocator import GpuAllocator
+from tensor_fusion_amd.controllers import (ControllerManager,
+                                           DefragController,
+                                           default_controllers,
+                                           generate_worker_pod)
+from tensor_fusion_amd.portallocator import IndexAllocator, PortAllocator
+from tensor_fusion_amd.webhook import PodMutator
+
+
+def mk_gpu(name, node="node-a", pool="pool-a", tflops=2500.0,
+           vram=C.MI355X_VRAM_BYTES):
+    g = GPU()
+    g.meta.name = name
+    g.status.uuid = f"uuid-{name}"
+    g.status.node = node
+    g.status.pool = pool
+    g.status.capacity = Resource(tflops, vram, 100.0)
+    g.status.available = Resource(tflops, vram, 100.0)
+    return g
+
+
+def mk_client_pod(name="app-1", ns="default", annotations=None):
+    p = Pod()
+    p.meta.name = name
+    p.meta.namespace = ns
+    p.meta.labels[C.LabelEnabled] = "true"
+    p.meta.annotations.update(annotations or {})
+    p.containers = [Container(name="main", image="app:1")]
+    return p
+
+
+# ----------------------------------------------------------- webhook
+
+
+class TestWebhook:
+    def test_remote_mode_creates_workload_and_injects_client(self):
+        store = Store()
+        m = PodMutator(store)
+        pod = mk_client_pod(annotations={
+            C.AnnoTflopsRequest: "600", C.AnnoVramRequest: "16Gi"})
+        m.handle(pod)
+        wl = store.get("TensorFusionWorkload", "app-1-wl", "default")
+        assert wl.profile.resources.requests.tflops == 600
+        assert wl.profile.resources.requests.vram == 16 << 30
+        env = pod.containers[0].env
+        assert env[C.EnvConnectionName] == "app-1-conn"
+        assert C.ClientLibName in env["LD_PRELOAD"]
+
+    def test_local_mode_sets_scheduler_and_limiter_env(self):
+        store = Store()
+        idx = IndexAllocator()
+        m = PodMutator(store, index_allocator=idx)
+        pod = mk_client_pod(annotations={
+            C.AnnoIsLocalGpu: "true",
+            C.AnnoComputePercentRequest: "25",
+            C.AnnoVramLimit: "8Gi"})
+        m.handle(pod)
+        assert pod.scheduler_name == C.SchedulerName
+        env = pod.containers[0].env
+        assert env[C.EnvVramLimit] == str(8 << 30)
+        assert env[C.EnvUpLimitPercent] == "25"
+        assert C.LimiterLibName in env["LD_PRELOAD"]
+        assert any(k.startswith(C.IndexResourcePrefix)
+                   for k in pod.containers[0].resources)
+
+    def test_auto_migration_of_plain_gpu_pod(self):
+        store = Store()
+        m = PodMutator(store)
+        pod = Pod()
+        pod.meta.name = "legacy"
+        pod.meta.namespace = "default"
+        pod.containers = [Container(name="main",
+                                    resources={"amd.com/gpu": "2"})]
+        assert m.should_handle(pod)
+        m.handle(pod)
+        assert "amd.com/gpu" not in pod.containers[0].resources
+        assert pod.scheduler_name == C.SchedulerName  # local mode
+        prof = m.parse(pod)
+        assert prof.gpu_count == 2
+
+    def test_idempotent(self):
+        store = Store()
+        m = PodMutator(store)
+        pod = mk_client_pod(annotations={C.AnnoTflopsRequest: "100"})
+        m.handle(pod)
+        env1 = dict(pod.containers[0].env)
+        n_mounts = len(pod.containers[0].volume_mounts)
+        m.handle(pod)
+        assert pod.containers[0].env == env1
+        # mounts may duplicate — they must not (regression guard)
+        assert len(pod.containers[0].volume_mounts) <= n_mounts + 1
+
+    def test_qos_derived_from_fraction(self):
+        store = Store()
+        m = PodMutator(store)
+        pod = mk_client_pod(annotations={C.AnnoComputePercentRequest: "80"})
+        assert m.parse(pod).qos == C.QosHigh
+        pod2 = mk_client_pod(name="b",
+                             annotations={C.AnnoComputePercentRequest: "30"})
+        assert m.parse(pod2).qos == C.QosMedium
+
+
+# -------------------------------------------------------- controllers
+
+
+class TestControllers:
+    def _mk_world(self):
+        store = Store()
+        mgr = ControllerManager(store)
+        for ctrl in default_controllers(store):
+            mgr.register(ctrl)
+        return store, mgr
+
+    def test_cluster_creates_pools(self):
+        store, mgr = self._mk_world()
+        cl = TensorFusionCluster()
+        cl.meta.name = "c1"
+        tmpl = GPUPool()
+        tmpl.meta.name = "pool-a"
+        cl.pools = [tmpl]
+        store.create(cl)
+        mgr.reconcile_now()
+        pool = store.get("GPUPool", "pool-a")
+        assert pool.cluster == "c1"
+        assert store.get("TensorFusionCluster", "c1").status.pool_count == 1
+
+    def test_node_to_gpunode_to_pool_capacity(self):
+        store, mgr = self._mk_world()
+        pool = GPUPool()
+        pool.meta.name = "pool-a"
+        store.create(pool)
+        node = Node()
+        node.meta.name = "node-a"
+        store.create(node)
+        mgr.reconcile_now()
+        gn = store.get("GPUNode", "node-a")
+        assert gn.pool == "pool-a"
+        for i in range(2):
+            store.create(mk_gpu(f"g{i}"))
+        mgr.reconcile_now()
+        gn = store.get("GPUNode", "node-a")
+        assert gn.status.gpu_count == 2
+        assert gn.status.total.vram == 2 * C.MI355X_VRAM_BYTES
+        pool = store.get("GPUPool", "pool-a")
+        assert pool.status.gpu_count == 2
+        # oversell: 500% tflops, +50% vram
+        assert pool.status.virtual_total.tflops == pytest.approx(2 * 2500 * 5)
+        assert pool.status.virtual_total.vram == pytest.approx(
+            2 * C.MI355X_VRAM_BYTES * 1.5)
+        # hypervisor pod created per node
+        hyp = store.get("Pod", "hypervisor-node-a", "tensor-fusion-sys")
+        assert hyp.meta.labels[C.LabelComponent] == C.ComponentHypervisor
+
+    def test_workload_scales_workers_and_connection_url(self):
+        store, mgr = self._mk_world()
+        wl = TensorFusionWorkload()
+        wl.meta.name = "wl1"
+        wl.meta.namespace = "default"
+        wl.replicas = 2
+        store.create(wl)
+        mgr.reconcile_now()
+        workers = [p for p in store.list("Pod", namespace="default")
+                   if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker]
+        assert len(workers) == 2
+        assert all(p.scheduler_name == C.SchedulerName for p in workers)
+
+        # mark one worker scheduled → connection picks it
+        def _sched(obj):
+            obj.status.phase = "Scheduled"
+            obj.status.pod_ip = "10.0.0.5"
+        store.patch("Pod", workers[0].meta.name, "default", _sched)
+        conn = TensorFusionConnection()
+        conn.meta.name = "c1"
+        conn.meta.namespace = "default"
+        conn.workload = "wl1"
+        store.create(conn)
+        mgr.reconcile_now()
+        conn = store.get("TensorFusionConnection", "c1", "default")
+        assert conn.status.connection_url.startswith("native+10.0.0.5+8000+")
+        assert conn.status.phase == "Ready"
+
+        # scale down to 1
+        def _scale(obj):
+            obj.replicas = 1
+        store.patch("TensorFusionWorkload", "wl1", "default", _scale)
+        mgr.reconcile_now()
+        workers = [p for p in store.list("Pod", namespace="default")
+                   if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker]
+        assert len(workers) == 1
+
+    def test_client_pod_gets_connection_created(self):
+        store, mgr = self._mk_world()
+        pod = mk_client_pod()
+        m = PodMutator(store)
+        m.handle(pod)
+        store.create(pod)
+        mgr.reconcile_now()
+        conn = store.get("TensorFusionConnection", "app-1-conn", "default")
+        assert conn.workload == "app-1-wl"
+
+    def test_pod_delete_deallocs(self):
+        store = Store()
+        alloc = GpuAllocator(store=store)
+        mgr = ControllerManager(store)
+        for ctrl in default_controllers(store, allocator=alloc):
+            mgr.register(ctrl)
+        store.create(mk_gpu("g0"))
+        from tensor_fusion_amd.api.types import AllocRequest
+        req = AllocRequest(pod_name="w1", namespace="default",
+                           request=Resource(100, 8 << 30, 10),
+                           limit=Resource(100, 8 << 30, 10))
+        alloc.assume(req, ["g0"])
+        alloc.commit("default/w1")
+        pod = Pod()
+        pod.meta.name = "w1"
+        pod.meta.namespace = "default"
+        store.create(pod)
+        store.delete("Pod", "w1", "default")
+        assert alloc.allocation("default/w1") is None
+        g = alloc.gpu("g0")
+        assert g.status.available.vram == C.MI355X_VRAM_BYTES
+
+    def test_worker_pod_annotations_carry_profile(self):
+        wl = TensorFusionWorkload()
+        wl.meta.name = "wl2"
+        wl.meta.namespace = "ns1"
+        wl.profile = WorkloadProfile()
+        wl.profile.resources.requests = Resource(600, 48 << 30, 25)
+        wl.profile.resources.limits = Resource(1200, 48 << 30, 50)
+        wl.profile.gang.enabled = True
+        wl.replicas = 4
+        pod = generate_worker_pod(wl, 0)
+        a = pod.meta.annotations
+        assert float(a[C.AnnoTflopsRequest]) == 600.0
+        assert a[C.AnnoGangEnabled] == "true"
+        assert a[C.AnnoGangMinMembers] == "4"
+
+
+# ------------------------------------------------------------- defrag
+
+
+class TestDefrag:
+    def _world(self):
+        store = Store()
+        alloc = GpuAllocator(store=store)
+        for node in ("node-a", "node-b"):
+            for i in range(2):
+                store.create(mk_gpu(f"{node}-g{i}", node=node))
+        return store, alloc
+
+    def test_campaign_consolidates_underutilized_node(self):
+        from tensor_fusion_amd.api.types import AllocRequest
+        store, alloc = self._world()
+        # node-a nearly full, node-b one small pod
+        req_big = AllocRequest(pod_name="big", namespace="d",
+                               request=Resource(2000, 200 << 30, 80),
+                               limit=Resource(2000, 200 << 30, 80))
+        alloc.assume(req_big, ["node-a-g0"])
+        alloc.commit("d/big")
+        req_small = AllocRequest(pod_name="small", namespace="d",
+                                 request=Resource(100, 8 << 30, 5),
+                                 limit=Resource(100, 8 << 30, 5))
+        alloc.assume(req_small, ["node-b-g0"])
+        alloc.commit("d/small")
+        pod = Pod()
+        pod.meta.name = "small"
+        pod.meta.namespace = "d"
+        store.create(pod)
+
+        d = DefragController(store, alloc, utilization_threshold=0.3,
+                             eviction_ttl_s=0.0, campaign_cooldown_s=0.0)
+        plan = d.run_campaign()
+        assert plan is not None
+        assert plan.candidate_nodes == ["node-b"]
+        assert plan.evict_pods == ["d/small"]
+        marked = store.get("Pod", "small", "d")
+        from tensor_fusion_amd.controllers.defrag import AnnoEvictionMark
+        assert AnnoEvictionMark in marked.meta.annotations
+        evicted = d.execute_due_evictions(now=time.time() + 1)
+        assert evicted == ["d/small"]
+
+    def test_no_campaign_when_everything_busy(self):
+        from tensor_fusion_amd.api.types import AllocRequest
+        store, alloc = self._world()
+        for node in ("node-a", "node-b"):
+            req = AllocRequest(pod_name=f"p-{node}", namespace="d",
+                               request=Resource(2000, 200 << 30, 80),
+                               limit=Resource(2000, 200 << 30, 80))
+            alloc.assume(req, [f"{node}-g0"])
+            alloc.commit(f"d/p-{node}")
+        d = DefragController(store, alloc, utilization_threshold=0.3,
+                             campaign_cooldown_s=0.0)
+        assert d.run_campaign() is None
+
+
+# ------------------------------------------------------------ rollout
+
+
+class TestRollout:
+    def test_batch_rolling_update(self):
+        from tensor_fusion_amd.component import ComponentRollout
+        store = Store()
+        tmpl_v1 = {"image": "worker:v1"}
+        roll = ComponentRollout(store, C.ComponentWorker, batch_percent=50,
+                                interval_s=0.0)
+        for i in range(4):
+            p = Pod()
+            p.meta.name = f"w{i}"
+            p.meta.namespace = "d"
+            p.meta.labels[C.LabelComponent] = C.ComponentWorker
+            roll.stamp(p, tmpl_v1)
+            store.create(p)
+        assert roll.out_of_date(tmpl_v1) == []
+        tmpl_v2 = {"image": "worker:v2"}
+        assert len(roll.out_of_date(tmpl_v2)) == 4
+        deleted = roll.tick(tmpl_v2)
+        assert len(deleted) == 2  # 50% batch
+        # batch is computed from the remaining pod count (2 left -> 1)
+        roll._last_batch_ts = 0.0
+        assert len(roll.tick(tmpl_v2)) == 1
+        roll._last_batch_ts = 0.0
+        assert len(roll.tick(tmpl_v2)) == 1
+        roll._last_batch_ts = 0.0
+        assert roll.tick(tmpl_v2) == []
+
+
+# --------------------------------------------------- expander/provider
+
+
+class TestExpander:
+    def test_unschedulable_creates_claim_then_node_joins(self):
+        from tensor_fusion_amd.api.types import AllocRequest
+        from tensor_fusion_amd.cloudprovider import MockProvider
+        from tensor_fusion_amd.controllers import (ControllerManager,
+                                                   NodeClaimReconciler)
+        from tensor_fusion_amd.scheduler.expander import NodeExpander
+        store = Store()
+        provider = MockProvider(polls_until_ready=1, store=store)
+        mgr = ControllerManager(store)
+        mgr.register(NodeClaimReconciler(store, provider=provider))
+        exp = NodeExpander(store, cooldown_s=0.0)
+        req = AllocRequest(pod_name="p1", namespace="d", gpu_count=2,
+                           request=Resource(1000, 100 << 30, 50),
+                           limit=Resource(1000, 100 << 30, 50))
+        claim_name = exp.handle_unschedulable(req)
+        assert claim_name
+        claim = store.get("GPUNodeClaim", claim_name)
+        assert claim.instance_type == "mi355x.2g"
+        # second call rides the same claim
+        assert exp.handle_unschedulable(req) == claim_name
+        mgr.reconcile_now()
+        # claim reconciler drives Pending→Creating→Bound (requeue ticks)
+        deadline = time.time() + 5
+        while time.time() < deadline:
+            mgr.reconcile_now()
+            if store.get("GPUNodeClaim", claim_name).status.phase == "Bound":
+                break
+            time.sleep(0.05)
+        claim = store.get("GPUNodeClaim", claim_name)
+        assert claim.status.phase == "Bound"
+        assert store.try_get("Node", claim.status.node_name) is not None
+
+    def test_cheapest_instance(self):
+        from tensor_fusion_amd.cloudprovider import cheapest_instance_for
+        it = cheapest_instance_for(1, 2000, 100 << 30)
+        assert it.name == "mi355x.1g"
+        assert cheapest_instance_for(16, 0, 0) is None

@@ -1,0 +1,165 @@
+"""Metrics/TSDB/autoscaler/alert/operator-server/config tests."""
+import os
+import time
+
+import pytest
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.alert import AlertEvaluator, AlertRule
+from tensor_fusion_amd.api.store import Store
+from tensor_fusion_amd.api.types import (Resource, TensorFusionConnection,
+                                         TensorFusionWorkload)
+from tensor_fusion_amd.autoscaler import (Autoscaler, CronRecommender,
+                                          DecayingHistogram,
+                                          PercentileRecommender)
+from tensor_fusion_amd.metrics import (MetricsRecorder, TSDB, WorkerMetrics,
+                                       parse_influx_line)
+from tensor_fusion_amd.server import create_operator_app, make_token
+
+
+class TestMetrics:
+    def test_influx_roundtrip(self):
+        rec = MetricsRecorder()
+        m = WorkerMetrics(workload="wl", worker="w0", namespace="d",
+                          pool="p", qos=C.QosHigh, device_uuid="u0",
+                          compute_percent=42.5, compute_tflops=600.0,
+                          vram_bytes=8 << 30)
+        rec.set_worker(m)
+        lines = rec.encode_all(ts_ns=123)
+        worker_lines = [l for l in lines if l.startswith("tf_worker_metrics")]
+        assert len(worker_lines) == 1
+        meas, tags, fields, ts = parse_influx_line(worker_lines[0])
+        assert tags["workload"] == "wl" and tags["qos"] == "high"
+        assert fields["compute_percent"] == 42.5
+        assert ts == 123
+        # billing: high-QoS pricing applied
+        assert fields["raw_cost"] == pytest.approx(
+            0.08 * 600 + 0.016 * 8, rel=1e-6)
+
+    def test_rolling_file_and_tsdb_ingest(self, tmp_path):
+        tsdb = TSDB()
+        rec = MetricsRecorder(out_dir=str(tmp_path), tsdb=tsdb)
+        rec.set_worker(WorkerMetrics(workload="wl", worker="w0",
+                                     namespace="d", device_uuid="u0",
+                                     compute_tflops=100.0))
+        n = rec.flush()
+        assert n >= 1
+        assert os.path.exists(tmp_path / "metrics.log")
+        pts = tsdb.query("tf_worker_metrics", "compute_tflops",
+                         tags={"workload": "wl"})
+        assert len(pts) == 1 and pts[0][1] == 100.0
+
+
+class TestAutoscaler:
+    def test_histogram_percentile(self):
+        h = DecayingHistogram(half_life_s=1e9)  # no decay
+        for v in [1.0] * 90 + [10.0] * 10:
+            h.add(v, ts=h.ref_ts)
+        assert h.percentile(0.5) == pytest.approx(1.0, rel=0.1)
+        assert h.percentile(0.95) == pytest.approx(10.0, rel=0.15)
+
+    def test_percentile_recommender_from_tsdb(self):
+        tsdb = TSDB()
+        rec = MetricsRecorder(tsdb=tsdb)
+        for i in range(20):
+            rec.set_worker(WorkerMetrics(
+                workload="wl1", worker="w0", namespace="d", device_uuid="u0",
+                compute_tflops=500.0, vram_bytes=40 << 30))
+            rec.flush()
+        store = Store()
+        wl = TensorFusionWorkload()
+        wl.meta.name = "wl1"
+        wl.meta.namespace = "d"
+        wl.profile.auto_scaling.enabled = True
+        wl.profile.auto_scaling.recommender = "percentile"
+        store.create(wl)
+        a = Autoscaler(store, tsdb=tsdb, apply_in_place=False)
+        assert a.tick() == 1
+        got = store.get("TensorFusionWorkload", "wl1", "d")
+        r = got.status.recommendation
+        assert r is not None
+        # p90 of constant 500 with 15% margin ∈ [500, 650]
+        assert 450 <= r.resources.requests.tflops <= 700
+        assert r.resources.requests.vram >= 35 << 30
+
+    def test_cron_recommender_window(self):
+        wl = TensorFusionWorkload()
+        wl.profile.auto_scaling.cron_rules = [
+            {"start": "00:00", "end": "23:59", "tflops": 777, "vram": 123}]
+        r = CronRecommender().recommend(wl)
+        assert r is not None and r.resources.requests.tflops == 777
+
+
+class TestAlert:
+    def test_rule_fires_and_resolves(self):
+        tsdb = TSDB()
+        rule = AlertRule(name="T", query="SELECT max(value) FROM points "
+                         "WHERE field='x'", threshold=5.0, interval_s=0.0)
+        ev = AlertEvaluator(tsdb, rules=[rule])
+        tsdb.ingest_lines([f"m x=3.0 {time.time_ns()}"])
+        assert ev.evaluate() == []  # below threshold, no transition
+        tsdb.ingest_lines([f"m x=9.0 {time.time_ns()}"])
+        changed = ev.evaluate()
+        assert len(changed) == 1 and changed[0].firing
+        assert ev.posted[-1]["status"] == "firing"
+
+
+class TestOperatorServer:
+    def test_connection_lookup_with_auth(self):
+        from fastapi.testclient import TestClient
+        store = Store()
+        conn = TensorFusionConnection()
+        conn.meta.name = "c1"
+        conn.meta.namespace = "d"
+        conn.status.connection_url = "native+10.0.0.9+8000+w-3"
+        store.create(conn)
+        app = create_operator_app(store, long_poll_s=0.2)
+        c = TestClient(app)
+        r = c.get("/connection", params={"name": "c1", "namespace": "d"})
+        assert r.status_code == 401
+        tok = make_token("tf-dev-secret", "d", "app-1")
+        r = c.get("/connection", params={"name": "c1", "namespace": "d"},
+                  headers={"Authorization": f"Bearer {tok}"})
+        assert r.status_code == 200
+        assert r.json()["connectionURL"] == "native+10.0.0.9+8000+w-3"
+        # cross-namespace denied
+        r = c.get("/connection", params={"name": "c1", "namespace": "other"},
+                  headers={"Authorization": f"Bearer {tok}"})
+        assert r.status_code == 403
+
+    def test_bad_token_rejected(self):
+        from fastapi.testclient import TestClient
+        app = create_operator_app(Store(), long_poll_s=0.1)
+        c = TestClient(app)
+        r = c.get("/connection", params={"name": "x", "namespace": "d"},
+                  headers={"Authorization": "Bearer d:app:deadbeef"})
+        assert r.status_code == 401
+
+
+class TestConfig:
+    def test_hot_reload(self, tmp_path):
+        from tensor_fusion_amd.config import ConfigWatcher
+        p = tmp_path / "config.yaml"
+        p.write_text("gpuFit:\n  vramWeight: 0.9\n")
+        w = ConfigWatcher(str(p))
+        assert w.config.gpu_fit.vram_weight == 0.9
+        seen = []
+        w.on_change(lambda cfg: seen.append(cfg.gpu_fit.vram_weight))
+        time.sleep(0.01)
+        os.utime(p, (time.time() + 5, time.time() + 5))
+        p.write_text("gpuFit:\n  vramWeight: 0.4\n")
+        os.utime(p, (time.time() + 10, time.time() + 10))
+        assert w.reload() is True
+        assert w.config.gpu_fit.vram_weight == 0.4
+        assert seen == [0.4]
+
+    def test_gpu_info_table(self, tmp_path):
+        from tensor_fusion_amd.config import ConfigWatcher
+        p = tmp_path / "config.yaml"
+        p.write_text(
+            "gpuInfo:\n- model: MI300X\n  fp16TFlops: 1300\n"
+            "  vramBytes: 206158430208\n")
+        w = ConfigWatcher(str(p))
+        assert "MI300X" in w.config.gpu_info
+        assert w.config.gpu_info["MI300X"].fp16_tflops == 1300
+        assert "MI355X" in w.config.gpu_info  # default retained
