@@ -38,6 +38,7 @@ def run_child(mode: str, args, local_rank: int) -> dict:
     for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT",
               "GROUP_RANK", "LOCAL_WORLD_SIZE", "TORCHELASTIC_RUN_ID"):
         env.pop(k, None)
+    worker = None
     if mode == "vgpu":
         vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "limiter")
         if vgpu_mode == "limiter":
@@ -46,13 +47,25 @@ def run_child(mode: str, args, local_rank: int) -> dict:
             env["TF_UP_LIMIT_PERCENT"] = "100"  # full vGPU of the device
             env["TF_VRAM_LIMIT_BYTES"] = str(288 << 30)
         elif vgpu_mode == "remote":
-            env["TF_BENCH_REMOTE"] = "1"  # remoting client path (stage 6)
+            # GPU-over-IP: the child is GPU-less; a per-rank vGPU worker
+            # owns the device and executes the forwarded HIP stream.
+            sys.path.insert(0, REPO)
+            from tensor_fusion_amd.client.runtime import (client_env,
+                                                          start_worker)
+            sock = f"/tmp/tf-bench-vgpu-{os.getpid()}-{local_rank}.sock"
+            worker = start_worker(
+                sock, device_index=int(env["HIP_VISIBLE_DEVICES"] or "0"))
+            env = client_env(sock, base=env)
     cmd = [sys.executable, "-m", "tensor_fusion_amd.models.llama",
            "--model", args.model, "--batch", str(args.batch),
            "--ctx", str(args.ctx), "--steps", str(args.steps),
            "--warmup", str(args.warmup)]
-    out = subprocess.run(cmd, env=env, cwd=REPO, capture_output=True,
-                         text=True, timeout=3600)
+    try:
+        out = subprocess.run(cmd, env=env, cwd=REPO, capture_output=True,
+                             text=True, timeout=3600)
+    finally:
+        if worker is not None:
+            worker.stop()
     if out.returncode != 0:
         raise RuntimeError(f"{mode} child failed:\n{out.stdout}\n{out.stderr}")
     return json.loads(out.stdout.strip().splitlines()[-1])

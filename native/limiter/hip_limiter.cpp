@@ -69,6 +69,12 @@ struct AllocMap {
   std::unordered_map<const void*, size_t> m;
 };
 
+// One oversubscribed (managed-memory) allocation and its current tier.
+struct ExpandedRange {
+  size_t bytes = 0;
+  bool device_resident = false;  // last prefetch target was HBM
+};
+
 struct Limiter {
   TfSharedState* shm = nullptr;  // mapped page (file or private standalone)
   bool standalone = false;
@@ -79,6 +85,19 @@ struct Limiter {
   double tokens_per_memcpy = 1.0;
   AllocMap allocs;
   std::atomic<long> blocked_threads{0};
+
+  // ---- VRAM oversubscription (BASELINE config 4) ----
+  // Over-cap hipMalloc falls back to HSA managed memory placed in host
+  // DRAM (preferred-location CPU, accessed-by GPU): the workload keeps
+  // running past the HBM cap; a tier thread promotes ranges into free
+  // HBM headroom and demotes them on hypervisor VRAM-pressure signal.
+  bool expand_enabled = false;
+  uint64_t expand_limit = 0;  // 0 = unlimited host expansion
+  std::mutex expand_mu;
+  std::unordered_map<const void*, ExpandedRange> expanded;
+  std::atomic<uint64_t> expanded_bytes{0};
+  std::atomic<uint64_t> promoted_bytes{0};
+  std::atomic<int> tier_thread_running{0};
 
   Limiter();  // all init lives in the constructor: the instance is a Meyers
               // singleton so there is no static-init-order hazard between
@@ -240,10 +259,130 @@ void record_free(const void* p) {
   tfshm::at(&e->pod_memory_used)->fetch_sub(bytes, std::memory_order_relaxed);
 }
 
+// ----------------------------------------------------- VRAM expansion
+
+// Real-HIP entry points the expansion path needs (resolved lazily so the
+// limiter still loads in GPU-less CI). resolve_real is defined below.
+void* resolve_real(const char* name);
+
+struct ExpandHip {
+  hipError_t (*MallocManaged)(void**, size_t, unsigned) = nullptr;
+  hipError_t (*MemAdvise)(const void*, size_t, int, int) = nullptr;
+  hipError_t (*MemPrefetchAsync)(const void*, size_t, int, void*) = nullptr;
+  hipError_t (*StreamSynchronize)(void*) = nullptr;
+  bool ok = false;
+  void init() {
+    MallocManaged = (decltype(MallocManaged))resolve_real("hipMallocManaged");
+    MemAdvise = (decltype(MemAdvise))resolve_real("hipMemAdvise");
+    MemPrefetchAsync =
+        (decltype(MemPrefetchAsync))resolve_real("hipMemPrefetchAsync");
+    StreamSynchronize =
+        (decltype(StreamSynchronize))resolve_real("hipStreamSynchronize");
+    ok = MallocManaged && MemAdvise && MemPrefetchAsync;
+  }
+};
+ExpandHip& ehip() {
+  static ExpandHip e;
+  return e;
+}
+
+static const int kHipCpuDeviceId = -1;
+static const int kAdviseSetPreferredLocation = 3;
+static const int kAdviseSetAccessedBy = 5;
+
+// Allocate an over-cap range in host-DRAM-backed managed memory.
+hipError_t expand_alloc(void** p, size_t sz) {
+  ExpandHip& eh = ehip();
+  if (!eh.MallocManaged) eh.init();
+  if (!eh.ok) return hipErrorOutOfMemory;
+  uint64_t lim = g.expand_limit;
+  if (lim && g.expanded_bytes.load(std::memory_order_relaxed) + sz > lim)
+    return hipErrorOutOfMemory;  // host expansion budget exhausted too
+  hipError_t r = eh.MallocManaged(p, sz, 1 /*hipMemAttachGlobal*/);
+  if (r != hipSuccess) return r;
+  // host tier: pages prefer DRAM; GPU keeps direct access (no fault storm
+  // on XNACK-disabled parts — fine-grain host access over PCIe).
+  eh.MemAdvise(*p, sz, kAdviseSetPreferredLocation, kHipCpuDeviceId);
+  eh.MemAdvise(*p, sz, kAdviseSetAccessedBy, tls_device);
+  {
+    std::lock_guard<std::mutex> l(g.expand_mu);
+    g.expanded[*p] = ExpandedRange{sz, false};
+  }
+  g.expanded_bytes.fetch_add(sz, std::memory_order_relaxed);
+  dbg("expand_alloc %zu B -> host tier (total expanded %lu)", sz,
+      g.expanded_bytes.load());
+  return hipSuccess;
+}
+
+// Returns bytes whose residency changed. to_device: promote cold→HBM up to
+// `budget` bytes; else demote device-resident ranges back to host DRAM.
+uint64_t tier_migrate(bool to_device, uint64_t budget) {
+  ExpandHip& eh = ehip();
+  if (!eh.ok) return 0;
+  uint64_t moved = 0;
+  std::lock_guard<std::mutex> l(g.expand_mu);
+  for (auto& [ptr, r] : g.expanded) {
+    if (to_device == r.device_resident) continue;
+    if (to_device && moved + r.bytes > budget) continue;
+    int dst = to_device ? tls_device : kHipCpuDeviceId;
+    eh.MemAdvise(ptr, r.bytes, kAdviseSetPreferredLocation, dst);
+    if (eh.MemPrefetchAsync(ptr, r.bytes, dst, nullptr) == hipSuccess) {
+      r.device_resident = to_device;
+      moved += r.bytes;
+      if (!to_device && moved >= budget && budget) break;
+    }
+  }
+  if (moved && eh.StreamSynchronize) eh.StreamSynchronize(nullptr);
+  if (to_device)
+    g.promoted_bytes.fetch_add(moved, std::memory_order_relaxed);
+  else
+    g.promoted_bytes.fetch_sub(
+        moved > g.promoted_bytes.load() ? g.promoted_bytes.load() : moved,
+        std::memory_order_relaxed);
+  return moved;
+}
+
+void* tier_thread_main(void*) {
+  // Hypervisor-driven loop: VRAM pressure flag → demote; otherwise promote
+  // expanded ranges into free cap headroom (hot data earns HBM residency).
+  while (true) {
+    timespec ts{0, 250000000};  // 250 ms
+    nanosleep(&ts, nullptr);
+    if (!g.shm) continue;
+    uint32_t flags = tfshm::at(&g.shm->flags)->load(std::memory_order_relaxed);
+    if (flags & TF_FLAG_VRAM_PRESSURE) {
+      tier_migrate(false, 0 /*all*/);
+      continue;
+    }
+    TfDeviceEntry* e = cur_dev();
+    if (!e) continue;
+    uint64_t lim = tfshm::at(&e->mem_limit_bytes)->load(std::memory_order_relaxed);
+    uint64_t used = tfshm::at(&e->pod_memory_used)->load(std::memory_order_relaxed);
+    uint64_t promoted = g.promoted_bytes.load(std::memory_order_relaxed);
+    if (lim > used + promoted) {
+      uint64_t headroom = lim - used - promoted;
+      if (headroom > (64u << 20)) tier_migrate(true, headroom);
+    }
+  }
+  return nullptr;
+}
+
+void ensure_tier_thread() {
+  int expect = 0;
+  if (g.tier_thread_running.compare_exchange_strong(expect, 1)) {
+    pthread_t t;
+    pthread_create(&t, nullptr, tier_thread_main, nullptr);
+    pthread_detach(t);
+  }
+}
+
 // ----------------------------------------------------------------- init
 
 Limiter::Limiter() {
   debug = getenv("TF_LIMITER_DEBUG") != nullptr;
+  expand_enabled = getenv("TF_VRAM_EXPAND") != nullptr &&
+                   atoi(getenv("TF_VRAM_EXPAND")) != 0;
+  expand_limit = (uint64_t)env_f("TF_VRAM_EXPAND_LIMIT_BYTES", 0);
   tokens_per_launch = env_f("TF_TOKENS_PER_LAUNCH", 1.0);
   tokens_per_graph = env_f("TF_TOKENS_PER_GRAPH", 8.0);
   tokens_per_memcpy = env_f("TF_TOKENS_PER_MEMCPY", 1.0);
@@ -387,7 +526,15 @@ int hipSetDevice(int dev) {
 // ----- memory
 hipError_t hipMalloc(void** p, size_t sz) {
   REAL(hipError_t, hipMalloc, void**, size_t);
-  if (g.enabled && !admit_alloc(sz)) return hipErrorOutOfMemory;
+  if (g.enabled && !admit_alloc(sz)) {
+    if (g.expand_enabled) {
+      // VRAM oversubscription: land the range in the host-DRAM tier
+      hipError_t r = expand_alloc(p, sz);
+      if (r == hipSuccess) ensure_tier_thread();
+      return r;
+    }
+    return hipErrorOutOfMemory;
+  }
   hipError_t r = call_hipMalloc(p, sz);
   if (g.enabled && r == hipSuccess) record_alloc(*p, sz);
   return r;
@@ -420,7 +567,18 @@ hipError_t hipMallocPitch(void** p, size_t* pitch, size_t w, size_t h) {
 hipError_t hipFree(void* p) {
   REAL(hipError_t, hipFree, void*);
   hipError_t r = call_hipFree(p);
-  if (g.enabled && r == hipSuccess) record_free(p);
+  if (g.enabled && r == hipSuccess) {
+    record_free(p);
+    std::lock_guard<std::mutex> l(g.expand_mu);
+    auto it = g.expanded.find(p);
+    if (it != g.expanded.end()) {
+      g.expanded_bytes.fetch_sub(it->second.bytes, std::memory_order_relaxed);
+      if (it->second.device_resident)
+        g.promoted_bytes.fetch_sub(it->second.bytes,
+                                   std::memory_order_relaxed);
+      g.expanded.erase(it);
+    }
+  }
   return r;
 }
 
@@ -587,6 +745,37 @@ void tf_limiter_freeze(int on) {
   if (g.shm) {
     uint32_t f = g.shm->flags;
     g.shm->flags = on ? (f | TF_FLAG_FREEZE) : (f & ~TF_FLAG_FREEZE);
+  }
+}
+
+// ---- VRAM tiering introspection / control (tests + fake amd-smi) ----
+
+int tf_limiter_tier_stats(unsigned long long* expanded,
+                          unsigned long long* promoted,
+                          unsigned* n_ranges) {
+  if (expanded) *expanded = g.expanded_bytes.load(std::memory_order_relaxed);
+  if (promoted) *promoted = g.promoted_bytes.load(std::memory_order_relaxed);
+  if (n_ranges) {
+    std::lock_guard<std::mutex> l(g.expand_mu);
+    *n_ranges = (unsigned)g.expanded.size();
+  }
+  return g.expand_enabled ? 1 : 0;
+}
+
+// Force-demote every expanded range to host DRAM (pressure-trap path);
+// returns bytes moved.
+unsigned long long tf_limiter_demote_all() { return tier_migrate(false, 0); }
+
+// Promote up to budget bytes into HBM; returns bytes moved.
+unsigned long long tf_limiter_promote(unsigned long long budget) {
+  return tier_migrate(true, budget);
+}
+
+void tf_limiter_set_vram_pressure(int on) {
+  if (g.shm) {
+    uint32_t f = g.shm->flags;
+    g.shm->flags =
+        on ? (f | TF_FLAG_VRAM_PRESSURE) : (f & ~TF_FLAG_VRAM_PRESSURE);
   }
 }
 

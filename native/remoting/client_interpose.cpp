@@ -512,7 +512,12 @@ hipError_t hipDeviceCanAccessPeer(int* can, int dev, int peer) {
 }
 
 hipError_t hipDeviceSynchronize(void) {
-  return send_sync(OP_DEVICE_SYNC, nullptr, 0, nullptr, 0);
+  hipError_t e = send_sync(OP_DEVICE_SYNC, nullptr, 0, nullptr, 0);
+  if (e == hipSuccess) {
+    // surface async launch/copy failures at sync points (CUDA semantics)
+    e = (hipError_t)at(&C().hdr->sticky_error)->exchange(0);
+  }
+  return e;
 }
 
 hipError_t hipDriverGetVersion(int* v) {
@@ -849,7 +854,10 @@ hipError_t hipStreamDestroy(void* s) {
   return hipSuccess;
 }
 hipError_t hipStreamSynchronize(void* s) {
-  return send_sync(OP_STREAM_SYNC, &s, 8, nullptr, 0);
+  hipError_t e = send_sync(OP_STREAM_SYNC, &s, 8, nullptr, 0);
+  if (e == hipSuccess)
+    e = (hipError_t)at(&C().hdr->sticky_error)->exchange(0);
+  return e;
 }
 hipError_t hipStreamQuery(void* s) {
   return send_sync(OP_STREAM_QUERY, &s, 8, nullptr, 0);
@@ -1242,5 +1250,154 @@ hipError_t hipDevicePrimaryCtxRetain(void** ctx, int) {
 }
 
 int tf_client_connected(void) { return C().connected ? 1 : 0; }
+
+// ------------------------------------------------- call configuration
+// <<<>>> launches compile to __hipPushCallConfiguration + stub +
+// __hipPopCallConfiguration + hipLaunchKernel. These MUST be interposed:
+// the real runtime's implementations fail (or return garbage) in a
+// GPU-less process, which silently corrupts every launch config.
+
+struct CallConfig {
+  dim3u grid, block;
+  size_t shmem;
+  void* stream;
+};
+
+static thread_local std::vector<CallConfig> tls_call_stack;
+
+hipError_t __hipPushCallConfiguration(dim3u grid, dim3u block, size_t shmem,
+                                      void* stream) {
+  tls_call_stack.push_back({grid, block, shmem, stream});
+  return hipSuccess;
+}
+
+hipError_t __hipPopCallConfiguration(dim3u* grid, dim3u* block, size_t* shmem,
+                                     void** stream) {
+  if (tls_call_stack.empty()) return hipErrorInvalidValue;
+  CallConfig c = tls_call_stack.back();
+  tls_call_stack.pop_back();
+  if (grid) *grid = c.grid;
+  if (block) *block = c.block;
+  if (shmem) *shmem = c.shmem;
+  if (stream) *stream = c.stream;
+  return hipSuccess;
+}
+
+// ------------------------------------------------- small local shims
+
+const char* hipGetErrorName(hipError_t e) {
+  switch (e) {
+    case 0: return "hipSuccess";
+    case 1: return "hipErrorInvalidValue";
+    case 2: return "hipErrorOutOfMemory";
+    case 3: return "hipErrorNotInitialized";
+    case 98: return "hipErrorInvalidDeviceFunction";
+    case 100: return "hipErrorNoDevice";
+    case 101: return "hipErrorInvalidDevice";
+    case 600: return "hipErrorNotReady";
+    case 801: return "hipErrorNotSupported";
+    default: return "hipErrorUnknown(remote)";
+  }
+}
+
+const char* hipGetErrorString(hipError_t e) { return hipGetErrorName(e); }
+
+hipError_t hipDeviceGetPCIBusId(char* id, int len, int dev) {
+  if (!id || len < 13) return hipErrorInvalidValue;
+  snprintf(id, len, "0000:%02x:00.0", dev & 0xff);
+  return hipSuccess;
+}
+
+hipError_t hipGetStreamDeviceId(void*) { return Client::cur_device; }
+
+struct hipFuncAttributes_small {
+  int binaryVersion, cacheModeCA;
+  size_t constSizeBytes, localSizeBytes;
+  int maxDynamicSharedSizeBytes, maxThreadsPerBlock, numRegs;
+  int preferredShmemCarveout, ptxVersion;
+  size_t sharedSizeBytes;
+};
+
+hipError_t hipFuncGetAttributes(hipFuncAttributes_small* a, const void*) {
+  if (!a) return hipErrorInvalidValue;
+  memset(a, 0, sizeof *a);
+  a->maxThreadsPerBlock = 1024;
+  a->maxDynamicSharedSizeBytes = 160 * 1024 - 16;  // CDNA4 LDS
+  a->numRegs = 64;
+  a->binaryVersion = 950;
+  return hipSuccess;
+}
+
+hipError_t hipOccupancyMaxPotentialBlockSize(int* minGrid, int* blockSize,
+                                             const void*, size_t, int) {
+  if (minGrid) *minGrid = 2048;  // 256 CUs × 8 blocks
+  if (blockSize) *blockSize = 256;
+  return hipSuccess;
+}
+
+hipError_t hipThreadExchangeStreamCaptureMode(int* mode) {
+  if (mode) *mode = 0;
+  return hipSuccess;
+}
+
+hipError_t hipDeviceEnablePeerAccess(int, unsigned) {
+  return hipErrorNotSupported;  // one vGPU per worker; RCCL rides xGMI
+}
+
+hipError_t hipStreamCreateWithFlags(void**, unsigned);  // fwd decl (above)
+
+hipError_t hipExtStreamCreateWithCUMask(void** stream, unsigned,
+                                        const unsigned*) {
+  // CU masking of a remote vGPU is enforced worker-side by the hypervisor
+  // (HSA_CU_MASK on the worker process); the client gets a normal stream.
+  return hipStreamCreateWithFlags(stream, 0);
+}
+
+hipError_t hipExtStreamGetCUMask(void*, unsigned n, unsigned* mask) {
+  if (!mask) return hipErrorInvalidValue;
+  for (unsigned i = 0; i < n; ++i) mask[i] = 0xffffffffu;
+  return hipSuccess;
+}
+
+// not-yet-supported surfaces: fail loudly rather than run locally
+#define TF_NOTSUP(name, ...)                         \
+  hipError_t name(__VA_ARGS__) {                     \
+    fprintf(stderr, "[tf-client] %s: not supported over remoting yet\n", \
+            #name);                                  \
+    return hipErrorNotSupported;                     \
+  }
+
+TF_NOTSUP(hipGetSymbolAddress, void**, const void*)
+TF_NOTSUP(hipMemcpyToSymbol, const void*, const void*, size_t, size_t, int)
+TF_NOTSUP(hipIpcGetMemHandle, void*, void*)
+TF_NOTSUP(hipIpcOpenMemHandle, void**, const void*, unsigned)
+TF_NOTSUP(hipIpcCloseMemHandle, void*)
+TF_NOTSUP(hipMemAddressReserve, void**, size_t, size_t, void*, unsigned long long)
+TF_NOTSUP(hipMemCreate, void*, size_t, const void*, unsigned long long)
+TF_NOTSUP(hipMemMap, void*, size_t, size_t, void*, unsigned long long)
+TF_NOTSUP(hipMemUnmap, void*, size_t)
+TF_NOTSUP(hipMemRelease, void*)
+TF_NOTSUP(hipMemSetAccess, void*, size_t, const void*, size_t)
+TF_NOTSUP(hipMemGetAllocationGranularity, size_t*, const void*, int)
+TF_NOTSUP(hipMemExportToShareableHandle, void*, void*, int, unsigned long long)
+TF_NOTSUP(hipMemImportFromShareableHandle, void*, void*, int)
+TF_NOTSUP(hipDeviceGetDefaultMemPool, void**, int)
+TF_NOTSUP(hipMemPoolGetAttribute, void*, int, void*)
+TF_NOTSUP(hipMemPoolSetAttribute, void*, int, void*)
+TF_NOTSUP(hipMemPoolSetAccess, void*, const void*, size_t)
+TF_NOTSUP(hipMemPoolTrimTo, void*, size_t)
+TF_NOTSUP(hipMemcpyPeerAsync, void*, int, const void*, int, size_t, void*)
+TF_NOTSUP(hipStreamWriteValue32, void*, void*, unsigned, unsigned)
+TF_NOTSUP(hipStreamBeginCapture, void*, int)
+TF_NOTSUP(hipStreamEndCapture, void*, void**)
+TF_NOTSUP(hipGraphInstantiate, void**, void*, void*, char*, size_t)
+TF_NOTSUP(hipGraphInstantiateWithFlags, void**, void*, unsigned long long)
+TF_NOTSUP(hipGraphLaunch, void*, void*)
+TF_NOTSUP(hipGraphDestroy, void*)
+TF_NOTSUP(hipGraphExecDestroy, void*)
+TF_NOTSUP(hipGraphGetNodes, void*, void**, size_t*)
+TF_NOTSUP(hipGraphNodeGetDependencies, void*, void**, size_t*)
+TF_NOTSUP(hipGraphDebugDotPrint, void*, const char*, unsigned)
+#undef TF_NOTSUP
 
 }  // extern "C"

@@ -149,8 +149,10 @@ struct Worker {
 
 Worker W;
 
-void set_sticky(hipError_t e) {
+void set_sticky(hipError_t e, const char* what = "") {
   if (e != 0) {
+    if (W.verbose)
+      fprintf(stderr, "[worker] %s failed: hipError %d\n", what, e);
     uint64_t expect = 0;
     tfrpc::at(&W.hdr->sticky_error)
         ->compare_exchange_strong(expect, (uint64_t)e);
@@ -213,7 +215,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&dev, body, 4);
       hipError_t e = hip.SetDevice(dev);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "async-op");
       break;
     }
     case OP_GET_PROPS: {
@@ -247,7 +249,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&p, body, 8);
       hipError_t e = hip.Free((void*)p);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "async-op");
       break;
     }
     case OP_MEMCPY_H2D: {
@@ -270,7 +272,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
         track_arena_use(m->arena_off + m->size, (hipStream_t)m->stream);
       }
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "async-op");
       break;
     }
     case OP_MEMCPY_D2H: {
@@ -287,7 +289,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       hipError_t e = hip.MemcpyAsync((void*)m->dst, (const void*)m->src,
                                      m->size, 3, (hipStream_t)m->stream);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "async-op");
       break;
     }
     case OP_MEMSET: {
@@ -295,7 +297,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       hipError_t e = hip.MemsetD8Async((void*)m->dst, (unsigned char)m->kind,
                                        m->size, (hipStream_t)m->stream);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "async-op");
       break;
     }
     case OP_LAUNCH: {
@@ -309,8 +311,15 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
           (hipFunction_t)l->func, l->grid[0], l->grid[1], l->grid[2],
           l->block[0], l->block[1], l->block[2], l->shmem,
           (hipStream_t)l->stream, nullptr, extra);
+      if (W.verbose)
+        fprintf(stderr,
+                "[worker] launch fn=%llx grid=%u,%u,%u block=%u,%u,%u "
+                "kernarg=%u stream=%llx -> %d\n",
+                (unsigned long long)l->func, l->grid[0], l->grid[1],
+                l->grid[2], l->block[0], l->block[1], l->block[2],
+                l->kernarg_size, (unsigned long long)l->stream, e);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e);
+      else set_sticky(e, "ModuleLaunchKernel");
       break;
     }
     case OP_STREAM_CREATE: {
@@ -327,7 +336,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_STREAM_DESTROY: {
       uint64_t st;
       memcpy(&st, body, 8);
-      set_sticky(hip.StreamDestroy((hipStream_t)st));
+      set_sticky(hip.StreamDestroy((hipStream_t)st), "StreamDestroy");
       break;
     }
     case OP_STREAM_SYNC: {
@@ -356,7 +365,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       uint64_t ev, st;
       memcpy(&ev, body, 8);
       memcpy(&st, body + 8, 8);
-      set_sticky(hip.EventRecord((hipEvent_t)ev, (hipStream_t)st));
+      set_sticky(hip.EventRecord((hipEvent_t)ev, (hipStream_t)st), "EventRecord");
       break;
     }
     case OP_EVENT_SYNC: {
@@ -383,7 +392,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_EVENT_DESTROY: {
       uint64_t ev;
       memcpy(&ev, body, 8);
-      set_sticky(hip.EventDestroy((hipEvent_t)ev));
+      set_sticky(hip.EventDestroy((hipEvent_t)ev), "EventDestroy");
       break;
     }
     case OP_DEVICE_SYNC: {
@@ -478,7 +487,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&st, body, 8);
       memcpy(&ev, body + 8, 8);
       memcpy(&flags, body + 16, 4);
-      set_sticky(hip.StreamWaitEvent((hipStream_t)st, (hipEvent_t)ev, flags));
+      set_sticky(hip.StreamWaitEvent((hipStream_t)st, (hipEvent_t)ev, flags), "StreamWaitEvent");
       break;
     }
     case OP_SHUTDOWN:

@@ -73,3 +73,56 @@ def test_erl_throttles_real_launches():
                                 "TF_ERL_CAPACITY": "20"})
     assert slow["elapsed_s"] > 4 * fast["elapsed_s"]
     assert slow["elapsed_s"] > 1.5
+
+
+EXPAND_CHILD = r"""
+import ctypes, json, sys
+import torch
+torch.cuda.init()
+# 1 GiB cap + expansion: allocate 3 x 0.75 GiB, all must succeed and stay
+# correct even though 2.25 GiB > cap (over-cap slabs live in host DRAM).
+tensors = []
+for i in range(3):
+    t = torch.full((int(0.75 * (1 << 30)) // 4,), float(i + 1),
+                   device="cuda", dtype=torch.float32)
+    tensors.append(t)
+torch.cuda.synchronize()
+ok = all(float(t.sum()) == (i + 1) * t.numel() for i, t in enumerate(tensors))
+lim = ctypes.CDLL(None)  # limiter is LD_PRELOADed into this process
+expanded = ctypes.c_ulonglong()
+promoted = ctypes.c_ulonglong()
+nranges = ctypes.c_uint()
+enabled = lim.tf_limiter_tier_stats(ctypes.byref(expanded),
+                                    ctypes.byref(promoted),
+                                    ctypes.byref(nranges))
+# demote everything to host, verify data still intact
+lim.tf_limiter_demote_all.restype = ctypes.c_ulonglong
+demoted = lim.tf_limiter_demote_all()
+torch.cuda.synchronize()
+ok2 = all(float(t.sum()) == (i + 1) * t.numel() for i, t in enumerate(tensors))
+print(json.dumps({"ok": bool(ok), "ok_after_demote": bool(ok2),
+                  "enabled": int(enabled),
+                  "expanded": int(expanded.value),
+                  "n_ranges": int(nranges.value),
+                  "demoted": int(demoted)}))
+"""
+
+
+def test_vram_expansion_over_cap():
+    """BASELINE config 4: allocations past the cap land in the host tier
+    and the workload keeps running (no OOM), with migration preserving
+    data."""
+
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = LIMITER
+    env.pop("TF_SHM_PATH", None)
+    env.update({"TF_VRAM_LIMIT_BYTES": str(1 << 30), "TF_VRAM_EXPAND": "1"})
+    out = subprocess.run([sys.executable, "-c", EXPAND_CHILD], env=env,
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["enabled"] == 1
+    assert r["ok"] is True
+    assert r["ok_after_demote"] is True
+    assert r["expanded"] >= int(1.4 * (1 << 30)), r  # ≥2 slabs expanded
+    assert r["n_ranges"] >= 2
