@@ -16,6 +16,7 @@
 // -less) libamdhip64 and fails loudly — deliberately: silent local
 // execution would falsify the remoting claim.
 
+#include <dlfcn.h>
 #include <fcntl.h>
 #include <stdarg.h>
 #include <stdio.h>
@@ -1250,6 +1251,54 @@ hipError_t hipDevicePrimaryCtxRetain(void** ctx, int) {
 }
 
 int tf_client_connected(void) { return C().connected ? 1 : 0; }
+
+// --------------------------------------------------- dlopen redirection
+// PyTorch's driver-API shim dlopens libamdhip64.so and dlsyms entry points
+// directly, which would bypass LD_PRELOAD and hit the GPU-less local
+// runtime. Redirect any dlopen of the HIP runtime to THIS library so those
+// dlsyms resolve to the interposers above.
+
+void* dlopen(const char* filename, int flags) {
+  using fn_t = void* (*)(const char*, int);
+  static fn_t real_dlopen = nullptr;
+  if (!real_dlopen) {
+    // glibc ≥2.34 moved dlopen into libc (version GLIBC_2.34); older lives
+    // in libdl (GLIBC_2.2.5). dlvsym avoids recursing into any interposer.
+    real_dlopen = (fn_t)dlvsym(RTLD_NEXT, "dlopen", "GLIBC_2.34");
+    if (!real_dlopen)
+      real_dlopen = (fn_t)dlvsym(RTLD_NEXT, "dlopen", "GLIBC_2.2.5");
+    if (!real_dlopen) real_dlopen = (fn_t)dlsym(RTLD_NEXT, "dlopen");
+  }
+  if (filename && strstr(filename, "libamdhip64")) {
+    Dl_info info;
+    if (dladdr((void*)&tf_client_connected, &info) && info.dli_fname) {
+      void* h = real_dlopen(info.dli_fname, flags);
+      if (h) return h;
+    }
+  }
+  return real_dlopen(filename, flags);
+}
+
+// ------------------------------------------- driver-API context symbols
+
+hipError_t hipDevicePrimaryCtxGetState(int dev, unsigned* flags,
+                                       int* active) {
+  if (dev < 0 || dev >= C().device_count) return hipErrorInvalidDevice;
+  if (flags) *flags = 0;
+  if (active) *active = 1;  // the remote worker's context is always live
+  return hipSuccess;
+}
+
+hipError_t hipDevicePrimaryCtxSetFlags(int, unsigned) { return hipSuccess; }
+hipError_t hipDevicePrimaryCtxRelease(int) { return hipSuccess; }
+
+hipError_t hipDeviceGet(int* dev, int ordinal) {
+  if (!dev) return hipErrorInvalidValue;
+  if (ordinal < 0 || ordinal >= C().device_count)
+    return hipErrorInvalidDevice;
+  *dev = ordinal;
+  return hipSuccess;
+}
 
 // ------------------------------------------------- call configuration
 // <<<>>> launches compile to __hipPushCallConfiguration + stub +
