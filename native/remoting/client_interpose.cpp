@@ -17,6 +17,8 @@
 // execution would falsify the remoting claim.
 
 #include <dlfcn.h>
+#include <pthread.h>
+#include <errno.h>
 #include <fcntl.h>
 #include <stdarg.h>
 #include <stdio.h>
@@ -73,6 +75,8 @@ struct Client {
   bool connected = false;
   bool debug = false;
   int sock = -1;
+  int seg_fd = -1;          // memfd of the segment, kept for re-handshake
+  char sock_path[256] = {0};
   Header* hdr = nullptr;
   RingView cmd;  // producer
   RingView cpl;  // consumer
@@ -138,6 +142,35 @@ bool send_fd(int sock, int fd) {
   return sendmsg(sock, &msg, 0) == 1;
 }
 
+void* reconnect_main(void* arg) {
+  Client* c = reinterpret_cast<Client*>(arg);
+  for (;;) {
+    char b;
+    ssize_t n = recv(c->sock, &b, 1, 0);
+    if (n > 0) continue;          // workers never send on this socket
+    if (n < 0 && (errno == EINTR || errno == EAGAIN)) continue;
+    // EOF: worker gone (migration or crash)
+    fprintf(stderr, "[tf-client] worker disconnected; waiting for successor on %s\n",
+            c->sock_path);
+    close(c->sock);
+    for (;;) {
+      int s2 = socket(AF_UNIX, SOCK_STREAM, 0);
+      sockaddr_un addr{};
+      addr.sun_family = AF_UNIX;
+      strncpy(addr.sun_path, c->sock_path, sizeof addr.sun_path - 1);
+      if (connect(s2, (sockaddr*)&addr, sizeof addr) == 0 &&
+          send_fd(s2, c->seg_fd)) {
+        c->sock = s2;
+        fprintf(stderr, "[tf-client] re-attached to worker\n");
+        break;
+      }
+      close(s2);
+      usleep(200000);
+    }
+  }
+  return nullptr;
+}
+
 Client::Client() {
   debug = getenv("TF_CLIENT_DEBUG") != nullptr;
   const char* sock_path = getenv("TF_WORKER_SOCKET");
@@ -179,7 +212,8 @@ Client::Client() {
     close(fd);
     return;
   }
-  close(fd);
+  seg_fd = fd;  // kept open: live migration re-sends it to the new worker
+  strncpy(this->sock_path, sock_path, sizeof this->sock_path - 1);
   cmd = RingView(&hdr->cmd, cmd_buf(hdr), CMD_RING_BYTES);
   cpl = RingView(&hdr->cpl, cpl_buf(hdr), CPL_RING_BYTES);
   arena_base = arena(hdr);
@@ -198,6 +232,12 @@ Client::Client() {
   // static's guarded construction; re-entering C() throws recursive_init_error.
   if (debug)
     fprintf(stderr, "[tf-client %d] connected via %s\n", getpid(), sock_path);
+  // watcher: when the worker exits (live migration), re-handshake with its
+  // successor on the same socket path — the shared segment carries all
+  // in-flight protocol state, so the app never notices beyond a pause.
+  pthread_t t;
+  pthread_create(&t, nullptr, reconnect_main, this);
+  pthread_detach(t);
 }
 
 // ------------------------------------------------------------ transport
@@ -363,10 +403,14 @@ uint64_t ship_image(FatBin* fb) {
   return 0;
 }
 
-KernLayout* fetch_layout(FatBin* fb, const char* name) {
+KernLayout* fetch_layout(FatBin* fb, const char* name,
+                         hipError_t* err_out = nullptr) {
   Client& c = C();
   uint64_t mod = ship_image(fb);
-  if (!mod) return nullptr;
+  if (!mod) {
+    if (err_out) *err_out = hipErrorInvalidValue;
+    return nullptr;
+  }
   struct {
     uint64_t image_id, module;
   } b{fb->image_id, mod};
@@ -376,7 +420,11 @@ KernLayout* fetch_layout(FatBin* fb, const char* name) {
   hipError_t e = send_sync(OP_GET_FUNCTION, &b, sizeof b, out.data(),
                            out.size(), &out_len, name, nlen);
   if (e != hipSuccess || out_len < 20) {
-    fprintf(stderr, "[tf-client] GET_FUNCTION %s failed: %d\n", name, e);
+    // 500 = hipErrorNotFound: NORMAL control flow — Tensile probes its
+    // lazily-loaded code objects for each kernel. Stay quiet.
+    if (e != 500)
+      fprintf(stderr, "[tf-client] GET_FUNCTION %s failed: %d\n", name, e);
+    if (err_out) *err_out = e ? e : hipErrorInvalidValue;
     return nullptr;
   }
   auto* kl = new KernLayout();
@@ -1080,8 +1128,9 @@ hipError_t hipModuleLoadDataEx(void** module, const void* image, unsigned,
 hipError_t hipModuleGetFunction(void** fn, void* module, const char* name) {
   auto* pr = reinterpret_cast<std::pair<FatBin*, uint64_t>*>(module);
   if (!pr) return hipErrorInvalidValue;
-  KernLayout* kl = fetch_layout(pr->first, name);
-  if (!kl) return hipErrorInvalidValue;
+  hipError_t err = hipSuccess;
+  KernLayout* kl = fetch_layout(pr->first, name, &err);
+  if (!kl) return err ? err : hipErrorInvalidValue;
   Client& c = C();
   {
     std::lock_guard<std::mutex> l(c.reg_mu);

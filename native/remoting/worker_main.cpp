@@ -16,6 +16,7 @@
 // and starts in GPU-less CI, failing only on first real command).
 
 #include <dlfcn.h>
+#include <signal.h>
 #include <errno.h>
 #include <fcntl.h>
 #include <stdio.h>
@@ -82,6 +83,21 @@ struct Hip {
   hipError_t (*DeviceCanAccessPeer)(int*, int, int);
   hipError_t (*GetLastError)(void);
 
+  // VMM surface (optional — absent on very old runtimes; worker falls back
+  // to plain hipMalloc and snapshot/restore is disabled).
+  hipError_t (*MemAddressReserve)(void**, size_t, size_t, void*,
+                                  unsigned long long) = nullptr;
+  hipError_t (*MemAddressFree)(void*, size_t) = nullptr;
+  hipError_t (*MemCreate)(void**, size_t, const void*,
+                          unsigned long long) = nullptr;
+  hipError_t (*MemRelease)(void*) = nullptr;
+  hipError_t (*MemMap)(void*, size_t, size_t, void*,
+                       unsigned long long) = nullptr;
+  hipError_t (*MemUnmap)(void*, size_t) = nullptr;
+  hipError_t (*MemSetAccess)(void*, size_t, const void*, size_t) = nullptr;
+  hipError_t (*MemGetAllocationGranularity)(size_t*, const void*,
+                                            int) = nullptr;
+
   bool load() {
     const char* names[] = {"libamdhip64.so", "libamdhip64.so.7",
                            "/opt/rocm/lib/libamdhip64.so"};
@@ -125,15 +141,202 @@ struct Hip {
     R(DeviceCanAccessPeer, "hipDeviceCanAccessPeer")
     R(GetLastError, "hipGetLastError")
 #undef R
+#define O(f, sym) f = reinterpret_cast<decltype(f)>(dlsym(h, sym));
+    O(MemAddressReserve, "hipMemAddressReserve")
+    O(MemAddressFree, "hipMemAddressFree")
+    O(MemCreate, "hipMemCreate")
+    O(MemRelease, "hipMemRelease")
+    O(MemMap, "hipMemMap")
+    O(MemUnmap, "hipMemUnmap")
+    O(MemSetAccess, "hipMemSetAccess")
+    O(MemGetAllocationGranularity, "hipMemGetAllocationGranularity")
+#undef O
     return true;
+  }
+
+  bool vmm_available() const {
+    return MemAddressReserve && MemAddressFree && MemCreate && MemRelease &&
+           MemMap && MemUnmap && MemSetAccess && MemGetAllocationGranularity;
   }
 };
 
 Hip hip;
 
+// ------------------------------------------------- VA-stable VMM heap
+// Device memory is suballocated from ONE VA reservation at a fixed base,
+// mapped page-by-page with hipMemCreate/hipMemMap. Snapshot/restore and
+// live migration rely on this: a resumed worker reserves the same base, so
+// every device pointer the client holds stays valid — no pointer
+// translation (the part the reference left closed-source and returned 501
+// for).
+
+struct hipMemLocation_ {
+  int type;
+  int id;
+};
+struct hipMemAllocationProp_ {
+  int type;                 // hipMemAllocationTypePinned = 1 (device mem)
+  int requestedHandleType;  // none
+  hipMemLocation_ location;
+  void* win32HandleMetaData;
+  struct {
+    unsigned char compressionType, gpuDirectRDMACapable;
+    unsigned short usage;
+  } allocFlags;
+};
+struct hipMemAccessDesc_ {
+  hipMemLocation_ location;
+  int flags;  // hipMemAccessFlagsProtReadWrite = 3
+};
+
+constexpr uint64_t VMM_BASE_HINT = 0x7a0000000000ull;
+
+struct VmmRange {
+  void* handle = nullptr;
+  size_t bytes = 0;  // granularity-rounded mapped size
+  size_t req_bytes = 0;  // what the client asked for (snapshot copies this)
+};
+
+struct Vmm {
+  bool enabled = false;
+  int device = 0;
+  uint64_t base = 0;
+  size_t heap_bytes = 0;
+  size_t gran = 2u << 20;
+  std::map<uint64_t, size_t> free_spans;          // va → len
+  std::map<uint64_t, VmmRange> mapped;            // va → range
+
+  bool init(int dev, uint64_t base_hint, size_t heap) {
+    if (!hip.vmm_available()) return false;
+    device = dev;
+    hipMemAllocationProp_ prop{};
+    prop.type = 1;
+    prop.location = {1 /*device*/, dev};
+    size_t g = 0;
+    if (hip.MemGetAllocationGranularity(&g, &prop, 1 /*recommended*/) != 0 ||
+        g == 0)
+      return false;
+    gran = g;
+    void* p = nullptr;
+    if (hip.MemAddressReserve(&p, heap, 0, (void*)base_hint, 0) != 0)
+      return false;
+    if (base_hint && (uint64_t)p != base_hint) {
+      // restore REQUIRES the exact base; fresh starts accept any
+      if (base_hint != VMM_BASE_HINT) {
+        hip.MemAddressFree(p, heap);
+        return false;
+      }
+    }
+    base = (uint64_t)p;
+    heap_bytes = heap;
+    free_spans[base] = heap;
+    enabled = true;
+    return true;
+  }
+
+  uint64_t round_up(uint64_t n) const { return (n + gran - 1) & ~(gran - 1); }
+
+  hipError_t map_at(uint64_t va, size_t len, size_t req) {
+    hipMemAllocationProp_ prop{};
+    prop.type = 1;
+    prop.location = {1, device};
+    void* handle = nullptr;
+    hipError_t e = hip.MemCreate(&handle, len, &prop, 0);
+    if (e != 0) return e;
+    e = hip.MemMap((void*)va, len, 0, handle, 0);
+    if (e != 0) {
+      hip.MemRelease(handle);
+      return e;
+    }
+    hipMemAccessDesc_ acc{{1, device}, 3 /*RW*/};
+    e = hip.MemSetAccess((void*)va, len, &acc, 1);
+    if (e != 0) {
+      hip.MemUnmap((void*)va, len);
+      hip.MemRelease(handle);
+      return e;
+    }
+    mapped[va] = VmmRange{handle, len, req};
+    return 0;
+  }
+
+  void* alloc(size_t sz, hipError_t* err) {
+    size_t len = round_up(sz ? sz : 1);
+    for (auto it = free_spans.begin(); it != free_spans.end(); ++it) {
+      if (it->second < len) continue;
+      uint64_t va = it->first;
+      size_t span = it->second;
+      free_spans.erase(it);
+      if (span > len) free_spans[va + len] = span - len;
+      hipError_t e = map_at(va, len, sz);
+      if (e != 0) {
+        free_spans[va] = len;  // give back (coalesce lazily)
+        *err = e;
+        return nullptr;
+      }
+      *err = 0;
+      return (void*)va;
+    }
+    *err = 2;  // hipErrorOutOfMemory (VA heap exhausted)
+    return nullptr;
+  }
+
+  // Restore path: map at an exact VA recorded in the snapshot.
+  hipError_t alloc_exact(uint64_t va, size_t req) {
+    size_t len = round_up(req ? req : 1);
+    // carve [va, va+len) out of the containing free span
+    auto it = free_spans.upper_bound(va);
+    if (it == free_spans.begin()) return 1;
+    --it;
+    uint64_t sb = it->first;
+    size_t slen = it->second;
+    if (va < sb || va + len > sb + slen) return 1;
+    free_spans.erase(it);
+    if (va > sb) free_spans[sb] = va - sb;
+    if (sb + slen > va + len) free_spans[va + len] = sb + slen - (va + len);
+    return map_at(va, len, req);
+  }
+
+  hipError_t free_(void* p) {
+    auto it = mapped.find((uint64_t)p);
+    if (it == mapped.end()) return 1;
+    hip.MemUnmap(p, it->second.bytes);
+    hip.MemRelease(it->second.handle);
+    uint64_t va = it->first;
+    size_t len = it->second.bytes;
+    mapped.erase(it);
+    // coalesce with neighbours
+    auto nxt = free_spans.find(va + len);
+    if (nxt != free_spans.end()) {
+      len += nxt->second;
+      free_spans.erase(nxt);
+    }
+    if (!free_spans.empty()) {
+      auto prv = free_spans.upper_bound(va);
+      if (prv != free_spans.begin()) {
+        --prv;
+        if (prv->first + prv->second == va) {
+          prv->second += len;
+          return 0;
+        }
+      }
+    }
+    free_spans[va] = len;
+    return 0;
+  }
+};
+
 struct PendingChunk {
   hipEvent_t ev;
   uint64_t end_off;  // arena_freed advances to this when ev completes
+};
+
+struct FuncRec {
+  uint64_t image_id;
+  std::string name;
+};
+struct StreamRec {
+  uint32_t flags;
+  int prio;
 };
 
 struct Worker {
@@ -145,9 +348,29 @@ struct Worker {
   std::unordered_map<uint64_t, std::map<std::string, tfrpc::KernelSig>> sigs;
   std::unordered_map<uint64_t, hipModule_t> modules;  // image_id → module
   bool verbose = getenv("TF_WORKER_DEBUG") != nullptr;
+
+  // ---- migratable state (snapshot/restore) ----
+  Vmm vmm;
+  int device = 0;
+  std::map<uint64_t, size_t> plain_allocs;  // fallback when !vmm.enabled
+  std::unordered_map<uint64_t, std::vector<uint8_t>> images;  // id → bytes
+  std::unordered_map<uint64_t, uint64_t> module_image;  // handle → image_id
+  std::unordered_map<uint64_t, FuncRec> funcs;    // func handle → origin
+  std::unordered_map<uint64_t, StreamRec> streams;
+  std::unordered_map<uint64_t, uint32_t> events;  // handle → flags
+  std::unordered_map<uint64_t, uint64_t> tr;  // old handle → new (restore)
 };
 
 Worker W;
+
+volatile sig_atomic_t g_snapshot_req = 0;
+
+// Translate a handle from a pre-migration snapshot to its live value.
+inline uint64_t xl(uint64_t h) {
+  if (W.tr.empty() || h == 0) return h;
+  auto it = W.tr.find(h);
+  return it == W.tr.end() ? h : it->second;
+}
 
 void set_sticky(hipError_t e, const char* what = "") {
   if (e != 0) {
@@ -239,7 +462,13 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       uint64_t size;
       memcpy(&size, body, 8);
       void* p = nullptr;
-      hipError_t e = hip.Malloc(&p, size);
+      hipError_t e;
+      if (W.vmm.enabled) {
+        p = W.vmm.alloc(size, &e);
+      } else {
+        e = hip.Malloc(&p, size);
+        if (e == 0) W.plain_allocs[(uint64_t)p] = size;
+      }
       uint64_t r = (uint64_t)p;
       reply(c->seq, e, &r, 8);
       break;
@@ -247,13 +476,20 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_FREE: {
       uint64_t p;
       memcpy(&p, body, 8);
-      hipError_t e = hip.Free((void*)p);
+      hipError_t e;
+      if (W.vmm.enabled) {
+        e = W.vmm.free_((void*)p);
+      } else {
+        e = hip.Free((void*)p);
+        W.plain_allocs.erase(p);
+      }
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
-      else set_sticky(e, "async-op");
+      else set_sticky(e, "hipFree");
       break;
     }
     case OP_MEMCPY_H2D: {
       auto* m = reinterpret_cast<MemcpyBody*>(body);
+      m->stream = xl(m->stream);
       const void* src;
       if (c->flags & F_INLINE_DATA)
         src = body + sizeof(MemcpyBody);
@@ -277,6 +513,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     }
     case OP_MEMCPY_D2H: {
       auto* m = reinterpret_cast<MemcpyBody*>(body);
+      m->stream = xl(m->stream);
       void* dst = W.arena + (m->arena_off % ARENA_BYTES);
       hipError_t e = hip.MemcpyAsync(dst, (const void*)m->src, m->size,
                                      2 /*D2H*/, (hipStream_t)m->stream);
@@ -286,6 +523,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     }
     case OP_MEMCPY_D2D: {
       auto* m = reinterpret_cast<MemcpyBody*>(body);
+      m->stream = xl(m->stream);
       hipError_t e = hip.MemcpyAsync((void*)m->dst, (const void*)m->src,
                                      m->size, 3, (hipStream_t)m->stream);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
@@ -294,6 +532,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     }
     case OP_MEMSET: {
       auto* m = reinterpret_cast<MemcpyBody*>(body);
+      m->stream = xl(m->stream);
       hipError_t e = hip.MemsetD8Async((void*)m->dst, (unsigned char)m->kind,
                                        m->size, (hipStream_t)m->stream);
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
@@ -302,6 +541,8 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     }
     case OP_LAUNCH: {
       auto* l = reinterpret_cast<LaunchBody*>(body);
+      l->func = xl(l->func);
+      l->stream = xl(l->stream);
       void* kernarg = body + sizeof(LaunchBody);
       size_t sz = l->kernarg_size;
       void* extra[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, kernarg,
@@ -329,6 +570,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&prio, body + 4, 4);
       hipStream_t st = nullptr;
       hipError_t e = hip.StreamCreateWithPriority(&st, flags, prio);
+      if (e == 0) W.streams[(uint64_t)st] = StreamRec{flags, prio};
       uint64_t r = (uint64_t)st;
       reply(c->seq, e, &r, 8);
       break;
@@ -336,12 +578,15 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_STREAM_DESTROY: {
       uint64_t st;
       memcpy(&st, body, 8);
+      st = xl(st);
+      W.streams.erase(st);
       set_sticky(hip.StreamDestroy((hipStream_t)st), "StreamDestroy");
       break;
     }
     case OP_STREAM_SYNC: {
       uint64_t st;
       memcpy(&st, body, 8);
+      st = xl(st);
       hipError_t e = hip.StreamSynchronize((hipStream_t)st);
       reply(c->seq, e, nullptr, 0);
       break;
@@ -349,6 +594,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_STREAM_QUERY: {
       uint64_t st;
       memcpy(&st, body, 8);
+      st = xl(st);
       reply(c->seq, hip.StreamQuery((hipStream_t)st), nullptr, 0);
       break;
     }
@@ -357,6 +603,7 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&flags, body, 4);
       hipEvent_t ev = nullptr;
       hipError_t e = hip.EventCreateWithFlags(&ev, flags);
+      if (e == 0) W.events[(uint64_t)ev] = flags;
       uint64_t r = (uint64_t)ev;
       reply(c->seq, e, &r, 8);
       break;
@@ -365,18 +612,22 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       uint64_t ev, st;
       memcpy(&ev, body, 8);
       memcpy(&st, body + 8, 8);
+      ev = xl(ev);
+      st = xl(st);
       set_sticky(hip.EventRecord((hipEvent_t)ev, (hipStream_t)st), "EventRecord");
       break;
     }
     case OP_EVENT_SYNC: {
       uint64_t ev;
       memcpy(&ev, body, 8);
+      ev = xl(ev);
       reply(c->seq, hip.EventSynchronize((hipEvent_t)ev), nullptr, 0);
       break;
     }
     case OP_EVENT_QUERY: {
       uint64_t ev;
       memcpy(&ev, body, 8);
+      ev = xl(ev);
       reply(c->seq, hip.EventQuery((hipEvent_t)ev), nullptr, 0);
       break;
     }
@@ -384,6 +635,8 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       uint64_t e0, e1;
       memcpy(&e0, body, 8);
       memcpy(&e1, body + 8, 8);
+      e0 = xl(e0);
+      e1 = xl(e1);
       float ms = 0;
       hipError_t e = hip.EventElapsedTime(&ms, (hipEvent_t)e0, (hipEvent_t)e1);
       reply(c->seq, e, &ms, 4);
@@ -392,6 +645,8 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_EVENT_DESTROY: {
       uint64_t ev;
       memcpy(&ev, body, 8);
+      ev = xl(ev);
+      W.events.erase(ev);
       set_sticky(hip.EventDestroy((hipEvent_t)ev), "EventDestroy");
       break;
     }
@@ -416,6 +671,10 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
         e = hip.ModuleLoadData(&mod, img);
         if (e == 0) {
           W.modules[b.image_id] = mod;
+          W.module_image[(uint64_t)mod] = b.image_id;
+          // snapshot needs the bytes after the arena chunk is recycled
+          W.images[b.image_id].assign((const uint8_t*)img,
+                                      (const uint8_t*)img + b.size);
           std::string err;
           auto& m = W.sigs[b.image_id];
           if (!tfrpc::parse_kernel_signatures(img, b.size, &m, &err) &&
@@ -435,9 +694,11 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       uint64_t image_id, mod;
       memcpy(&image_id, body, 8);
       memcpy(&mod, body + 8, 8);
+      mod = xl(mod);
       const char* name = (const char*)(body + 16);
       hipFunction_t fn = nullptr;
       hipError_t e = hip.ModuleGetFunction(&fn, (hipModule_t)mod, name);
+      if (e == 0) W.funcs[(uint64_t)fn] = FuncRec{image_id, name};
       // reply: {func u64, kernarg_size u32, explicit u32, nargs u32,
       //         {size,offset} x nargs}
       std::vector<uint8_t> out(20);
@@ -487,6 +748,8 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&st, body, 8);
       memcpy(&ev, body + 8, 8);
       memcpy(&flags, body + 16, 4);
+      st = xl(st);
+      ev = xl(ev);
       set_sticky(hip.StreamWaitEvent((hipStream_t)st, (hipEvent_t)ev, flags), "StreamWaitEvent");
       break;
     }
@@ -499,6 +762,225 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       break;
   }
 }
+
+
+// --------------------------------------------------- snapshot / restore
+// Device-level snapshot: quiesce, dump every allocation (VA + contents),
+// module images, function/stream/event registries. Restore rebuilds the
+// identical VA layout (VMM heap at the same base) and loads a handle
+// translation table for pre-snapshot stream/event/module/function values.
+// Triggered by SIGUSR1 with TF_WORKER_SNAPSHOT_PATH set; a worker started
+// with TF_WORKER_RESTORE_PATH resumes serving the SAME client segment —
+// live migration of a vGPU under a live client.
+
+constexpr uint64_t SNAP_MAGIC = 0x50414e53465431ull;  // "TFSNAP1"
+
+bool write_all(FILE* f, const void* p, size_t n) {
+  return fwrite(p, 1, n, f) == n;
+}
+bool read_all(FILE* f, void* p, size_t n) {
+  return fread(p, 1, n, f) == n;
+}
+
+int do_snapshot(const char* path) {
+  if (!W.vmm.enabled) {
+    fprintf(stderr, "[worker] snapshot requires the VMM heap (TF_WORKER_NO_VMM unset)\n");
+    return -1;
+  }
+  hipError_t e = hip.DeviceSynchronize();
+  if (e != 0) {
+    fprintf(stderr, "[worker] snapshot: device sync failed %d\n", e);
+    return -1;
+  }
+  FILE* f = fopen(path, "wb");
+  if (!f) return -1;
+  uint64_t magic = SNAP_MAGIC;
+  write_all(f, &magic, 8);
+  write_all(f, &W.vmm.base, 8);
+  uint64_t heap = W.vmm.heap_bytes;
+  write_all(f, &heap, 8);
+  int32_t dev = W.device;
+  write_all(f, &dev, 4);
+
+  uint32_t n_allocs = (uint32_t)W.vmm.mapped.size();
+  write_all(f, &n_allocs, 4);
+  std::vector<uint8_t> staging;
+  for (auto& [va, r] : W.vmm.mapped) {
+    uint64_t v = va, req = r.req_bytes;
+    write_all(f, &v, 8);
+    write_all(f, &req, 8);
+    staging.resize(r.req_bytes);
+    e = hip.MemcpyAsync(staging.data(), (void*)va, r.req_bytes, 2 /*D2H*/,
+                        nullptr);
+    if (e == 0) e = hip.StreamSynchronize(nullptr);
+    if (e != 0) {
+      fclose(f);
+      fprintf(stderr, "[worker] snapshot: D2H %llx failed %d\n",
+              (unsigned long long)va, e);
+      return -1;
+    }
+    write_all(f, staging.data(), r.req_bytes);
+  }
+
+  uint32_t n_mod = (uint32_t)W.module_image.size();
+  write_all(f, &n_mod, 4);
+  for (auto& [handle, image_id] : W.module_image) {
+    auto& img = W.images[image_id];
+    uint64_t h = handle, id = image_id, sz = img.size();
+    write_all(f, &h, 8);
+    write_all(f, &id, 8);
+    write_all(f, &sz, 8);
+    write_all(f, img.data(), img.size());
+  }
+
+  uint32_t n_fn = (uint32_t)W.funcs.size();
+  write_all(f, &n_fn, 4);
+  for (auto& [handle, rec] : W.funcs) {
+    uint64_t h = handle, id = rec.image_id;
+    uint32_t nl = (uint32_t)rec.name.size();
+    write_all(f, &h, 8);
+    write_all(f, &id, 8);
+    write_all(f, &nl, 4);
+    write_all(f, rec.name.data(), nl);
+  }
+
+  uint32_t n_st = (uint32_t)W.streams.size();
+  write_all(f, &n_st, 4);
+  for (auto& [handle, rec] : W.streams) {
+    uint64_t h = handle;
+    int32_t prio = rec.prio;
+    write_all(f, &h, 8);
+    write_all(f, &rec.flags, 4);
+    write_all(f, &prio, 4);
+  }
+
+  uint32_t n_ev = (uint32_t)W.events.size();
+  write_all(f, &n_ev, 4);
+  for (auto& [handle, flags] : W.events) {
+    uint64_t h = handle;
+    write_all(f, &h, 8);
+    write_all(f, &flags, 4);
+  }
+  bool ok = fflush(f) == 0;
+  fclose(f);
+  fprintf(stderr, "[worker] snapshot -> %s (%u allocs, %u modules, %u fns, "
+          "%u streams, %u events)\n", path, n_allocs, n_mod, n_fn, n_st,
+          n_ev);
+  return ok ? 0 : -1;
+}
+
+int do_restore(const char* path) {
+  FILE* f = fopen(path, "rb");
+  if (!f) {
+    fprintf(stderr, "[worker] restore: cannot open %s\n", path);
+    return -1;
+  }
+  uint64_t magic = 0, base = 0, heap = 0;
+  int32_t dev = 0;
+  if (!read_all(f, &magic, 8) || magic != SNAP_MAGIC ||
+      !read_all(f, &base, 8) || !read_all(f, &heap, 8) ||
+      !read_all(f, &dev, 4)) {
+    fclose(f);
+    return -1;
+  }
+  if (!W.vmm.init(W.device, base, heap)) {
+    fprintf(stderr, "[worker] restore: cannot re-reserve VA base %llx\n",
+            (unsigned long long)base);
+    fclose(f);
+    return -1;
+  }
+  uint32_t n_allocs = 0;
+  read_all(f, &n_allocs, 4);
+  std::vector<uint8_t> staging;
+  for (uint32_t i = 0; i < n_allocs; ++i) {
+    uint64_t va = 0, req = 0;
+    if (!read_all(f, &va, 8) || !read_all(f, &req, 8)) goto fail;
+    if (W.vmm.alloc_exact(va, req) != 0) {
+      fprintf(stderr, "[worker] restore: map %llx failed\n",
+              (unsigned long long)va);
+      goto fail;
+    }
+    staging.resize(req);
+    if (!read_all(f, staging.data(), req)) goto fail;
+    if (hip.MemcpyAsync((void*)va, staging.data(), req, 1, nullptr) != 0)
+      goto fail;
+  }
+  hip.StreamSynchronize(nullptr);
+
+  {
+    uint32_t n_mod = 0;
+    read_all(f, &n_mod, 4);
+    for (uint32_t i = 0; i < n_mod; ++i) {
+      uint64_t old_h = 0, id = 0, sz = 0;
+      if (!read_all(f, &old_h, 8) || !read_all(f, &id, 8) ||
+          !read_all(f, &sz, 8))
+        goto fail;
+      auto& img = W.images[id];
+      img.resize(sz);
+      if (!read_all(f, img.data(), sz)) goto fail;
+      hipModule_t mod = nullptr;
+      if (hip.ModuleLoadData(&mod, img.data()) != 0) goto fail;
+      W.modules[id] = mod;
+      W.module_image[(uint64_t)mod] = id;
+      W.tr[old_h] = (uint64_t)mod;
+      std::string err;
+      tfrpc::parse_kernel_signatures(img.data(), sz, &W.sigs[id], &err);
+    }
+    uint32_t n_fn = 0;
+    read_all(f, &n_fn, 4);
+    for (uint32_t i = 0; i < n_fn; ++i) {
+      uint64_t old_h = 0, id = 0;
+      uint32_t nl = 0;
+      if (!read_all(f, &old_h, 8) || !read_all(f, &id, 8) ||
+          !read_all(f, &nl, 4))
+        goto fail;
+      std::string name(nl, 0);
+      if (!read_all(f, name.data(), nl)) goto fail;
+      hipFunction_t fn = nullptr;
+      auto mit = W.modules.find(id);
+      if (mit == W.modules.end() ||
+          hip.ModuleGetFunction(&fn, mit->second, name.c_str()) != 0)
+        goto fail;
+      W.tr[old_h] = (uint64_t)fn;
+      W.funcs[(uint64_t)fn] = FuncRec{id, name};
+    }
+    uint32_t n_st = 0;
+    read_all(f, &n_st, 4);
+    for (uint32_t i = 0; i < n_st; ++i) {
+      uint64_t old_h = 0;
+      uint32_t flags = 0;
+      int32_t prio = 0;
+      if (!read_all(f, &old_h, 8) || !read_all(f, &flags, 4) ||
+          !read_all(f, &prio, 4))
+        goto fail;
+      hipStream_t st = nullptr;
+      if (hip.StreamCreateWithPriority(&st, flags, prio) != 0) goto fail;
+      W.tr[old_h] = (uint64_t)st;
+      W.streams[(uint64_t)st] = StreamRec{flags, prio};
+    }
+    uint32_t n_ev = 0;
+    read_all(f, &n_ev, 4);
+    for (uint32_t i = 0; i < n_ev; ++i) {
+      uint64_t old_h = 0;
+      uint32_t flags = 0;
+      if (!read_all(f, &old_h, 8) || !read_all(f, &flags, 4)) goto fail;
+      hipEvent_t ev = nullptr;
+      if (hip.EventCreateWithFlags(&ev, flags) != 0) goto fail;
+      W.tr[old_h] = (uint64_t)ev;
+      W.events[(uint64_t)ev] = flags;
+    }
+  }
+  fclose(f);
+  fprintf(stderr, "[worker] restored %u allocs from %s (VA base %llx)\n",
+          n_allocs, path, (unsigned long long)base);
+  return 0;
+fail:
+  fclose(f);
+  fprintf(stderr, "[worker] restore: truncated/invalid snapshot %s\n", path);
+  return -1;
+}
+
+void on_sigusr1(int) { g_snapshot_req = 1; }
 
 int serve(tfrpc::Header* hdr) {
   W.hdr = hdr;
@@ -516,6 +998,20 @@ int serve(tfrpc::Header* hdr) {
   tfrpc::at(&hdr->worker_ready)->store(1, std::memory_order_release);
   fprintf(stderr, "[worker] serving\n");
   while (!hdr->shutdown) {
+    if (g_snapshot_req) {
+      g_snapshot_req = 0;
+      const char* sp = getenv("TF_WORKER_SNAPSHOT_PATH");
+      if (sp && *sp) {
+        retire_pending(true);
+        if (do_snapshot(sp) == 0) {
+          // migration handoff: mark not-ready and exit; the client's
+          // reconnect thread re-attaches to whoever binds the socket next
+          tfrpc::at(&hdr->worker_ready)->store(0, std::memory_order_release);
+          fprintf(stderr, "[worker] exiting for migration\n");
+          exit(0);
+        }
+      }
+    }
     size_t len;
     uint8_t* p = W.cmd.try_next(&len);
     if (!p) {
@@ -568,6 +1064,20 @@ int main(int argc, char** argv) {
     return 3;
   }
   hip.SetDevice(0);
+  W.device = 0;
+  signal(SIGUSR1, on_sigusr1);
+
+  const char* restore_path = getenv("TF_WORKER_RESTORE_PATH");
+  bool want_vmm = !getenv("TF_WORKER_NO_VMM");
+  if (want_vmm && !restore_path) {
+    // fresh start: VA-stable heap (restore re-reserves from the snapshot)
+    size_t heap = 192ull << 30;  // VA only; physical commits per mapping
+    const char* hb = getenv("TF_WORKER_VMM_BYTES");
+    if (hb) heap = strtoull(hb, nullptr, 10);
+    if (!W.vmm.init(0, VMM_BASE_HINT, heap))
+      fprintf(stderr, "[worker] VMM unavailable — plain hipMalloc, "
+                      "snapshot/migration disabled\n");
+  }
 
   unlink(sock_path);
   int srv = socket(AF_UNIX, SOCK_STREAM, 0);
@@ -599,6 +1109,15 @@ int main(int argc, char** argv) {
       munmap(seg, tfrpc::SEG_BYTES);
       close(cli);
       continue;
+    }
+    if (restore_path && *restore_path) {
+      if (do_restore(restore_path) != 0) {
+        fprintf(stderr, "[worker] restore failed — refusing to serve\n");
+        munmap(seg, tfrpc::SEG_BYTES);
+        close(cli);
+        return 4;
+      }
+      restore_path = nullptr;  // one restore per process
     }
     serve(hdr);
     hip.HostUnregister(tfrpc::arena(hdr));

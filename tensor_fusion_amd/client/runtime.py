@@ -56,7 +56,8 @@ class WorkerHandle:
 
 def start_worker(socket_path: str, device_index: int = 0,
                  env: Optional[Dict[str, str]] = None,
-                 oneshot: bool = False, wait_s: float = 30.0) -> WorkerHandle:
+                 oneshot: bool = False, wait_s: float = 30.0,
+                 snapshot_path: Optional[str] = None) -> WorkerHandle:
     """Spawn tf_vgpu_worker bound to one GPU, wait for its socket."""
 
     exe = os.path.join(_NATIVE, "tf_vgpu_worker")
@@ -65,6 +66,8 @@ def start_worker(socket_path: str, device_index: int = 0,
     e = dict(os.environ)
     e["HIP_VISIBLE_DEVICES"] = str(device_index)
     e["ROCR_VISIBLE_DEVICES"] = str(device_index)
+    if snapshot_path:
+        e["TF_WORKER_SNAPSHOT_PATH"] = snapshot_path
     if oneshot:
         e["TF_WORKER_ONESHOT"] = "1"
     e.update(env or {})
@@ -97,3 +100,36 @@ def client_env(socket_path: str, base: Optional[Dict[str, str]] = None,
     if debug:
         e["TF_CLIENT_DEBUG"] = "1"
     return e
+
+
+def snapshot_and_stop(handle: WorkerHandle, snapshot_path: str,
+                      timeout_s: float = 120.0) -> str:
+    """Quiesce + snapshot the worker's device state (VA-stable), then let
+    the process exit. The worker must have been started with
+    TF_WORKER_SNAPSHOT_PATH pointing at `snapshot_path` (start_worker's
+    snapshot_path parameter does this)."""
+
+    import signal as _signal
+    handle.proc.send_signal(_signal.SIGUSR1)
+    handle.proc.wait(timeout=timeout_s)
+    if not os.path.exists(snapshot_path):
+        raise RuntimeError(
+            f"worker exited rc={handle.proc.returncode} without writing "
+            f"{snapshot_path}")
+    return snapshot_path
+
+
+def migrate_worker(handle: WorkerHandle, snapshot_path: str,
+                   new_device_index: int = 0,
+                   env: Optional[Dict[str, str]] = None) -> WorkerHandle:
+    """Live-migrate a vGPU: snapshot worker A, start worker B on
+    `new_device_index` restoring that snapshot on the SAME socket path.
+    The client process re-attaches automatically (its reconnect thread
+    re-sends the shared segment); device pointers stay valid because the
+    worker's VMM heap is re-reserved at the same VA base."""
+
+    sock = handle.socket_path
+    snapshot_and_stop(handle, snapshot_path)
+    e = dict(env or {})
+    e["TF_WORKER_RESTORE_PATH"] = snapshot_path
+    return start_worker(sock, device_index=new_device_index, env=e)
