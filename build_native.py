@@ -116,6 +116,10 @@ def build_remoting():
         out.append(_cc("tf_vgpu_worker",
                        [worker, os.path.join(cdir, "codeobj.cpp")],
                        extra=["-I", cdir, "-ldl", "-pthread"], shared=False))
+    probe = os.path.join(cdir, "vmm_probe.cpp")
+    if os.path.exists(probe):
+        out.append(_cc("tf_vmm_probe", [probe],
+                       extra=["-I", cdir, "-ldl"], shared=False))
     ring_test = os.path.join(cdir, "ring_test.cpp")
     if os.path.exists(ring_test):
         out.append(_cc("tf_ring_test", [ring_test],

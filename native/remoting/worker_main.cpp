@@ -242,15 +242,22 @@ struct Vmm {
     prop.location = {1, device};
     void* handle = nullptr;
     hipError_t e = hip.MemCreate(&handle, len, &prop, 0);
-    if (e != 0) return e;
+    if (e != 0) {
+      fprintf(stderr, "[worker] vmm: hipMemCreate(%zu) -> %d\n", len, e);
+      return e;
+    }
     e = hip.MemMap((void*)va, len, 0, handle, 0);
     if (e != 0) {
+      fprintf(stderr, "[worker] vmm: hipMemMap(%llx,%zu) -> %d\n",
+              (unsigned long long)va, len, e);
       hip.MemRelease(handle);
       return e;
     }
     hipMemAccessDesc_ acc{{1, device}, 3 /*RW*/};
     e = hip.MemSetAccess((void*)va, len, &acc, 1);
     if (e != 0) {
+      fprintf(stderr, "[worker] vmm: hipMemSetAccess(%llx,%zu) -> %d\n",
+              (unsigned long long)va, len, e);
       hip.MemUnmap((void*)va, len);
       hip.MemRelease(handle);
       return e;

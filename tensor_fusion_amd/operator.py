@@ -1,0 +1,192 @@
+"""Operator entry — wires the whole control plane in one process.
+
+Reference: cmd/main.go:131-297 (manager wiring: provider manager, metrics
+recorder, index+GPU+port allocators, autoscaler, webhook, scheduler, 16
+controllers, client HTTP server, TSDB+alerts). `build_operator()` returns
+the assembled Operator for embedding (tests, single-node runtime);
+`python -m tensor_fusion_amd.operator` runs it standalone against the
+embedded store with optional persistence.
+"""
+from __future__ import annotations
+
+import argparse
+import threading
+import time
+from dataclasses import dataclass, field
+from typing import List, Optional
+
+from . import constants as C
+from .alert import AlertEvaluator
+from .allocator.gpuallocator import GpuAllocator
+from .api.store import Store
+from .autoscaler import Autoscaler
+from .cloudprovider import MockProvider
+from .config import ConfigWatcher, GlobalConfig
+from .controllers import ControllerManager, default_controllers
+from .controllers.defrag import DefragController
+from .gang.manager import GangManager
+from .metrics import MetricsRecorder, PoolMetrics, TSDB
+from .portallocator import IndexAllocator, PortAllocator
+from .quota.quota_store import QuotaStore
+from .scheduler.expander import NodeExpander
+from .scheduler.framework import Scheduler
+from .scheduler.gpuresources import GPUResourcesFit
+from .scheduler.gputopo import GPUNetworkTopologyAware
+from .webhook import PodMutator
+
+
+@dataclass
+class Operator:
+    store: Store
+    allocator: GpuAllocator
+    quota: QuotaStore
+    gang: GangManager
+    scheduler: Scheduler
+    mutator: PodMutator
+    controllers: ControllerManager
+    autoscaler: Autoscaler
+    metrics: MetricsRecorder
+    tsdb: TSDB
+    alerts: AlertEvaluator
+    port_allocator: PortAllocator
+    index_allocator: IndexAllocator
+    expander: NodeExpander
+    defrag: DefragController
+    config: Optional[ConfigWatcher] = None
+    _threads: List[threading.Thread] = field(default_factory=list)
+    _stop: threading.Event = field(default_factory=threading.Event)
+
+    # ------------------------------------------------------ admission
+
+    def admit(self, pod):
+        """Webhook path: mutate + persist the pod (the kube-apiserver's
+        admission hook, embedded)."""
+
+        self.mutator.handle(pod)
+        existing = self.store.try_get("Pod", pod.meta.name,
+                                      pod.meta.namespace)
+        if existing is None:
+            self.store.create(pod)
+        else:
+            pod.meta.resource_version = existing.meta.resource_version
+            self.store.update(pod)
+        return pod
+
+    # -------------------------------------------------------- lifecycle
+
+    def tick(self):
+        """One deterministic pass of every loop (tests + single-node)."""
+
+        self.controllers.reconcile_now()
+        self.scheduler.schedule_pending()
+        self.controllers.reconcile_now()
+        self.allocator.sync_dirty()
+        self.quota.sync_dirty()
+        self.autoscaler.tick()
+        self._record_pool_metrics()
+        self.metrics.flush()
+        self.alerts.evaluate()
+
+    def _record_pool_metrics(self):
+        for pool in self.store.list("GPUPool"):
+            s = pool.status
+            self.metrics.set_pool(PoolMetrics(
+                pool=pool.meta.name, node_count=s.node_count,
+                gpu_count=s.gpu_count, total_tflops=s.total.tflops,
+                total_vram=s.total.vram,
+                allocated_tflops=s.total.tflops - s.available.tflops,
+                allocated_vram=s.total.vram - s.available.vram))
+
+    def start(self, scheduler_interval_s: float = 0.5):
+        self.controllers.start()
+        self.autoscaler.start()
+        self.metrics.start()
+        self.alerts.start()
+        if self.config:
+            self.config.start()
+
+        def sched_loop():
+            while not self._stop.wait(scheduler_interval_s):
+                try:
+                    self.scheduler.schedule_pending()
+                    self.allocator.sync_dirty()
+                    self.quota.sync_dirty()
+                    self.allocator.sweep_stale_assumed(
+                        gang_active=self.gang.active_groups())
+                except Exception:
+                    pass
+        t = threading.Thread(target=sched_loop, daemon=True,
+                             name="scheduler-loop")
+        t.start()
+        self._threads.append(t)
+
+    def stop(self):
+        self._stop.set()
+        self.controllers.stop()
+        self.autoscaler.stop()
+        self.metrics.stop()
+        self.alerts.stop()
+        if self.config:
+            self.config.stop()
+
+
+def build_operator(persist_dir: Optional[str] = None,
+                   metrics_dir: str = "",
+                   config_path: Optional[str] = None,
+                   provider=None) -> Operator:
+    store = Store(persist_dir=persist_dir)
+    quota = QuotaStore(store)
+    allocator = GpuAllocator(store=store, quota=quota)
+    gang = GangManager(store)
+    tsdb = TSDB()
+    metrics = MetricsRecorder(out_dir=metrics_dir, tsdb=tsdb)
+    alerts = AlertEvaluator(tsdb)
+    ports = PortAllocator(store)
+    indexes = IndexAllocator(store)
+    expander = NodeExpander(store)
+    mutator = PodMutator(store, index_allocator=indexes,
+                         port_allocator=ports)
+    fit = GPUResourcesFit(store, allocator, gang=gang,
+                          index_allocator=indexes, expander=expander,
+                          metrics=metrics)
+    topo = GPUNetworkTopologyAware(store)
+    scheduler = Scheduler(store, [fit, topo])
+    provider = provider or MockProvider(store=store)
+    mgr = ControllerManager(store)
+    for ctrl in default_controllers(store, allocator=allocator,
+                                    provider=provider):
+        mgr.register(ctrl)
+    autoscaler = Autoscaler(store, tsdb=tsdb, allocator=allocator)
+    defrag = DefragController(store, allocator)
+    cfg = ConfigWatcher(config_path) if config_path else None
+    return Operator(
+        store=store, allocator=allocator, quota=quota, gang=gang,
+        scheduler=scheduler, mutator=mutator, controllers=mgr,
+        autoscaler=autoscaler, metrics=metrics, tsdb=tsdb, alerts=alerts,
+        port_allocator=ports, index_allocator=indexes, expander=expander,
+        defrag=defrag, config=cfg)
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--persist-dir", default="")
+    ap.add_argument("--metrics-dir", default="")
+    ap.add_argument("--config", default="")
+    ap.add_argument("--http-port", type=int, default=C.OperatorHTTPPort)
+    args = ap.parse_args()
+    op = build_operator(persist_dir=args.persist_dir or None,
+                        metrics_dir=args.metrics_dir,
+                        config_path=args.config or None)
+    op.start()
+
+    import uvicorn
+
+    from .server import create_operator_app
+    app = create_operator_app(op.store, allocator=op.allocator,
+                              port_allocator=op.port_allocator,
+                              index_allocator=op.index_allocator)
+    uvicorn.run(app, host="0.0.0.0", port=args.http_port, log_level="warning")
+
+
+if __name__ == "__main__":
+    main()

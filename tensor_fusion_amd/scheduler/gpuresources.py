@@ -32,12 +32,15 @@ class GPUResourcesFit(Plugin):
 
     def __init__(self, store: Store, allocator: GpuAllocator,
                  gang: Optional[GangManager] = None,
-                 index_allocator=None, port_allocator=None):
+                 index_allocator=None, port_allocator=None,
+                 expander=None, metrics=None):
         self.store = store
         self.allocator = allocator
         self.gang = gang
         self.index_allocator = index_allocator
         self.port_allocator = port_allocator
+        self.expander = expander  # NodeExpander: auto-provision on no-fit
+        self.metrics = metrics
         self._mu = threading.Lock()
 
     # -------------------------------------------------------------- hooks
@@ -74,6 +77,13 @@ class GPUResourcesFit(Plugin):
             return None, Status.unschedulable(str(e))
         state[S_SCORES] = scores
         if not scores:
+            if self.metrics:
+                self.metrics.record_unschedulable()
+            if self.expander is not None:
+                claim = self.expander.handle_unschedulable(req)
+                if claim:
+                    reasons = dict(reasons)
+                    reasons["expander"] = f"node claim {claim} in flight"
             msg = "; ".join(f"{k}: {v}" for k, v in reasons.items()) or \
                 "no eligible GPU"
             return None, Status.unschedulable(msg)
@@ -104,6 +114,8 @@ class GPUResourcesFit(Plugin):
             except NotFound:
                 pass
             self.allocator.dealloc(vk)
+        if self.metrics:
+            self.metrics.record_preempted()
         return Status(Code.Success,
                       [f"evicted {len(victims)} lower-QoS pods on {node}"])
 
@@ -190,7 +202,7 @@ class GPUResourcesFit(Plugin):
             C.AnnoContainerGpus: ",".join(alloc.gpu_names),
         }
         if self.index_allocator is not None:
-            idx = self.index_allocator.assign(node, pod.meta.key)
+            idx = self.index_allocator.occupy(node, pod.meta.key)
             annos[C.AnnoPodIndex] = str(idx)
         if req.isolation_mode == C.IsolationHard:
             gpu0 = self.allocator.gpu(alloc.gpu_names[0])
@@ -213,6 +225,10 @@ class GPUResourcesFit(Plugin):
     def post_bind(self, state: CycleState, pod: Pod, node: str) -> None:
         req = state[S_REQ]
         self.allocator.notify_bound(req.pod_key)
+        if self.expander is not None:
+            self.expander.forget_pod(req.pod_key)
+        if self.metrics:
+            self.metrics.record_scheduled(0.0)
         if self.gang:
             self.gang.mark_scheduled(pod)
 
