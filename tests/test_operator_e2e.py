@@ -76,10 +76,16 @@ class TestE2E:
         pts = op.tsdb.query("tf_pool_metrics", "allocated_tflops")
         assert pts and pts[-1][1] >= 600
 
-        # teardown: deleting the worker pod releases the devices
-        op.store.delete("Pod", w.meta.name, "default")
+        # teardown: scaling the workload to 0 deletes workers and
+        # releases the devices
+        def _scale0(obj):
+            obj.replicas = 0
+        op.store.patch("TensorFusionWorkload", "app-1-wl", "default", _scale0)
         for _ in range(3):
             op.tick()
+        workers = [p for p in op.store.list("Pod", namespace="default")
+                   if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker]
+        assert workers == []
         assert op.allocator.allocation(w.meta.key) is None
 
     def test_gang_workload_schedules_atomically(self):
