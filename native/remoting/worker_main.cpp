@@ -201,6 +201,7 @@ struct Vmm {
   bool enabled = false;
   int device = 0;
   uint64_t base = 0;
+  uint64_t high_water = 0;  // top of the mapped region (union SetAccess)
   size_t heap_bytes = 0;
   size_t gran = 2u << 20;
   std::map<uint64_t, size_t> free_spans;          // va → len
@@ -253,8 +254,16 @@ struct Vmm {
       hip.MemRelease(handle);
       return e;
     }
+    // ROCm 7.2 quirk (characterized by tf_vmm_probe): per-range
+    // SetAccess at offsets deep inside a reservation intermittently
+    // returns InvalidValue, but SetAccess over the union
+    // [base, high_water) always succeeds. Apply the union first; fall
+    // back to the exact range.
     hipMemAccessDesc_ acc{{1, device}, 3 /*RW*/};
-    e = hip.MemSetAccess((void*)va, len, &acc, 1);
+    uint64_t end = va + len;
+    uint64_t hw = end > high_water ? end : high_water;
+    e = hip.MemSetAccess((void*)base, hw - base, &acc, 1);
+    if (e != 0) e = hip.MemSetAccess((void*)va, len, &acc, 1);
     if (e != 0) {
       fprintf(stderr, "[worker] vmm: hipMemSetAccess(%llx,%zu) -> %d\n",
               (unsigned long long)va, len, e);
@@ -262,6 +271,7 @@ struct Vmm {
       hip.MemRelease(handle);
       return e;
     }
+    high_water = hw;
     mapped[va] = VmmRange{handle, len, req};
     return 0;
   }

@@ -167,11 +167,12 @@ class GpuAllocator:
             for node, lst in by_node.items():
                 if len(lst) < req.gpu_count:
                     continue
+                # gpu_scores stay lazy: pick_gpus re-scores only the chosen
+                # node at Reserve (eager per-GPU dicts dominated PreFilter)
                 out[node] = NodeScore(
                     node=node,
                     score=self.strategy.score_node(lst, req),
-                    gpu_scores={g.meta.name: self.strategy.score_gpu(g, req)
-                                for g in lst},
+                    gpu_scores={},
                 )
             return out, reasons
 

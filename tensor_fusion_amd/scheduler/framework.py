@@ -167,6 +167,21 @@ class Scheduler:
         self._bind_fn = bind_fn or self._default_bind
         self._unsched_backoff: Dict[str, float] = {}
         self.results: Dict[str, ScheduleResult] = {}
+        # event-maintained ready-node cache: store.list deepcopies every
+        # object, which dominated the scheduling hot path at 1k nodes
+        self._node_mu = threading.Lock()
+        self._ready_nodes: Dict[str, bool] = {
+            n.meta.name: n.status_phase == "Ready"
+            for n in store.list("Node")}
+        store.on_change("Node", self._on_node_event)
+
+    def _on_node_event(self, event: str, obj):
+        with self._node_mu:
+            if event == "DELETED":
+                self._ready_nodes.pop(obj.meta.name, None)
+            else:
+                self._ready_nodes[obj.meta.name] = (
+                    obj.status_phase == "Ready")
 
     # ------------------------------------------------------------- binding
 
@@ -187,8 +202,8 @@ class Scheduler:
             fn(wp)
 
     def _nodes(self) -> List[str]:
-        return [n.meta.name for n in self.store.list("Node")
-                if n.status_phase == "Ready"]
+        with self._node_mu:
+            return [n for n, ready in self._ready_nodes.items() if ready]
 
     # -------------------------------------------------------------- cycle
 
