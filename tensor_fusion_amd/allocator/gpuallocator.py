@@ -19,6 +19,8 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Set, Tuple
 
+import numpy as np
+
 from .. import constants as C
 from ..api.store import NotFound, Store
 from ..api.types import (GPU, AllocRequest, GPUPartition, PartitionTemplate,
@@ -70,6 +72,8 @@ class GpuAllocator:
         self._allocations: Dict[str, Allocation] = {}  # pod key -> alloc
         self._dirty: Set[str] = set()
         self._filters = default_registry()
+        self._soa: Optional[dict] = None  # numpy mirror (lazy, see _soa_get)
+        self._soa_rows: Dict[str, int] = {}
         if store:
             store.on_change("GPU", self._on_gpu_event)
             for g in store.list("GPU"):
@@ -94,18 +98,21 @@ class GpuAllocator:
                     cur.status.capacity = obj.status.capacity
                     cur.status.topology = obj.status.topology
                     cur.status.isolation_mode = obj.status.isolation_mode
+                    self._soa_sync(obj.meta.name)
 
     def _ingest(self, g: GPU):
         g = g.deepcopy()
         self._gpus[g.meta.name] = g
         self._node_gpus.setdefault(g.status.node, set()).add(g.meta.name)
         self._pool_gpus.setdefault(g.status.pool, set()).add(g.meta.name)
+        self._soa_invalidate()
 
     def _evict(self, name: str):
         g = self._gpus.pop(name, None)
         if g:
             self._node_gpus.get(g.status.node, set()).discard(name)
             self._pool_gpus.get(g.status.pool, set()).discard(name)
+        self._soa_invalidate()
 
     def upsert_gpu_for_testing(self, g: GPU):
         """Direct injection (reference UpsertGPUForTesting :421)."""
@@ -134,6 +141,170 @@ class GpuAllocator:
         with self._mu:
             return self._allocations.get(pod_key)
 
+    # ------------------------------------------------------ numpy mirror
+    # Structure-of-arrays mirror of the device fleet: the per-pod filter +
+    # score pass touches every GPU, and pure-Python loops capped the
+    # scheduler at ~12 pods/s on 8k GPUs. Rebuilt on topology changes,
+    # updated in place on assume/release.
+
+    _ISO_CODES = {C.IsolationShared: 0, C.IsolationSoft: 1,
+                  C.IsolationHard: 2, C.IsolationPartitioned: 3}
+
+    def _soa_invalidate(self):
+        self._soa = None
+
+    def _soa_get(self) -> dict:
+        if self._soa is not None:
+            return self._soa
+        names = list(self._gpus.keys())
+        n = len(names)
+        node_names: List[str] = []
+        node_idx_of: Dict[str, int] = {}
+        soa = {
+            "names": names,
+            "av_t": np.zeros(n), "av_v": np.zeros(n), "av_c": np.zeros(n),
+            "cap_t": np.ones(n), "cap_v": np.ones(n),
+            "ready": np.zeros(n, dtype=bool),
+            "n_apps": np.zeros(n, dtype=np.int32),
+            "n_parts": np.zeros(n, dtype=np.int32),
+            "iso": np.zeros(n, dtype=np.int8),
+            "node_id": np.zeros(n, dtype=np.int32),
+            "node_names": node_names,
+        }
+        self._soa_rows = {}
+        for i, name in enumerate(names):
+            self._soa_rows[name] = i
+            g = self._gpus[name]
+            s = g.status
+            nid = node_idx_of.setdefault(s.node, len(node_names))
+            if nid == len(node_names):
+                node_names.append(s.node)
+            soa["node_id"][i] = nid
+            self._soa_fill_row(soa, i, g)
+        self._soa = soa
+        return soa
+
+    def _soa_fill_row(self, soa: dict, i: int, g: GPU):
+        s = g.status
+        soa["av_t"][i] = s.available.tflops
+        soa["av_v"][i] = s.available.vram
+        soa["av_c"][i] = s.available.compute_percent
+        soa["cap_t"][i] = s.capacity.tflops or 1.0
+        soa["cap_v"][i] = s.capacity.vram or 1.0
+        soa["ready"][i] = (s.phase == "Ready"
+                           and s.used_by == "tensor-fusion")
+        soa["n_apps"][i] = len(s.running_apps)
+        soa["n_parts"][i] = len(s.allocated_partitions)
+        soa["iso"][i] = self._ISO_CODES.get(s.isolation_mode, 1)
+
+    def _soa_sync(self, name: str):
+        if self._soa is None:
+            return
+        i = self._soa_rows.get(name)
+        g = self._gpus.get(name)
+        if i is None or g is None:
+            self._soa = None
+            return
+        self._soa_fill_row(self._soa, i, g)
+
+    def _fast_filter_score(self, req: AllocRequest
+                           ) -> Optional[Tuple[Dict[str, NodeScore],
+                                               Dict[str, str]]]:
+        """Vectorized filter+score for the common request shape. Returns
+        None when the request needs the full chain."""
+
+        if (req.partitioned or req.gpu_model or req.gpu_vendor
+                or req.gpu_indices or req.node_affinity or req.pool
+                or type(self.strategy).__name__ not in (
+                    "NodeCompactGPULowLoad", "CompactFirst", "LowLoadFirst")):
+            return None
+        soa = self._soa_get()
+        if not soa["names"]:
+            return {}, {"resource": "no GPUs"}
+        r = req.request
+        fits = ((soa["av_t"] >= r.tflops - 1e-9)
+                & (soa["av_v"] >= r.vram)
+                & (soa["av_c"] >= r.compute_percent - 1e-9))
+        empty = (soa["n_apps"] == 0) & (soa["n_parts"] == 0)
+        iso_ok = empty | ((soa["iso"] == self._ISO_CODES.get(
+            req.isolation_mode, 1)) & (soa["n_parts"] == 0))
+        ok = soa["ready"] & fits & iso_ok
+        if not ok.any():
+            reasons = {}
+            if not soa["ready"].any():
+                reasons["phase"] = "no GPU in Ready phase"
+            elif not (soa["ready"] & fits).any():
+                reasons["resource"] = (
+                    f"insufficient resources (need {r.tflops:.0f} tflops / "
+                    f"{r.vram >> 30} GiB)")
+            else:
+                reasons["isolation"] = \
+                    f"isolation mode {req.isolation_mode} conflicts"
+            return {}, reasons
+        idx = np.nonzero(ok)[0]
+        nid = soa["node_id"][idx]
+        # per-node eligible count (same-node constraint for gpu_count>1)
+        counts = np.bincount(nid, minlength=len(soa["node_names"]))
+        good_nodes = np.nonzero(counts >= req.gpu_count)[0]
+        if len(good_nodes) == 0:
+            return {}, {"same_node":
+                        f"no node with {req.gpu_count} eligible GPUs"}
+        good_mask = np.isin(nid, good_nodes)
+        idx = idx[good_mask]
+        nid = nid[good_mask]
+        # scores
+        sname = type(self.strategy).__name__
+        tw, vw = self.strategy.tflops_weight, self.strategy.vram_weight
+        af = (tw * (1.0 - (soa["av_t"][idx] - r.tflops) / soa["cap_t"][idx])
+              + vw * (1.0 - (soa["av_v"][idx] - r.vram) / soa["cap_v"][idx]))
+        af = np.clip(af, 0.0, 1.0)
+        if sname == "CompactFirst":
+            gpu_score = 100.0 * af
+        else:
+            gpu_score = 100.0 * (1.0 - af)
+        nn = len(soa["node_names"])
+        if sname == "NodeCompactGPULowLoad":
+            usage = (0.5 * (1.0 - soa["av_t"][idx] / soa["cap_t"][idx])
+                     + 0.5 * (1.0 - soa["av_v"][idx] / soa["cap_v"][idx]))
+            sums = np.bincount(nid, weights=usage, minlength=nn)
+            cnts = np.bincount(nid, minlength=nn)
+            with np.errstate(invalid="ignore"):
+                node_score = np.where(cnts > 0, 100.0 * sums /
+                                      np.maximum(cnts, 1), 0.0)
+        else:
+            # mean of top gpu_count scores per node; k==1 → per-node max
+            if req.gpu_count == 1:
+                node_score = np.full(nn, -1.0)
+                np.maximum.at(node_score, nid, gpu_score)
+            else:
+                order = np.lexsort((-gpu_score, nid))
+                node_score = np.zeros(nn)
+                taken = np.zeros(nn, dtype=np.int32)
+                for j in order:
+                    b = nid[j]
+                    if taken[b] < req.gpu_count:
+                        node_score[b] += gpu_score[j]
+                        taken[b] += 1
+                node_score = np.where(taken > 0,
+                                      node_score / np.maximum(taken, 1), 0.0)
+        out: Dict[str, NodeScore] = {}
+        for b in good_nodes:
+            out[soa["node_names"][b]] = NodeScore(
+                node=soa["node_names"][b], score=float(node_score[b]),
+                gpu_scores={})
+        return out, {}
+
+    def eligible_gpu_names(self, req: AllocRequest, node: str) -> List[str]:
+        """Names of devices on `node` passing the filter chain for `req`
+        (gputopo's combo source when PreFilter left gpu_scores lazy)."""
+
+        with self._mu:
+            names = [n for n in self._node_gpus.get(node, set())
+                     if n in self._gpus]
+            eligible, _ = self._filters.apply(
+                req, [self._gpus[n] for n in names])
+            return [g.meta.name for g in eligible]
+
     # --------------------------------------------- CheckQuotaAndFilter
 
     def check_quota_and_filter(self, req: AllocRequest
@@ -147,6 +318,9 @@ class GpuAllocator:
 
         self.quota.check(req)  # raises QuotaExceeded
         with self._mu:
+            fast = self._fast_filter_score(req)
+            if fast is not None:
+                return fast
             pool_names = self._pool_gpus.get(req.pool) if req.pool else None
             cands = [self._gpus[n] for n in (pool_names if pool_names is not None
                                              else self._gpus.keys())
@@ -232,6 +406,7 @@ class GpuAllocator:
                 if partition is not None and g is gpus[0]:
                     g.status.allocated_partitions.append(partition)
                 self._dirty.add(g.meta.name)
+                self._soa_sync(g.meta.name)
             self.quota.assume(req)
             alloc = Allocation(req=req, gpu_names=list(gpu_names),
                                partition=partition)
@@ -333,6 +508,7 @@ class GpuAllocator:
                     p for p in g.status.allocated_partitions
                     if p.partition_id != alloc.partition.partition_id]
             self._dirty.add(n)
+            self._soa_sync(n)
 
     def dealloc(self, pod_key: str) -> None:
         """Release a committed allocation (pod deleted/failed;

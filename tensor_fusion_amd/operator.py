@@ -149,7 +149,7 @@ def build_operator(persist_dir: Optional[str] = None,
     fit = GPUResourcesFit(store, allocator, gang=gang,
                           index_allocator=indexes, expander=expander,
                           metrics=metrics)
-    topo = GPUNetworkTopologyAware(store)
+    topo = GPUNetworkTopologyAware(allocator)
     scheduler = Scheduler(store, [fit, topo])
     provider = provider or MockProvider(store=store)
     mgr = ControllerManager(store)

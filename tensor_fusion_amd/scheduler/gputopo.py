@@ -44,7 +44,10 @@ class GPUNetworkTopologyAware(Plugin):
         combos: Dict[str, List[str]] = {}
         bonuses: Dict[str, float] = {}
         for node, ns in scores.items():
-            combo, same_numa = self._best_combo(node, list(ns.gpu_scores),
+            names = list(ns.gpu_scores)
+            if not names:  # allocator's fast path leaves gpu_scores lazy
+                names = self.allocator.eligible_gpu_names(req, node)
+            combo, same_numa = self._best_combo(node, names,
                                                 ns.gpu_scores, req.gpu_count)
             if combo:
                 combos[node] = combo

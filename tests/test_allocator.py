@@ -32,7 +32,10 @@ def test_filter_and_score_basic():
     req = make_request(tflops=100, vram=16 << 30)
     scores, reasons = a.check_quota_and_filter(req)
     assert set(scores) == {"node-0", "node-1"}
-    assert len(scores["node-0"].gpu_scores) == 8
+    # gpu_scores are computed lazily at Reserve (pick_gpus); PreFilter
+    # returns node-level scores only
+    picked = a.pick_gpus(req, "node-0")
+    assert len(picked) == req.gpu_count
 
 
 def test_assume_commit_dealloc_cycle():
