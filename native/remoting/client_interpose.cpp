@@ -303,7 +303,7 @@ hipError_t send_sync(uint32_t op, const void* body, size_t blen, void* out,
 #endif
       }
       if (!p) {
-        at(&c.hdr->futex_cpl)->store(0, std::memory_order_release);
+        at(&c.hdr->futex_cpl)->exchange(0, std::memory_order_acq_rel);
         p = c.cpl.try_next(&len);
         if (!p) {
           futex_wait(&c.hdr->futex_cpl, 0, 200);

@@ -73,6 +73,11 @@ class RingView {
   }
 
   void wake_consumer() {
+    // load-before-RMW: during bursts the flag is already 1 and the plain
+    // load keeps the line shared instead of ping-ponging an exchange
+    // between the producer and consumer cores on every record
+    if (at(&r_->futex_nonempty)->load(std::memory_order_relaxed) == 1)
+      return;
     if (at(&r_->futex_nonempty)->exchange(1, std::memory_order_release) == 0)
       futex_wake(&r_->futex_nonempty);
   }
@@ -117,7 +122,10 @@ class RingView {
       __builtin_ia32_pause();
 #endif
     }
-    at(&r_->futex_nonempty)->store(0, std::memory_order_release);
+    // exchange (full barrier), not a plain store: pairs with the
+    // producer's cheap load in wake_consumer — the barrier guarantees the
+    // head re-check below observes any commit whose wake was skipped
+    at(&r_->futex_nonempty)->exchange(0, std::memory_order_acq_rel);
     uint64_t tail = at(&r_->tail)->load(std::memory_order_relaxed);
     uint64_t head = at(&r_->head)->load(std::memory_order_acquire);
     if (head > tail) return;

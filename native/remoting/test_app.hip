@@ -97,6 +97,42 @@ int main() {
   CK(hipMemGetInfo(&free_b, &total_b));
   printf("vram_total_gb=%.0f\n", total_b / 1073741824.0);
 
+  // launch-rate microbench: per-launch overhead of the dispatch path
+  // (native HIP vs the remoting ring) — drives the <4% overhead budget.
+  {
+    const int NL = 20000;
+    hipEvent_t b0, b1;
+    CK(hipEventCreateWithFlags(&b0, 0));
+    CK(hipEventCreateWithFlags(&b1, 0));
+    // warmup
+    for (int r = 0; r < 200; ++r)
+      hipLaunchKernelGGL(saxpy, dim3(1), dim3(64), 0, st, 1.0f, dx, dy, 64);
+    CK(hipStreamSynchronize(st));
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    for (int r = 0; r < NL; ++r)
+      hipLaunchKernelGGL(saxpy, dim3(1), dim3(64), 0, st, 1.0f, dx, dy, 64);
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    double enq_us = ((t1.tv_sec - t0.tv_sec) * 1e9 +
+                     (t1.tv_nsec - t0.tv_nsec)) / 1e3 / NL;
+    CK(hipStreamSynchronize(st));
+    struct timespec t2;
+    clock_gettime(CLOCK_MONOTONIC, &t2);
+    double wall_us = ((t2.tv_sec - t0.tv_sec) * 1e9 +
+                      (t2.tv_nsec - t0.tv_nsec)) / 1e3 / NL;
+    printf("launch_enqueue_us=%.3f launch_wall_us=%.3f\n", enq_us, wall_us);
+    // small-H2D rate (the decode loop's per-token pattern)
+    const int NC = 2000;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    for (int r = 0; r < NC; ++r)
+      CK(hipMemcpyAsync(dy, hx.data(), 8, hipMemcpyHostToDevice, st));
+    CK(hipStreamSynchronize(st));
+    clock_gettime(CLOCK_MONOTONIC, &t1);
+    double h2d_us = ((t1.tv_sec - t0.tv_sec) * 1e9 +
+                     (t1.tv_nsec - t0.tv_nsec)) / 1e3 / NC;
+    printf("small_h2d_us=%.3f\n", h2d_us);
+  }
+
   CK(hipFree(dx));
   CK(hipFree(dy));
   CK(hipFree(dsum));
