@@ -40,7 +40,7 @@ def run_child(mode: str, args, local_rank: int) -> dict:
         env.pop(k, None)
     worker = None
     if mode == "vgpu":
-        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "limiter")
+        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "remote")
         if vgpu_mode == "limiter":
             env["LD_PRELOAD"] = os.path.join(
                 REPO, "tensor_fusion_amd", "_native", "libtfhip_limiter.so")
@@ -117,7 +117,7 @@ def main():
 
     overhead = 100.0 * (1.0 - vgpu_tok / native_tok)
     if rank == 0:
-        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "limiter")
+        vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "remote")
         print(json.dumps({
             "metric": "remote-vGPU overhead % vs native HIP (Llama-3-8B tok/s)",
             "value": round(overhead, 3),
