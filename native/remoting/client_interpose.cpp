@@ -125,6 +125,19 @@ void cdbg(const char* fmt, ...) {
   va_end(ap);
 }
 
+// hipGraph capture state (client-side mirror; worker owns the real capture)
+std::mutex g_capture_mu;
+std::map<void*, unsigned long long> g_capturing;  // stream → capture id
+unsigned long long g_capture_seq = 1;
+
+bool stream_capturing(void* stream, unsigned long long* id) {
+  std::lock_guard<std::mutex> l(g_capture_mu);
+  auto it = g_capturing.find(stream);
+  if (it == g_capturing.end()) return false;
+  if (id) *id = it->second;
+  return true;
+}
+
 bool send_fd(int sock, int fd) {
   char buf[1] = {0};
   iovec iov{buf, 1};
@@ -919,20 +932,21 @@ hipError_t hipStreamWaitEvent(void* s, void* ev, unsigned flags) {
   send_async(OP_STREAM_WAIT_EVENT, 0, &b, sizeof b);
   return hipSuccess;
 }
-hipError_t hipStreamIsCapturing(void*, int* status) {
-  if (status) *status = 0;  // hipStreamCaptureStatusNone
+hipError_t hipStreamIsCapturing(void* stream, int* status) {
+  if (status)
+    *status = stream_capturing(stream, nullptr) ? 1 : 0;
   return hipSuccess;
 }
-hipError_t hipStreamGetCaptureInfo(void*, int* status, unsigned long long* id) {
-  if (status) *status = 0;
+hipError_t hipStreamGetCaptureInfo(void* stream, int* status,
+                                   unsigned long long* id) {
+  if (status) *status = stream_capturing(stream, id) ? 1 : 0;
   if (id) *id = 0;
   return hipSuccess;
 }
-hipError_t hipStreamGetCaptureInfo_v2(void*, int* status,
+hipError_t hipStreamGetCaptureInfo_v2(void* stream, int* status,
                                       unsigned long long* id, void** g,
                                       const void*** deps, size_t* ndeps) {
-  if (status) *status = 0;
-  if (id) *id = 0;
+  if (status) *status = stream_capturing(stream, id) ? 1 : 0;
   if (g) *g = nullptr;
   if (deps) *deps = nullptr;
   if (ndeps) *ndeps = 0;
@@ -1486,16 +1500,74 @@ TF_NOTSUP(hipMemPoolSetAccess, void*, const void*, size_t)
 TF_NOTSUP(hipMemPoolTrimTo, void*, size_t)
 TF_NOTSUP(hipMemcpyPeerAsync, void*, int, const void*, int, size_t, void*)
 TF_NOTSUP(hipStreamWriteValue32, void*, void*, unsigned, unsigned)
-TF_NOTSUP(hipStreamBeginCapture, void*, int)
-TF_NOTSUP(hipStreamEndCapture, void*, void**)
-TF_NOTSUP(hipGraphInstantiate, void**, void*, void*, char*, size_t)
-TF_NOTSUP(hipGraphInstantiateWithFlags, void**, void*, unsigned long long)
-TF_NOTSUP(hipGraphLaunch, void*, void*)
-TF_NOTSUP(hipGraphDestroy, void*)
-TF_NOTSUP(hipGraphExecDestroy, void*)
 TF_NOTSUP(hipGraphGetNodes, void*, void**, size_t*)
 TF_NOTSUP(hipGraphNodeGetDependencies, void*, void**, size_t*)
 TF_NOTSUP(hipGraphDebugDotPrint, void*, const char*, unsigned)
 #undef TF_NOTSUP
+
+// ------------------------------------------------------------ hipGraphs
+// Capture runs worker-side on the real stream (every stream op already
+// flows through the ring in order); the client tracks capture state
+// locally so torch's per-op hipStreamIsCapturing checks stay free.
+
+hipError_t hipStreamBeginCapture(void* stream, int mode) {
+  struct {
+    uint64_t st;
+    uint32_t mode;
+  } b{(uint64_t)stream, (uint32_t)mode};
+  hipError_t e = send_sync(OP_BEGIN_CAPTURE, &b, 12, nullptr, 0);
+  if (e == hipSuccess) {
+    std::lock_guard<std::mutex> l(g_capture_mu);
+    g_capturing[stream] = g_capture_seq++;
+  }
+  return e;
+}
+
+hipError_t hipStreamEndCapture(void* stream, void** graph) {
+  uint64_t st = (uint64_t)stream, g = 0;
+  hipError_t e = send_sync(OP_END_CAPTURE, &st, 8, &g, 8);
+  {
+    std::lock_guard<std::mutex> l(g_capture_mu);
+    g_capturing.erase(stream);
+  }
+  if (graph) *graph = (void*)g;
+  return e;
+}
+
+hipError_t hipGraphInstantiateWithFlags(void** graphExec, void* graph,
+                                        unsigned long long flags) {
+  struct {
+    uint64_t graph, flags;
+  } b{(uint64_t)graph, flags};
+  uint64_t ge = 0;
+  hipError_t e = send_sync(OP_GRAPH_INSTANTIATE, &b, 16, &ge, 8);
+  if (graphExec) *graphExec = (void*)ge;
+  return e;
+}
+
+hipError_t hipGraphInstantiate(void** graphExec, void* graph, void*, char*,
+                               size_t) {
+  return hipGraphInstantiateWithFlags(graphExec, graph, 0);
+}
+
+hipError_t hipGraphLaunch(void* graphExec, void* stream) {
+  struct {
+    uint64_t ge, st;
+  } b{(uint64_t)graphExec, (uint64_t)stream};
+  send_async(OP_GRAPH_LAUNCH, 0, &b, 16);
+  return hipSuccess;
+}
+
+hipError_t hipGraphDestroy(void* graph) {
+  uint64_t g = (uint64_t)graph;
+  send_async(OP_GRAPH_DESTROY, 0, &g, 8);
+  return hipSuccess;
+}
+
+hipError_t hipGraphExecDestroy(void* graphExec) {
+  uint64_t g = (uint64_t)graphExec;
+  send_async(OP_GRAPH_EXEC_DESTROY, 0, &g, 8);
+  return hipSuccess;
+}
 
 }  // extern "C"

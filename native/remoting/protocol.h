@@ -123,6 +123,14 @@ enum Op : uint32_t {
   OP_CAN_ACCESS_PEER,    // {dev, peer} → reply {int}
   OP_STREAM_WAIT_EVENT,  // {stream, event, flags} async
   OP_SHUTDOWN,
+  // hipGraphs: capture happens worker-side on the real stream; the
+  // client only tracks "is capturing" locally for torch's hot checks.
+  OP_BEGIN_CAPTURE,      // {stream, mode} → reply {err}
+  OP_END_CAPTURE,        // {stream} → reply {graph}
+  OP_GRAPH_INSTANTIATE,  // {graph, flags} → reply {graphExec}
+  OP_GRAPH_LAUNCH,       // {graphExec, stream} async
+  OP_GRAPH_DESTROY,      // {graph} async
+  OP_GRAPH_EXEC_DESTROY, // {graphExec} async
 };
 
 constexpr uint32_t F_WANT_REPLY = 1u << 0;

@@ -82,6 +82,13 @@ struct Hip {
   hipError_t (*MemGetInfo)(size_t*, size_t*);
   hipError_t (*DeviceCanAccessPeer)(int*, int, int);
   hipError_t (*GetLastError)(void);
+  hipError_t (*StreamBeginCapture)(hipStream_t, int) = nullptr;
+  hipError_t (*StreamEndCapture)(hipStream_t, void**) = nullptr;
+  hipError_t (*GraphInstantiateWithFlags)(void**, void*,
+                                          unsigned long long) = nullptr;
+  hipError_t (*GraphLaunch)(void*, hipStream_t) = nullptr;
+  hipError_t (*GraphDestroy)(void*) = nullptr;
+  hipError_t (*GraphExecDestroy)(void*) = nullptr;
 
   // VMM surface (optional — absent on very old runtimes; worker falls back
   // to plain hipMalloc and snapshot/restore is disabled).
@@ -142,6 +149,12 @@ struct Hip {
     R(GetLastError, "hipGetLastError")
 #undef R
 #define O(f, sym) f = reinterpret_cast<decltype(f)>(dlsym(h, sym));
+    O(StreamBeginCapture, "hipStreamBeginCapture")
+    O(StreamEndCapture, "hipStreamEndCapture")
+    O(GraphInstantiateWithFlags, "hipGraphInstantiateWithFlags")
+    O(GraphLaunch, "hipGraphLaunch")
+    O(GraphDestroy, "hipGraphDestroy")
+    O(GraphExecDestroy, "hipGraphExecDestroy")
     O(MemAddressReserve, "hipMemAddressReserve")
     O(MemAddressFree, "hipMemAddressFree")
     O(MemCreate, "hipMemCreate")
@@ -839,6 +852,68 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       st = xl(st);
       ev = xl(ev);
       set_sticky(hip.StreamWaitEvent((hipStream_t)st, (hipEvent_t)ev, flags), "StreamWaitEvent");
+      break;
+    }
+    case OP_BEGIN_CAPTURE: {
+      uint64_t st;
+      uint32_t mode;
+      memcpy(&st, body, 8);
+      memcpy(&mode, body + 8, 4);
+      st = xl(st);
+      hipError_t e = hip.StreamBeginCapture
+                         ? hip.StreamBeginCapture((hipStream_t)st, (int)mode)
+                         : 801;
+      reply(c->seq, e, nullptr, 0);
+      break;
+    }
+    case OP_END_CAPTURE: {
+      uint64_t st;
+      memcpy(&st, body, 8);
+      st = xl(st);
+      void* graph = nullptr;
+      hipError_t e = hip.StreamEndCapture
+                         ? hip.StreamEndCapture((hipStream_t)st, &graph)
+                         : 801;
+      uint64_t r = (uint64_t)graph;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_GRAPH_INSTANTIATE: {
+      uint64_t graph;
+      uint64_t flags;
+      memcpy(&graph, body, 8);
+      memcpy(&flags, body + 8, 8);
+      void* ge = nullptr;
+      hipError_t e = hip.GraphInstantiateWithFlags
+                         ? hip.GraphInstantiateWithFlags(&ge, (void*)graph,
+                                                         flags)
+                         : 801;
+      uint64_t r = (uint64_t)ge;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_GRAPH_LAUNCH: {
+      uint64_t ge, st;
+      memcpy(&ge, body, 8);
+      memcpy(&st, body + 8, 8);
+      st = xl(st);
+      set_sticky(hip.GraphLaunch ? hip.GraphLaunch((void*)ge, (hipStream_t)st)
+                                 : 801,
+                 "GraphLaunch");
+      break;
+    }
+    case OP_GRAPH_DESTROY: {
+      uint64_t g;
+      memcpy(&g, body, 8);
+      if (hip.GraphDestroy) set_sticky(hip.GraphDestroy((void*)g),
+                                       "GraphDestroy");
+      break;
+    }
+    case OP_GRAPH_EXEC_DESTROY: {
+      uint64_t g;
+      memcpy(&g, body, 8);
+      if (hip.GraphExecDestroy) set_sticky(hip.GraphExecDestroy((void*)g),
+                                           "GraphExecDestroy");
       break;
     }
     case OP_SHUTDOWN:

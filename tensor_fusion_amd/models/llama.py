@@ -15,6 +15,7 @@ from __future__ import annotations
 import argparse
 import json
 import math
+import os
 import time
 from dataclasses import dataclass
 
@@ -91,7 +92,8 @@ class Attention(nn.Module):
         self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(cfg.heads * self.head_dim, cfg.dim, bias=False)
 
-    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None,
+                mask=None):
         B, T, _ = x.shape
         cfg = self.cfg
         q = self.wq(x).view(B, T, cfg.heads, self.head_dim).transpose(1, 2)
@@ -103,17 +105,25 @@ class Attention(nn.Module):
             k_cache, v_cache = cache
             k_cache[:, :, pos] = k
             v_cache[:, :, pos] = v
-            # pos_end is a host int: a device read here (`pos[-1].item()`)
-            # would force a D2H sync per layer per token.
-            end = pos_end if pos_end is not None else int(pos[-1].item()) + 1
-            k = k_cache[:, :, :end]
-            v = v_cache[:, :, :end]
+            if mask is not None:
+                # graph mode: fixed-shape full-cache attention; validity is
+                # the additive mask (updated outside the captured region)
+                k = k_cache
+                v = v_cache
+            else:
+                # pos_end is a host int: a device read (`pos[-1].item()`)
+                # would force a D2H sync per layer per token.
+                end = pos_end if pos_end is not None \
+                    else int(pos[-1].item()) + 1
+                k = k_cache[:, :, :end]
+                v = v_cache[:, :, :end]
         rep = cfg.heads // cfg.kv_heads
         if rep > 1:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
-        causal = T > 1
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=causal)
+        causal = T > 1 and mask is None
+        o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
+                                           is_causal=causal)
         o = o.transpose(1, 2).reshape(B, T, -1)
         return self.wo(o)
 
@@ -137,8 +147,10 @@ class Block(nn.Module):
         self.ln1 = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ln2 = RMSNorm(cfg.dim, cfg.norm_eps)
 
-    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
-        x = x + self.attn(self.ln1(x), cos, sin, pos, cache, pos_end)
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None,
+                mask=None):
+        x = x + self.attn(self.ln1(x), cos, sin, pos, cache, pos_end,
+                          mask=mask)
         x = x + self.mlp(self.ln2(x))
         return x
 
@@ -152,7 +164,8 @@ class Llama(nn.Module):
         self.norm = RMSNorm(cfg.dim, cfg.norm_eps)
         self.lm_head = nn.Linear(cfg.dim, cfg.vocab, bias=False)
 
-    def forward(self, tokens, pos=None, caches=None, pos_end=None):
+    def forward(self, tokens, pos=None, caches=None, pos_end=None,
+                mask=None):
         device = tokens.device
         if pos is None:
             pos = torch.arange(tokens.shape[1], device=device)
@@ -161,7 +174,8 @@ class Llama(nn.Module):
         x = self.embed(tokens)
         for i, blk in enumerate(self.blocks):
             x = blk(x, self._cos, self._sin, pos,
-                    caches[i] if caches is not None else None, pos_end)
+                    caches[i] if caches is not None else None, pos_end,
+                    mask=mask)
         return self.lm_head(self.norm(x))
 
     def make_kv_cache(self, batch: int, max_seq: int, device, dtype):
@@ -223,6 +237,66 @@ def decode_bench(model: Llama, batch: int, ctx: int, steps: int, warmup: int,
     return batch * steps / dt, dt / steps * 1000.0
 
 
+@torch.no_grad()
+def decode_bench_graphs(model: Llama, batch: int, ctx: int, steps: int,
+                        warmup: int, device="cuda", dtype=torch.bfloat16):
+    """hipGraph-captured decode: one graph replay + two 1-element H2Ds per
+    token instead of ~1,100 eager launches. Works natively and through the
+    GPU-over-IP client (capture ops are forwarded to the worker)."""
+
+    cfg = model.cfg
+    total = ctx + steps + warmup + 8
+    caches = model.make_kv_cache(batch, total, device, dtype)
+    toks = torch.randint(0, cfg.vocab, (batch, ctx), device=device)
+    model(toks, pos=torch.arange(ctx, device=device), caches=caches,
+          pos_end=ctx)
+
+    cur = torch.randint(0, cfg.vocab, (batch, 1), device=device)
+    pos_buf = torch.empty(1, dtype=torch.long, device=device)
+    # additive validity mask over the FULL cache length
+    mask = torch.full((1, 1, 1, total), float("-inf"), device=device,
+                      dtype=dtype)
+    mask[..., :ctx] = 0.0
+    zero = torch.zeros(1, device=device, dtype=dtype)
+
+    def step_graphable():
+        logits = model(cur, pos=pos_buf, caches=caches, mask=mask)
+        cur.copy_(logits.argmax(-1))
+
+    # warmup on a side stream (allocator steady-state), then capture
+    side = torch.cuda.Stream()
+    side.wait_stream(torch.cuda.current_stream())
+    with torch.cuda.stream(side):
+        pos_buf.copy_(torch.tensor([ctx]))
+        mask[..., ctx] = 0.0
+        for _ in range(3):
+            step_graphable()
+    torch.cuda.current_stream().wait_stream(side)
+    torch.cuda.synchronize()
+
+    g = torch.cuda.CUDAGraph()
+    with torch.cuda.graph(g):
+        step_graphable()
+
+    host_pos = torch.empty(1, dtype=torch.long, pin_memory=False)
+
+    def step(i):
+        host_pos[0] = ctx + i
+        pos_buf.copy_(host_pos, non_blocking=True)
+        mask[..., ctx + i:ctx + i + 1].copy_(zero.view(1), non_blocking=True)
+        g.replay()
+
+    for i in range(1, warmup):
+        step(i)
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for i in range(warmup, warmup + steps):
+        step(i)
+    torch.cuda.synchronize()
+    dt = time.perf_counter() - t0
+    return batch * steps / dt, dt / steps * 1000.0
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--model", default="llama3-8b")
@@ -232,12 +306,19 @@ def main():
     ap.add_argument("--warmup", type=int, default=8)
     ap.add_argument("--device", default="cuda")
     ap.add_argument("--dtype", default="bf16")
+    ap.add_argument("--graphs", action="store_true",
+                    help="hipGraph-captured decode loop")
     args = ap.parse_args()
     dtype = {"bf16": torch.bfloat16, "fp16": torch.float16,
              "fp32": torch.float32}[args.dtype]
     model = build_model(args.model, device=args.device, dtype=dtype)
-    tok_s, ms = decode_bench(model, args.batch, args.ctx, args.steps,
-                             args.warmup, device=args.device, dtype=dtype)
+    if args.graphs or os.environ.get("TF_BENCH_GRAPHS") == "1":
+        tok_s, ms = decode_bench_graphs(model, args.batch, args.ctx,
+                                        args.steps, args.warmup,
+                                        device=args.device, dtype=dtype)
+    else:
+        tok_s, ms = decode_bench(model, args.batch, args.ctx, args.steps,
+                                 args.warmup, device=args.device, dtype=dtype)
     print(json.dumps({"tok_s": tok_s, "ms_per_step": ms, "model": args.model,
                       "batch": args.batch, "ctx": args.ctx}), flush=True)
 

@@ -126,3 +126,63 @@ def test_vram_expansion_over_cap():
     assert r["ok_after_demote"] is True
     assert r["expanded"] >= int(1.4 * (1 << 30)), r  # ≥2 slabs expanded
     assert r["n_ranges"] >= 2
+
+
+def test_cu_mask_hard_isolation_slows_compute():
+    """Hard isolation on CDNA4 = ROCr CU masking (HSA_CU_MASK): a vGPU
+    restricted to 32 of 256 CUs must run compute-bound work measurably
+    slower (reference SetComputeUnitHardLimit → our accelerator lib
+    composes the mask)."""
+
+    full = run_child("matmul", {}, timeout=300)
+    # 32 CUs = XCD 0 only: mask bits 0..31 (4 words of 8 CUs... ROCr
+    # parses comma-separated 32-bit hex words, LSB = CU 0)
+    masked = run_child("matmul", {
+        "HSA_CU_MASK": "0:0xffffffff",
+        "TF_UP_LIMIT_PERCENT": "100",  # ERL off; only the CU mask acts
+    }, timeout=300)
+    # 1024³ bf16 matmul is compute-heavy: 1/8 of the CUs ⇒ ≥2× slower
+    assert masked["elapsed_s"] > 2.0 * full["elapsed_s"], (full, masked)
+
+
+def test_freeze_blocks_and_resume_continues():
+    """Auto-freeze/trap machinery: the hypervisor's freeze flag must stop
+    a worker's GPU submissions; clearing it resumes them (reference
+    FreezeWorker/ResumeWorker limiter.h:77-81)."""
+
+    import threading
+    CHILD_FREEZE = r"""
+import ctypes, json, sys, time
+import torch
+lim = ctypes.CDLL(None)
+a = torch.randn(512, 512, device="cuda")
+for _ in range(5):
+    a @ a
+torch.cuda.synchronize()
+lim.tf_limiter_freeze(1)
+t0 = time.perf_counter()
+done = {"v": False}
+import threading as th
+def work():
+    (a @ a).sum().item()
+    done["v"] = True
+w = th.Thread(target=work, daemon=True)
+w.start()
+w.join(timeout=1.0)
+frozen_blocked = not done["v"]
+lim.tf_limiter_freeze(0)
+w.join(timeout=30.0)
+resumed = done["v"]
+print(json.dumps({"frozen_blocked": frozen_blocked, "resumed": resumed}))
+"""
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = LIMITER
+    env.pop("TF_SHM_PATH", None)
+    env["TF_UP_LIMIT_PERCENT"] = "99"  # force shm page + bucket path
+    env["TF_ERL_RATE"] = "1000000"
+    out = subprocess.run([sys.executable, "-c", CHILD_FREEZE], env=env,
+                         capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stderr[-2000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["frozen_blocked"] is True
+    assert r["resumed"] is True
