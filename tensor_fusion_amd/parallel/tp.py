@@ -82,7 +82,8 @@ class TPAttention(nn.Module):
                                        tp)
         self.wo = RowParallelLinear(cfg.heads * self.head_dim, cfg.dim, tp)
 
-    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None,
+                mask=None):
         B, T, _ = x.shape
         q = self.wq(x).view(B, T, self.heads, self.head_dim).transpose(1, 2)
         k = self.wk(x).view(B, T, self.kv_heads, self.head_dim).transpose(1, 2)
@@ -93,14 +94,19 @@ class TPAttention(nn.Module):
             k_cache, v_cache = cache
             k_cache[:, :, pos] = k
             v_cache[:, :, pos] = v
-            end = pos_end if pos_end is not None else int(pos[-1].item()) + 1
-            k = k_cache[:, :, :end]
-            v = v_cache[:, :, :end]
+            if mask is not None:
+                k, v = k_cache, v_cache
+            else:
+                end = pos_end if pos_end is not None \
+                    else int(pos[-1].item()) + 1
+                k = k_cache[:, :, :end]
+                v = v_cache[:, :, :end]
         rep = self.heads // self.kv_heads
         if rep > 1:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
-        o = F.scaled_dot_product_attention(q, k, v, is_causal=T > 1)
+        o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
+                                           is_causal=T > 1 and mask is None)
         o = o.transpose(1, 2).reshape(B, T, -1)
         return self.wo(o)
 
@@ -124,8 +130,10 @@ class TPBlock(nn.Module):
         self.ln1 = RMSNorm(cfg.dim, cfg.norm_eps)
         self.ln2 = RMSNorm(cfg.dim, cfg.norm_eps)
 
-    def forward(self, x, cos, sin, pos, cache=None, pos_end=None):
-        x = x + self.attn(self.ln1(x), cos, sin, pos, cache, pos_end)
+    def forward(self, x, cos, sin, pos, cache=None, pos_end=None,
+                mask=None):
+        x = x + self.attn(self.ln1(x), cos, sin, pos, cache, pos_end,
+                          mask=mask)
         x = x + self.mlp(self.ln2(x))
         return x
 
