@@ -186,3 +186,30 @@ print(json.dumps({"frozen_blocked": frozen_blocked, "resumed": resumed}))
     r = json.loads(out.stdout.strip().splitlines()[-1])
     assert r["frozen_blocked"] is True
     assert r["resumed"] is True
+
+
+def test_fractional_vgpu_resnet50_config2():
+    """BASELINE config 2: ResNet-50 bf16 inference under a 25% TFLOPS /
+    8 GB fractional vGPU — must run inside the cap and be throttled
+    relative to the full device."""
+
+    def run(env_extra):
+        env = dict(os.environ)
+        env["LD_PRELOAD"] = LIMITER
+        env.pop("TF_SHM_PATH", None)
+        env.update(env_extra)
+        out = subprocess.run(
+            [sys.executable, "-m", "tensor_fusion_amd.models.resnet",
+             "--batch", "32", "--steps", "16", "--warmup", "4"],
+            env=env, capture_output=True, text=True, timeout=420, cwd=REPO)
+        assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-2500:]
+        return json.loads(out.stdout.strip().splitlines()[-1])
+
+    full = run({"TF_VRAM_LIMIT_BYTES": str(8 << 30),
+                "TF_UP_LIMIT_PERCENT": "100"})
+    frac = run({"TF_VRAM_LIMIT_BYTES": str(8 << 30),
+                "TF_UP_LIMIT_PERCENT": "25",
+                "TF_ERL_RATE": "2000", "TF_ERL_CAPACITY": "200"})
+    assert full["img_s"] > 0 and frac["img_s"] > 0
+    # the 25% vGPU must be meaningfully slower than the full device
+    assert frac["img_s"] < 0.7 * full["img_s"], (full, frac)
