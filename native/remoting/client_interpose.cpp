@@ -1500,7 +1500,6 @@ TF_NOTSUP(hipMemPoolSetAccess, void*, const void*, size_t)
 TF_NOTSUP(hipMemPoolTrimTo, void*, size_t)
 TF_NOTSUP(hipMemcpyPeerAsync, void*, int, const void*, int, size_t, void*)
 TF_NOTSUP(hipStreamWriteValue32, void*, void*, unsigned, unsigned)
-TF_NOTSUP(hipGraphGetNodes, void*, void**, size_t*)
 TF_NOTSUP(hipGraphNodeGetDependencies, void*, void**, size_t*)
 TF_NOTSUP(hipGraphDebugDotPrint, void*, const char*, unsigned)
 #undef TF_NOTSUP
@@ -1561,6 +1560,29 @@ hipError_t hipGraphLaunch(void* graphExec, void* stream) {
 hipError_t hipGraphDestroy(void* graph) {
   uint64_t g = (uint64_t)graph;
   send_async(OP_GRAPH_DESTROY, 0, &g, 8);
+  return hipSuccess;
+}
+
+hipError_t hipGraphGetNodes(void* graph, void** nodes, size_t* count) {
+  if (!count) return hipErrorInvalidValue;
+  struct {
+    uint64_t graph, cap;
+  } b{(uint64_t)graph, nodes ? (uint64_t)*count : 0};
+  std::vector<uint8_t> out(8 + 8 * 4096);
+  size_t out_len = 0;
+  hipError_t e = send_sync(OP_GRAPH_GET_NODES, &b, 16, out.data(),
+                           out.size(), &out_len);
+  if (e != hipSuccess || out_len < 8) return e ? e : hipErrorInvalidValue;
+  uint64_t cnt;
+  memcpy(&cnt, out.data(), 8);
+  if (nodes) {
+    size_t have = (out_len - 8) / 8;
+    size_t n = have < *count ? have : *count;
+    memcpy(nodes, out.data() + 8, n * 8);
+    *count = n;
+  } else {
+    *count = (size_t)cnt;
+  }
   return hipSuccess;
 }
 

@@ -89,6 +89,7 @@ struct Hip {
   hipError_t (*GraphLaunch)(void*, hipStream_t) = nullptr;
   hipError_t (*GraphDestroy)(void*) = nullptr;
   hipError_t (*GraphExecDestroy)(void*) = nullptr;
+  hipError_t (*GraphGetNodes)(void*, void**, size_t*) = nullptr;
 
   // VMM surface (optional — absent on very old runtimes; worker falls back
   // to plain hipMalloc and snapshot/restore is disabled).
@@ -155,6 +156,7 @@ struct Hip {
     O(GraphLaunch, "hipGraphLaunch")
     O(GraphDestroy, "hipGraphDestroy")
     O(GraphExecDestroy, "hipGraphExecDestroy")
+    O(GraphGetNodes, "hipGraphGetNodes")
     O(MemAddressReserve, "hipMemAddressReserve")
     O(MemAddressFree, "hipMemAddressFree")
     O(MemCreate, "hipMemCreate")
@@ -914,6 +916,28 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       memcpy(&g, body, 8);
       if (hip.GraphExecDestroy) set_sticky(hip.GraphExecDestroy((void*)g),
                                            "GraphExecDestroy");
+      break;
+    }
+    case OP_GRAPH_GET_NODES: {
+      uint64_t graph, cap;
+      memcpy(&graph, body, 8);
+      memcpy(&cap, body + 8, 8);
+      size_t count = 0;
+      hipError_t e = hip.GraphGetNodes
+                         ? hip.GraphGetNodes((void*)graph, nullptr, &count)
+                         : 801;
+      std::vector<uint8_t> outb(8);
+      uint64_t cnt = count;
+      memcpy(outb.data(), &cnt, 8);
+      if (e == 0 && cap > 0 && count > 0) {
+        size_t n = count < cap ? count : cap;
+        std::vector<void*> nodes(n);
+        size_t got = n;
+        e = hip.GraphGetNodes((void*)graph, nodes.data(), &got);
+        outb.resize(8 + got * 8);
+        memcpy(outb.data() + 8, nodes.data(), got * 8);
+      }
+      reply(c->seq, e, outb.data(), (uint32_t)outb.size());
       break;
     }
     case OP_SHUTDOWN:
