@@ -213,10 +213,17 @@ struct VmmRange {
 };
 
 struct Vmm {
+  // VA range manager WITHOUT one giant reservation: each allocation gets
+  // its own hipMemAddressReserve at an exact address inside [base,
+  // base+heap). Rationale (tf_vmm_probe characterization): ROCm 7.2's
+  // hipMemSetAccess intermittently rejects ranges deep inside a shared
+  // reservation, but the first SetAccess of a fresh reservation always
+  // succeeds — per-allocation reservations make every SetAccess exactly
+  // that. VA stability across snapshot/restore is preserved because WE
+  // choose the addresses.
   bool enabled = false;
   int device = 0;
   uint64_t base = 0;
-  uint64_t high_water = 0;  // top of the mapped region (union SetAccess)
   size_t heap_bytes = 0;
   size_t gran = 2u << 20;
   std::map<uint64_t, size_t> free_spans;          // va → len
@@ -233,17 +240,15 @@ struct Vmm {
         g == 0)
       return false;
     gran = g;
+    // probe that the base region is reservable, then release it — real
+    // reservations are per-allocation at exact addresses
+    uint64_t want = base_hint ? base_hint : VMM_BASE_HINT;
     void* p = nullptr;
-    if (hip.MemAddressReserve(&p, heap, 0, (void*)base_hint, 0) != 0)
+    if (hip.MemAddressReserve(&p, gran, 0, (void*)want, 0) != 0)
       return false;
-    if (base_hint && (uint64_t)p != base_hint) {
-      // restore REQUIRES the exact base; fresh starts accept any
-      if (base_hint != VMM_BASE_HINT) {
-        hip.MemAddressFree(p, heap);
-        return false;
-      }
-    }
-    base = (uint64_t)p;
+    hip.MemAddressFree(p, gran);
+    if ((uint64_t)p != want) return false;
+    base = want;
     heap_bytes = heap;
     free_spans[base] = heap;
     enabled = true;
@@ -253,13 +258,22 @@ struct Vmm {
   uint64_t round_up(uint64_t n) const { return (n + gran - 1) & ~(gran - 1); }
 
   hipError_t map_at(uint64_t va, size_t len, size_t req) {
+    void* got = nullptr;
+    hipError_t e = hip.MemAddressReserve(&got, len, 0, (void*)va, 0);
+    if (e != 0 || (uint64_t)got != va) {
+      fprintf(stderr, "[worker] vmm: reserve(%llx,%zu) -> %d (got %p)\n",
+              (unsigned long long)va, len, e, got);
+      if (e == 0) hip.MemAddressFree(got, len);
+      return e ? e : 1;
+    }
     hipMemAllocationProp_ prop{};
     prop.type = 1;
     prop.location = {1, device};
     void* handle = nullptr;
-    hipError_t e = hip.MemCreate(&handle, len, &prop, 0);
+    e = hip.MemCreate(&handle, len, &prop, 0);
     if (e != 0) {
       fprintf(stderr, "[worker] vmm: hipMemCreate(%zu) -> %d\n", len, e);
+      hip.MemAddressFree((void*)va, len);
       return e;
     }
     e = hip.MemMap((void*)va, len, 0, handle, 0);
@@ -267,26 +281,19 @@ struct Vmm {
       fprintf(stderr, "[worker] vmm: hipMemMap(%llx,%zu) -> %d\n",
               (unsigned long long)va, len, e);
       hip.MemRelease(handle);
+      hip.MemAddressFree((void*)va, len);
       return e;
     }
-    // ROCm 7.2 quirk (characterized by tf_vmm_probe): per-range
-    // SetAccess at offsets deep inside a reservation intermittently
-    // returns InvalidValue, but SetAccess over the union
-    // [base, high_water) always succeeds. Apply the union first; fall
-    // back to the exact range.
     hipMemAccessDesc_ acc{{1, device}, 3 /*RW*/};
-    uint64_t end = va + len;
-    uint64_t hw = end > high_water ? end : high_water;
-    e = hip.MemSetAccess((void*)base, hw - base, &acc, 1);
-    if (e != 0) e = hip.MemSetAccess((void*)va, len, &acc, 1);
+    e = hip.MemSetAccess((void*)va, len, &acc, 1);
     if (e != 0) {
       fprintf(stderr, "[worker] vmm: hipMemSetAccess(%llx,%zu) -> %d\n",
               (unsigned long long)va, len, e);
       hip.MemUnmap((void*)va, len);
       hip.MemRelease(handle);
+      hip.MemAddressFree((void*)va, len);
       return e;
     }
-    high_water = hw;
     mapped[va] = VmmRange{handle, len, req};
     return 0;
   }
@@ -333,6 +340,7 @@ struct Vmm {
     if (it == mapped.end()) return 1;
     hip.MemUnmap(p, it->second.bytes);
     hip.MemRelease(it->second.handle);
+    hip.MemAddressFree(p, it->second.bytes);
     uint64_t va = it->first;
     size_t len = it->second.bytes;
     mapped.erase(it);
