@@ -193,8 +193,54 @@ def build_app(devices: DeviceController, workers: WorkerController,
         if snapshot_fn is None:
             return JSONResponse(status_code=501, content={
                 "success": False,
-                "message": "process snapshot requires CRIU+ROCm host support"})
+                "message": "no vGPU worker manager attached"})
         res = snapshot_fn(st)
         return {"success": True, "data": res}
 
     return app
+
+
+def attach_vgpu_manager(app: FastAPI, manager) -> None:
+    """Mount vGPU worker snapshot/resume/migration routes (the endpoints
+    the reference returns 501 for — here backed by the VA-stable worker
+    snapshot machinery, hypervisor/vgpu_manager.py)."""
+
+    @app.get("/api/v1/vgpu")
+    def vgpu_status():
+        return {"success": True, "data": manager.status()}
+
+    @app.post("/api/v1/vgpu/{namespace}/{pod}/snapshot")
+    def vgpu_snapshot(namespace: str, pod: str):
+        try:
+            path = manager.snapshot(f"{namespace}/{pod}")
+        except KeyError:
+            return JSONResponse(status_code=404, content={"success": False})
+        except Exception as e:
+            return JSONResponse(status_code=500,
+                                content={"success": False, "message": str(e)})
+        return {"success": True, "data": {"snapshot": path}}
+
+    @app.post("/api/v1/vgpu/{namespace}/{pod}/resume")
+    def vgpu_resume(namespace: str, pod: str,
+                    device: Optional[int] = Query(None)):
+        try:
+            w = manager.resume(f"{namespace}/{pod}", device_index=device)
+        except KeyError:
+            return JSONResponse(status_code=404, content={"success": False})
+        except Exception as e:
+            return JSONResponse(status_code=500,
+                                content={"success": False, "message": str(e)})
+        return {"success": True, "data": {"device": w.device_index}}
+
+    @app.post("/api/v1/vgpu/{namespace}/{pod}/migrate")
+    def vgpu_migrate(namespace: str, pod: str, device: int = Query(...)):
+        try:
+            w = manager.migrate(f"{namespace}/{pod}", device)
+        except KeyError:
+            return JSONResponse(status_code=404, content={"success": False})
+        except Exception as e:
+            return JSONResponse(status_code=500,
+                                content={"success": False, "message": str(e)})
+        return {"success": True,
+                "data": {"device": w.device_index,
+                         "migrations": w.migrations}}

@@ -220,3 +220,25 @@ def test_http_api(hyp):
     assert c.post("/api/v1/workers/default/api/resume").json()["success"]
     r = c.post("/api/v1/workers/default/api/snapshot")
     assert r.status_code == 501  # honest: needs CRIU host support
+
+
+def test_vgpu_manager_routes(tmp_path):
+    """Snapshot/resume/migrate HTTP surface exists and 404s for unknown
+    workers (the reference's handlers return 501 unconditionally)."""
+
+    from fastapi.testclient import TestClient
+
+    from tensor_fusion_amd.hypervisor.main import build_hypervisor
+    from tensor_fusion_amd.hypervisor.server import (attach_vgpu_manager,
+                                                     build_app)
+    from tensor_fusion_amd.hypervisor.vgpu_manager import VgpuWorkerManager
+
+    devices, workers, erl, _ = build_hypervisor(
+        mock_devices=1, shm_root=str(tmp_path / "shm"))
+    app = build_app(devices, workers)
+    mgr = VgpuWorkerManager(run_dir=str(tmp_path / "vgpu"))
+    attach_vgpu_manager(app, mgr)
+    c = TestClient(app)
+    assert c.get("/api/v1/vgpu").json()["data"] == {}
+    assert c.post("/api/v1/vgpu/ns/pod/snapshot").status_code == 404
+    assert c.post("/api/v1/vgpu/ns/pod/migrate?device=1").status_code == 404
