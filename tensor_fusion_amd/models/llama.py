@@ -278,7 +278,9 @@ def decode_bench_graphs(model: Llama, batch: int, ctx: int, steps: int,
     with torch.cuda.graph(g):
         step_graphable()
 
-    host_pos = torch.empty(1, dtype=torch.long, pin_memory=False)
+    # pinned: torch issues a genuinely async H2D (a pageable source would
+    # fall back to a synchronous copy and serialize every remote replay)
+    host_pos = torch.empty(1, dtype=torch.long, pin_memory=True)
 
     def step(i):
         host_pos[0] = ctx + i
