@@ -1,0 +1,120 @@
+"""Admission webhook HTTP endpoint — the kube-apiserver wire contract.
+
+Reference: internal/webhook/v1 serves /mutate-v1-pod taking an
+AdmissionReview(v1) and returning a JSONPatch response. The embedded
+store path uses PodMutator directly; this endpoint exists so a real
+cluster can point a MutatingWebhookConfiguration at the operator: it
+converts corev1.Pod JSON ⇄ the internal Pod model, runs the same
+mutator, and responds with an RFC-6902 patch.
+"""
+from __future__ import annotations
+
+import base64
+import copy
+import json
+from typing import Any, Dict, List
+
+from fastapi import FastAPI, Request
+
+from ..api.types import Container, Pod
+from ..webhook import PodMutator
+
+
+def pod_from_k8s(obj: Dict[str, Any]) -> Pod:
+    p = Pod()
+    meta = obj.get("metadata", {})
+    p.meta.name = meta.get("name", "") or meta.get("generateName", "")
+    p.meta.namespace = meta.get("namespace", "default")
+    p.meta.labels = dict(meta.get("labels") or {})
+    p.meta.annotations = dict(meta.get("annotations") or {})
+    spec = obj.get("spec", {})
+    p.scheduler_name = spec.get("schedulerName", "default")
+    for c in spec.get("containers", []):
+        cc = Container(name=c.get("name", "main"),
+                       image=c.get("image", ""),
+                       command=list(c.get("command") or []))
+        for e in c.get("env") or []:
+            if "value" in e:
+                cc.env[e["name"]] = e["value"]
+        res = c.get("resources") or {}
+        for kind in ("limits", "requests"):
+            for k, v in (res.get(kind) or {}).items():
+                cc.resources[k] = str(v)
+        p.containers.append(cc)
+    return p
+
+
+def pod_to_k8s_patch(orig: Dict[str, Any], mutated: Pod) -> List[dict]:
+    """RFC-6902 patch from the original k8s pod to the mutated model."""
+
+    patch: List[dict] = []
+    meta = orig.get("metadata", {})
+    if (meta.get("labels") or {}) != mutated.meta.labels:
+        patch.append({"op": "add" if "labels" not in meta else "replace",
+                      "path": "/metadata/labels",
+                      "value": mutated.meta.labels})
+    if (meta.get("annotations") or {}) != mutated.meta.annotations:
+        patch.append({"op": "add" if "annotations" not in meta else "replace",
+                      "path": "/metadata/annotations",
+                      "value": mutated.meta.annotations})
+    if orig.get("spec", {}).get("schedulerName", "default") != \
+            mutated.scheduler_name:
+        patch.append({"op": "add", "path": "/spec/schedulerName",
+                      "value": mutated.scheduler_name})
+    for i, (oc, mc) in enumerate(zip(orig.get("spec", {}).get(
+            "containers", []), mutated.containers)):
+        oenv = {e["name"]: e.get("value") for e in (oc.get("env") or [])}
+        if oenv != mc.env:
+            patch.append({
+                "op": "add" if not oc.get("env") else "replace",
+                "path": f"/spec/containers/{i}/env",
+                "value": [{"name": k, "value": v} for k, v in mc.env.items()],
+            })
+        ores = {k: str(v) for kind in ("limits", "requests")
+                for k, v in (oc.get("resources", {}).get(kind) or {}).items()}
+        if ores != mc.resources:
+            patch.append({
+                "op": "add" if not oc.get("resources") else "replace",
+                "path": f"/spec/containers/{i}/resources",
+                "value": {"limits": dict(mc.resources),
+                          "requests": dict(mc.resources)},
+            })
+        if mc.volume_mounts and not oc.get("volumeMounts"):
+            patch.append({
+                "op": "add",
+                "path": f"/spec/containers/{i}/volumeMounts",
+                "value": mc.volume_mounts,
+            })
+    return patch
+
+
+def create_webhook_app(mutator: PodMutator) -> FastAPI:
+    app = FastAPI(title="tensor-fusion-webhook")
+
+    @app.post("/mutate-v1-pod")
+    async def mutate(request: Request):
+        review = await request.json()
+        req = review.get("request", {})
+        uid = req.get("uid", "")
+        obj = req.get("object", {})
+        resp = {"uid": uid, "allowed": True}
+        try:
+            pod = pod_from_k8s(obj)
+            if mutator.should_handle(pod):
+                mutator.handle(pod)
+                patch = pod_to_k8s_patch(obj, pod)
+                if patch:
+                    resp["patchType"] = "JSONPatch"
+                    resp["patch"] = base64.b64encode(
+                        json.dumps(patch).encode()).decode()
+        except Exception as e:
+            resp = {"uid": uid, "allowed": False,
+                    "status": {"message": f"mutation failed: {e}"}}
+        return {"apiVersion": "admission.k8s.io/v1",
+                "kind": "AdmissionReview", "response": resp}
+
+    @app.get("/healthz")
+    def healthz():
+        return {"ok": True}
+
+    return app

@@ -163,3 +163,60 @@ class TestConfig:
         assert "MI300X" in w.config.gpu_info
         assert w.config.gpu_info["MI300X"].fp16_tflops == 1300
         assert "MI355X" in w.config.gpu_info  # default retained
+
+
+class TestWebhookServer:
+    def test_admission_review_roundtrip(self):
+        import base64
+        import json as _json
+
+        from fastapi.testclient import TestClient
+
+        from tensor_fusion_amd.server.webhook_server import create_webhook_app
+        from tensor_fusion_amd.webhook import PodMutator
+
+        store = Store()
+        app = create_webhook_app(PodMutator(store))
+        c = TestClient(app)
+        review = {
+            "apiVersion": "admission.k8s.io/v1",
+            "kind": "AdmissionReview",
+            "request": {
+                "uid": "u-1",
+                "object": {
+                    "metadata": {"name": "app-1", "namespace": "default",
+                                 "labels": {C.LabelEnabled: "true"},
+                                 "annotations": {
+                                     C.AnnoTflopsRequest: "500",
+                                     C.AnnoVramRequest: "16Gi"}},
+                    "spec": {"containers": [{"name": "main",
+                                             "image": "app:1"}]},
+                },
+            },
+        }
+        r = c.post("/mutate-v1-pod", json=review)
+        assert r.status_code == 200
+        resp = r.json()["response"]
+        assert resp["allowed"] is True
+        patch = _json.loads(base64.b64decode(resp["patch"]))
+        paths = {p["path"] for p in patch}
+        assert "/spec/containers/0/env" in paths
+        # workload CR was created by the same mutator core
+        assert store.get("TensorFusionWorkload", "app-1-wl",
+                         "default") is not None
+
+    def test_non_tf_pod_untouched(self):
+        from fastapi.testclient import TestClient
+
+        from tensor_fusion_amd.server.webhook_server import create_webhook_app
+        from tensor_fusion_amd.webhook import PodMutator
+
+        app = create_webhook_app(PodMutator(Store()))
+        c = TestClient(app)
+        review = {"request": {"uid": "u-2", "object": {
+            "metadata": {"name": "plain", "namespace": "d"},
+            "spec": {"containers": [{"name": "c"}]}}}}
+        r = c.post("/mutate-v1-pod", json=review)
+        resp = r.json()["response"]
+        assert resp["allowed"] is True
+        assert "patch" not in resp
