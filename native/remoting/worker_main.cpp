@@ -438,6 +438,7 @@ struct Staging {
 
 struct Worker {
   tfrpc::Header* hdr = nullptr;
+  int cli_fd = -1;  // client socket: EOF ⇒ client exited
   tfrpc::RingView cmd;  // consumer
   tfrpc::RingView cpl;  // producer
   uint8_t* arena = nullptr;
@@ -1213,6 +1214,17 @@ int serve(tfrpc::Header* hdr) {
     uint8_t* p = W.cmd.try_next(&len);
     if (!p) {
       retire_pending();
+      // idle: is the client still alive? (EOF ⇒ tear down; a MIGRATING
+      // client keeps its socket open to the successor, so this only
+      // fires when the client process is gone)
+      if (W.cli_fd >= 0) {
+        char b;
+        ssize_t n = recv(W.cli_fd, &b, 1, MSG_DONTWAIT);
+        if (n == 0) {
+          fprintf(stderr, "[worker] client disconnected\n");
+          break;
+        }
+      }
       W.cmd.wait_nonempty();
       continue;
     }
@@ -1307,6 +1319,7 @@ int main(int argc, char** argv) {
       close(cli);
       continue;
     }
+    W.cli_fd = cli;
     if (restore_path && *restore_path) {
       if (do_restore(restore_path) != 0) {
         fprintf(stderr, "[worker] restore failed — refusing to serve\n");
