@@ -156,7 +156,11 @@ int tf_tier_gather(const void* base, void* staging, const uint32_t* idx_dev,
                    int nontemporal) {
   dim3 block(256);
   dim3 grid(copy_grid(page_bytes / 16, 256), npages);
-  if (grid.x > 64) grid.x = 64;  // npages * 64 blocks fills the chip
+  // target ≈2048 workgroups total: few big pages get wide x-grids, many
+  // small pages narrow ones (256 CUs × 8 XCDs need ≫256 WGs in flight)
+  unsigned cap = npages ? (2048u + npages - 1) / npages : 1;
+  if (cap < 1) cap = 1;
+  if (grid.x > cap) grid.x = cap;
   hipLaunchKernelGGL(gather_pages_kernel, grid, block, 0, (hipStream_t)stream,
                      (const uint8_t*)base, (uint8_t*)staging, idx_dev,
                      page_bytes, npages, nontemporal);
@@ -167,7 +171,9 @@ int tf_tier_scatter(const void* staging, void* base, const uint32_t* idx_dev,
                     size_t page_bytes, uint32_t npages, void* stream) {
   dim3 block(256);
   dim3 grid(copy_grid(page_bytes / 16, 256), npages);
-  if (grid.x > 64) grid.x = 64;
+  unsigned cap = npages ? (2048u + npages - 1) / npages : 1;
+  if (cap < 1) cap = 1;
+  if (grid.x > cap) grid.x = cap;
   hipLaunchKernelGGL(scatter_pages_kernel, grid, block, 0, (hipStream_t)stream,
                      (const uint8_t*)staging, (uint8_t*)base, idx_dev,
                      page_bytes, npages);
