@@ -252,23 +252,27 @@ def decode_bench_graphs(model: Llama, batch: int, ctx: int, steps: int,
           pos_end=ctx)
 
     cur = torch.randint(0, cfg.vocab, (batch, 1), device=device)
-    pos_buf = torch.empty(1, dtype=torch.long, device=device)
+    pos_buf = torch.tensor([ctx], dtype=torch.long, device=device)
     # additive validity mask over the FULL cache length
     mask = torch.full((1, 1, 1, total), float("-inf"), device=device,
                       dtype=dtype)
     mask[..., :ctx] = 0.0
     zero = torch.zeros(1, device=device, dtype=dtype)
+    one = torch.ones(1, dtype=torch.long, device=device)
 
     def step_graphable():
+        # fully device-driven: position advance and mask validity update
+        # are graph nodes, so a decode step is ONE graph replay with no
+        # host work at all (remote: one async ring record per token)
+        mask.view(-1).index_copy_(0, pos_buf, zero)
         logits = model(cur, pos=pos_buf, caches=caches, mask=mask)
         cur.copy_(logits.argmax(-1))
+        pos_buf.add_(one)
 
     # warmup on a side stream (allocator steady-state), then capture
     side = torch.cuda.Stream()
     side.wait_stream(torch.cuda.current_stream())
     with torch.cuda.stream(side):
-        pos_buf.copy_(torch.tensor([ctx]))
-        mask[..., ctx] = 0.0
         for _ in range(3):
             step_graphable()
     torch.cuda.current_stream().wait_stream(side)
@@ -278,22 +282,12 @@ def decode_bench_graphs(model: Llama, batch: int, ctx: int, steps: int,
     with torch.cuda.graph(g):
         step_graphable()
 
-    # pinned: torch issues a genuinely async H2D (a pageable source would
-    # fall back to a synchronous copy and serialize every remote replay)
-    host_pos = torch.empty(1, dtype=torch.long, pin_memory=True)
-
-    def step(i):
-        host_pos[0] = ctx + i
-        pos_buf.copy_(host_pos, non_blocking=True)
-        mask[..., ctx + i:ctx + i + 1].copy_(zero.view(1), non_blocking=True)
+    for _ in range(1, warmup):
         g.replay()
-
-    for i in range(1, warmup):
-        step(i)
     torch.cuda.synchronize()
     t0 = time.perf_counter()
-    for i in range(warmup, warmup + steps):
-        step(i)
+    for _ in range(steps):
+        g.replay()
     torch.cuda.synchronize()
     dt = time.perf_counter() - t0
     return batch * steps / dt, dt / steps * 1000.0
