@@ -66,19 +66,23 @@ __global__ void copy16_kernel(const u4* __restrict__ src, u4* __restrict__ dst,
 // lines as last-use in L2 so tiering traffic does not evict workload data.
 __global__ void copy16_nt_kernel(const u4* __restrict__ src,
                                  u4* __restrict__ dst, size_t n16) {
+  // blocked layout: 4 consecutive u4 per lane = 64 B/lane, so one
+  // wavefront streams 4 KB contiguously per iteration (DRAM-friendly)
   size_t stride = (size_t)gridDim.x * blockDim.x;
-  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  for (; i + 3 * stride < n16; i += 4 * stride) {
+  size_t t = blockIdx.x * blockDim.x + threadIdx.x;
+  size_t n4 = n16 / 4;
+  for (size_t j = t; j < n4; j += stride) {
+    size_t i = 4 * j;
     u4 a = __builtin_nontemporal_load(&src[i]);
-    u4 b = __builtin_nontemporal_load(&src[i + stride]);
-    u4 c = __builtin_nontemporal_load(&src[i + 2 * stride]);
-    u4 d = __builtin_nontemporal_load(&src[i + 3 * stride]);
+    u4 b = __builtin_nontemporal_load(&src[i + 1]);
+    u4 c = __builtin_nontemporal_load(&src[i + 2]);
+    u4 d = __builtin_nontemporal_load(&src[i + 3]);
     __builtin_nontemporal_store(a, &dst[i]);
-    __builtin_nontemporal_store(b, &dst[i + stride]);
-    __builtin_nontemporal_store(c, &dst[i + 2 * stride]);
-    __builtin_nontemporal_store(d, &dst[i + 3 * stride]);
+    __builtin_nontemporal_store(b, &dst[i + 1]);
+    __builtin_nontemporal_store(c, &dst[i + 2]);
+    __builtin_nontemporal_store(d, &dst[i + 3]);
   }
-  for (; i < n16; i += stride) {
+  for (size_t i = 4 * n4 + t; i < n16; i += stride) {
     u4 v = __builtin_nontemporal_load(&src[i]);
     __builtin_nontemporal_store(v, &dst[i]);
   }

@@ -242,3 +242,23 @@ def test_vgpu_manager_routes(tmp_path):
     assert c.get("/api/v1/vgpu").json()["data"] == {}
     assert c.post("/api/v1/vgpu/ns/pod/snapshot").status_code == 404
     assert c.post("/api/v1/vgpu/ns/pod/migrate?device=1").status_code == 404
+
+
+def test_hypervisor_metrics_loop(tmp_path):
+    """Hypervisor metrics: node + worker usage → influx file + TSDB
+    (reference pkg/hypervisor/metrics 60 s loop)."""
+
+    from tensor_fusion_amd.hypervisor.main import build_hypervisor
+    from tensor_fusion_amd.hypervisor.metrics import HypervisorMetrics
+    from tensor_fusion_amd.metrics import TSDB
+
+    devices, workers, erl, _ = build_hypervisor(
+        mock_devices=2, shm_root=str(tmp_path / "shm"))
+    tsdb = TSDB()
+    hm = HypervisorMetrics("node-x", devices, workers,
+                           out_dir=str(tmp_path / "m"), tsdb=tsdb)
+    n = hm.collect_once()
+    assert n >= 1
+    pts = tsdb.query("tf_node_metrics", "gpu_count", tags={"node": "node-x"})
+    assert pts and pts[-1][1] >= 1  # mock device count is process-sticky
+    assert (tmp_path / "m" / "metrics.log").exists()
