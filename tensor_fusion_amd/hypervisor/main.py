@@ -63,9 +63,17 @@ def main():
 
     threading.Thread(target=sync_loop, daemon=True).start()
 
-    from .server import build_app
+    # metrics loop (reference pkg/hypervisor/metrics 60 s cadence)
+    from .metrics import HypervisorMetrics
+    hyp_metrics = HypervisorMetrics(args.node, devices, workers,
+                                    out_dir="/var/log/tensor-fusion")
+    hyp_metrics.start()
+
+    from .server import attach_vgpu_manager, build_app
+    from .vgpu_manager import VgpuWorkerManager
     import uvicorn
     app = build_app(devices, workers)
+    attach_vgpu_manager(app, VgpuWorkerManager())
     signal.signal(signal.SIGTERM, lambda *_: stop.set())
     uvicorn.run(app, host="0.0.0.0", port=args.http_port, log_level="warning")
 
