@@ -440,6 +440,7 @@ class NodeClaimReconciler(Reconciler):
 
 def default_controllers(store: Store, allocator=None, provider=None
                         ) -> List[Reconciler]:
+    from .provider import ProviderConfigReconciler, ProviderManager
     return [
         ClusterReconciler(store),
         PoolReconciler(store),
@@ -449,4 +450,6 @@ def default_controllers(store: Store, allocator=None, provider=None
         ConnectionReconciler(store),
         PodReconciler(store, allocator=allocator),
         NodeClaimReconciler(store, provider=provider),
+        ProviderConfigReconciler(store,
+                                 ProviderManager(allocator=allocator)),
     ]

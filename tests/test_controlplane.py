@@ -380,3 +380,22 @@ class TestExpander:
         it = cheapest_instance_for(1, 2000, 100 << 30)
         assert it.name == "mi355x.1g"
         assert cheapest_instance_for(16, 0, 0) is None
+
+
+class TestProviderManager:
+    def test_partition_templates_hot_reload_into_allocator(self):
+        from tensor_fusion_amd.api.types import (ProviderConfig,
+                                                 default_mi355x_partition_templates)
+        store = Store()
+        alloc = GpuAllocator(store=store)
+        mgr = ControllerManager(store)
+        for ctrl in default_controllers(store, allocator=alloc):
+            mgr.register(ctrl)
+        pc = ProviderConfig()
+        pc.meta.name = "amd"
+        pc.partition_templates = default_mi355x_partition_templates()
+        store.create(pc)
+        mgr.reconcile_now()
+        assert len(alloc.partition_templates) == 4
+        ids = {t.id for t in alloc.partition_templates}
+        assert ids == {"xcd1", "xcd2", "xcd4", "xcd8"}
