@@ -32,19 +32,21 @@ with torch.cuda.stream(side):
         step()
 torch.cuda.current_stream().wait_stream(side)
 torch.cuda.synchronize()
-g = torch.cuda.CUDAGraph()
+try:
+    g = torch.cuda.CUDAGraph(keep_graph=True)
+except TypeError:
+    g = torch.cuda.CUDAGraph()
 with torch.cuda.graph(g):
     step()
 # node count via raw HIP
-lib = ctypes.CDLL("libamdhip64.so")
+lib = ctypes.CDLL(None)
 cnt = ctypes.c_size_t(0)
-rc = lib.hipGraphGetNodes(ctypes.c_void_p(g.cuda_graph.raw_cuda_graph() if hasattr(g, "cuda_graph") else 0), None, ctypes.byref(cnt)) if False else -1
+rc = -1
 try:
     raw = g.raw_cuda_graph()
-except AttributeError:
-    raw = None
-if raw is not None:
     rc = lib.hipGraphGetNodes(ctypes.c_void_p(raw), None, ctypes.byref(cnt))
+except Exception as e:
+    print("raw graph unavailable:", e)
 print("nodes_rc=", rc, "node_count=", cnt.value)
 # replay timing
 for _ in range(3):
