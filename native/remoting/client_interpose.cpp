@@ -1586,6 +1586,17 @@ hipError_t hipGraphGetNodes(void* graph, void** nodes, size_t* count) {
   return hipSuccess;
 }
 
+int tf_graph_node_types(void* graph, unsigned long long* hist17) {
+  std::vector<uint8_t> out(17 * 8);
+  size_t out_len = 0;
+  uint64_t g = (uint64_t)graph;
+  hipError_t e = send_sync(OP_GRAPH_NODE_TYPES, &g, 8, out.data(),
+                           out.size(), &out_len);
+  if (e != hipSuccess || out_len < 17 * 8) return -1;
+  memcpy(hist17, out.data(), 17 * 8);
+  return 0;
+}
+
 hipError_t hipGraphExecDestroy(void* graphExec) {
   uint64_t g = (uint64_t)graphExec;
   send_async(OP_GRAPH_EXEC_DESTROY, 0, &g, 8);

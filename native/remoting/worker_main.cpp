@@ -90,6 +90,7 @@ struct Hip {
   hipError_t (*GraphDestroy)(void*) = nullptr;
   hipError_t (*GraphExecDestroy)(void*) = nullptr;
   hipError_t (*GraphGetNodes)(void*, void**, size_t*) = nullptr;
+  hipError_t (*GraphNodeGetType)(void*, int*) = nullptr;
 
   // VMM surface (optional — absent on very old runtimes; worker falls back
   // to plain hipMalloc and snapshot/restore is disabled).
@@ -157,6 +158,7 @@ struct Hip {
     O(GraphDestroy, "hipGraphDestroy")
     O(GraphExecDestroy, "hipGraphExecDestroy")
     O(GraphGetNodes, "hipGraphGetNodes")
+    O(GraphNodeGetType, "hipGraphNodeGetType")
     O(MemAddressReserve, "hipMemAddressReserve")
     O(MemAddressFree, "hipMemAddressFree")
     O(MemCreate, "hipMemCreate")
@@ -657,13 +659,39 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
       l->stream = xl(l->stream);
       void* kernarg = body + sizeof(LaunchBody);
       size_t sz = l->kernarg_size;
-      void* extra[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, kernarg,
-                       HIP_LAUNCH_PARAM_BUFFER_SIZE, &sz,
-                       HIP_LAUNCH_PARAM_END};
-      hipError_t e = hip.ModuleLaunchKernel(
-          (hipFunction_t)l->func, l->grid[0], l->grid[1], l->grid[2],
-          l->block[0], l->block[1], l->block[2], l->shmem,
-          (hipStream_t)l->stream, nullptr, extra);
+      hipError_t e;
+      static int use_params = getenv("TF_WORKER_LAUNCH_PARAMS") ? 1 : 0;
+      const tfrpc::KernelSig* sig = nullptr;
+      if (use_params) {
+        auto fit = W.funcs.find(l->func);
+        if (fit != W.funcs.end()) {
+          auto mit = W.sigs.find(fit->second.image_id);
+          if (mit != W.sigs.end()) {
+            auto sit = mit->second.find(fit->second.name);
+            if (sit != mit->second.end()) sig = &sit->second;
+          }
+        }
+      }
+      if (sig && !sig->args.empty()) {
+        // per-arg params path: lets ROCr pack kernargs at enqueue instead
+        // of recording a kernarg staging copy as a second graph node
+        void* params[64];
+        size_t n = sig->args.size() < 64 ? sig->args.size() : 64;
+        for (size_t i = 0; i < n; ++i)
+          params[i] = (uint8_t*)kernarg + sig->args[i].offset;
+        e = hip.ModuleLaunchKernel(
+            (hipFunction_t)l->func, l->grid[0], l->grid[1], l->grid[2],
+            l->block[0], l->block[1], l->block[2], l->shmem,
+            (hipStream_t)l->stream, params, nullptr);
+      } else {
+        void* extra[] = {HIP_LAUNCH_PARAM_BUFFER_POINTER, kernarg,
+                         HIP_LAUNCH_PARAM_BUFFER_SIZE, &sz,
+                         HIP_LAUNCH_PARAM_END};
+        e = hip.ModuleLaunchKernel(
+            (hipFunction_t)l->func, l->grid[0], l->grid[1], l->grid[2],
+            l->block[0], l->block[1], l->block[2], l->shmem,
+            (hipStream_t)l->stream, nullptr, extra);
+      }
       if (W.verbose)
         fprintf(stderr,
                 "[worker] launch fn=%llx grid=%u,%u,%u block=%u,%u,%u "
@@ -947,6 +975,30 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
         memcpy(outb.data() + 8, nodes.data(), got * 8);
       }
       reply(c->seq, e, outb.data(), (uint32_t)outb.size());
+      break;
+    }
+    case OP_GRAPH_NODE_TYPES: {
+      uint64_t graph;
+      memcpy(&graph, body, 8);
+      uint64_t out[17] = {0};
+      hipError_t e = 801;
+      if (hip.GraphGetNodes && hip.GraphNodeGetType) {
+        size_t count = 0;
+        e = hip.GraphGetNodes((void*)graph, nullptr, &count);
+        if (e == 0 && count) {
+          std::vector<void*> nodes(count);
+          size_t got = count;
+          e = hip.GraphGetNodes((void*)graph, nodes.data(), &got);
+          out[0] = got;
+          for (size_t i = 0; i < got && e == 0; ++i) {
+            int t = 15;
+            hip.GraphNodeGetType(nodes[i], &t);
+            if (t < 0 || t > 15) t = 15;
+            out[1 + t]++;
+          }
+        }
+      }
+      reply(c->seq, e, out, sizeof out);
       break;
     }
     case OP_SHUTDOWN:

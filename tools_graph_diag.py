@@ -48,6 +48,14 @@ try:
 except Exception as e:
     print("raw graph unavailable:", e)
 print("nodes_rc=", rc, "node_count=", cnt.value)
+try:
+    hist = (ctypes.c_ulonglong * 17)()
+    fn = lib.tf_graph_node_types
+    if fn(ctypes.c_void_p(raw), hist) == 0:
+        print("node_types:", {i: int(hist[1 + i]) for i in range(16)
+                              if hist[1 + i]})
+except Exception as e:
+    print("no node type helper:", e)
 # replay timing
 for _ in range(3):
     g.replay()
