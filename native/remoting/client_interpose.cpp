@@ -1586,6 +1586,19 @@ hipError_t hipGraphGetNodes(void* graph, void** nodes, size_t* count) {
   return hipSuccess;
 }
 
+int tf_graph_kernel_histo(void* graph, char* out, size_t cap) {
+  std::vector<uint8_t> buf(64 << 10);
+  size_t out_len = 0;
+  uint64_t g = (uint64_t)graph;
+  hipError_t e = send_sync(OP_GRAPH_KERNEL_HISTO, &g, 8, buf.data(),
+                           buf.size(), &out_len);
+  if (e != hipSuccess) return -1;
+  size_t n = out_len < cap - 1 ? out_len : cap - 1;
+  memcpy(out, buf.data(), n);
+  out[n] = 0;
+  return 0;
+}
+
 int tf_graph_node_types(void* graph, unsigned long long* hist17) {
   std::vector<uint8_t> out(17 * 8);
   size_t out_len = 0;

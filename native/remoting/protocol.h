@@ -133,6 +133,7 @@ enum Op : uint32_t {
   OP_GRAPH_EXEC_DESTROY, // {graphExec} async
   OP_GRAPH_GET_NODES,    // {graph, cap} → reply {count, nodes[...]}
   OP_GRAPH_NODE_TYPES,   // {graph} → reply {count, type_histogram[16]}
+  OP_GRAPH_KERNEL_HISTO, // {graph} → reply {text histogram of node kernels}
 };
 
 constexpr uint32_t F_WANT_REPLY = 1u << 0;

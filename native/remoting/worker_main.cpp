@@ -27,6 +27,7 @@
 #include <sys/un.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <map>
 #include <string>
 #include <unordered_map>
@@ -91,6 +92,7 @@ struct Hip {
   hipError_t (*GraphExecDestroy)(void*) = nullptr;
   hipError_t (*GraphGetNodes)(void*, void**, size_t*) = nullptr;
   hipError_t (*GraphNodeGetType)(void*, int*) = nullptr;
+  hipError_t (*GraphKernelNodeGetParams)(void*, void*) = nullptr;
 
   // VMM surface (optional — absent on very old runtimes; worker falls back
   // to plain hipMalloc and snapshot/restore is disabled).
@@ -159,6 +161,7 @@ struct Hip {
     O(GraphExecDestroy, "hipGraphExecDestroy")
     O(GraphGetNodes, "hipGraphGetNodes")
     O(GraphNodeGetType, "hipGraphNodeGetType")
+    O(GraphKernelNodeGetParams, "hipGraphKernelNodeGetParams")
     O(MemAddressReserve, "hipMemAddressReserve")
     O(MemAddressFree, "hipMemAddressFree")
     O(MemCreate, "hipMemCreate")
@@ -999,6 +1002,65 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
         }
       }
       reply(c->seq, e, out, sizeof out);
+      break;
+    }
+    case OP_GRAPH_KERNEL_HISTO: {
+      struct KNP {  // hipKernelNodeParams (ROCm 7 layout)
+        uint32_t blockDim[3];
+        uint32_t _pad0;
+        void** extra;
+        void* func;
+        uint32_t gridDim[3];
+        uint32_t _pad1;
+        void** kernelParams;
+        unsigned int sharedMemBytes;
+        uint32_t _pad2;
+      };
+      uint64_t graph;
+      memcpy(&graph, body, 8);
+      std::string text;
+      hipError_t e = 801;
+      if (hip.GraphGetNodes && hip.GraphNodeGetType &&
+          hip.GraphKernelNodeGetParams) {
+        size_t count = 0;
+        e = hip.GraphGetNodes((void*)graph, nullptr, &count);
+        std::vector<void*> nodes(count);
+        if (e == 0 && count) {
+          size_t got = count;
+          e = hip.GraphGetNodes((void*)graph, nodes.data(), &got);
+          std::map<std::string, int> histo;
+          for (size_t i = 0; i < got; ++i) {
+            int t = -1;
+            hip.GraphNodeGetType(nodes[i], &t);
+            if (t != 0) {
+              histo["<node type " + std::to_string(t) + ">"]++;
+              continue;
+            }
+            KNP p{};
+            if (hip.GraphKernelNodeGetParams(nodes[i], &p) != 0) {
+              histo["<params failed>"]++;
+              continue;
+            }
+            auto fit = W.funcs.find((uint64_t)p.func);
+            char buf[160];
+            const char* nm = fit != W.funcs.end()
+                                 ? fit->second.name.c_str()
+                                 : "<unknown func>";
+            snprintf(buf, sizeof buf, "%.100s g=%u,%u,%u b=%u,%u,%u", nm,
+                     p.gridDim[0], p.gridDim[1], p.gridDim[2], p.blockDim[0],
+                     p.blockDim[1], p.blockDim[2]);
+            histo[buf]++;
+          }
+          std::vector<std::pair<int, std::string>> top;
+          for (auto& [k, v] : histo) top.push_back({v, k});
+          std::sort(top.rbegin(), top.rend());
+          for (size_t i = 0; i < top.size() && i < 40; ++i)
+            text += std::to_string(top[i].first) + "x " + top[i].second +
+                    "\n";
+          if (text.size() > 60000) text.resize(60000);
+        }
+      }
+      reply(c->seq, e, text.data(), (uint32_t)text.size());
       break;
     }
     case OP_SHUTDOWN:

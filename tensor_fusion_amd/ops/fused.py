@@ -1,0 +1,86 @@
+"""ctypes wrapper over libtfops.so (native/ops/fused_ops.hip).
+
+Fused normalization kernels for the decode hot path. Opt-in for the
+model via TF_FUSED_OPS=1 (numerics validated against the fp32 torch
+reference in tests/test_gpu_fused.py); fails loudly on a GPU box if the
+extension is missing.
+"""
+from __future__ import annotations
+
+import ctypes
+import os
+from typing import Optional
+
+_LIB: Optional[ctypes.CDLL] = None
+
+
+class FusedOpsMissing(RuntimeError):
+    pass
+
+
+def _native_dir() -> str:
+    return os.path.join(
+        os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
+        "_native")
+
+
+def lib() -> ctypes.CDLL:
+    global _LIB
+    if _LIB is None:
+        path = os.path.join(_native_dir(), "libtfops.so")
+        if not os.path.exists(path):
+            raise FusedOpsMissing(
+                f"{path} not built — run python build_native.py")
+        _LIB = ctypes.CDLL(path)
+        for fn in ("tf_rmsnorm", "tf_add_rmsnorm"):
+            getattr(_LIB, fn).restype = ctypes.c_int
+        _LIB.tf_rmsnorm.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+            ctypes.c_int, ctypes.c_float, ctypes.c_void_p]
+        _LIB.tf_add_rmsnorm.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p,
+            ctypes.c_void_p, ctypes.c_int, ctypes.c_int, ctypes.c_float,
+            ctypes.c_void_p]
+    return _LIB
+
+
+def available() -> bool:
+    return os.path.exists(os.path.join(_native_dir(), "libtfops.so"))
+
+
+def _stream() -> int:
+    import torch
+    return torch.cuda.current_stream().cuda_stream
+
+
+def rmsnorm(x, weight, eps: float):
+    """out = x * rsqrt(mean(x², -1) + eps) * weight — bf16, any leading
+    shape, last dim % 8 == 0."""
+
+    import torch
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    dim = x.shape[-1]
+    rows = x.numel() // dim
+    out = torch.empty_like(x)
+    rc = lib().tf_rmsnorm(x.data_ptr(), weight.data_ptr(), out.data_ptr(),
+                          rows, dim, eps, _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_rmsnorm failed: {rc}")
+    return out
+
+
+def add_rmsnorm(x, residual, weight, eps: float):
+    """residual += x (in place); returns rmsnorm(residual) * weight."""
+
+    import torch
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    assert residual.is_contiguous()
+    dim = x.shape[-1]
+    rows = x.numel() // dim
+    out = torch.empty_like(x)
+    rc = lib().tf_add_rmsnorm(x.data_ptr(), residual.data_ptr(),
+                              weight.data_ptr(), out.data_ptr(), rows, dim,
+                              eps, _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_add_rmsnorm failed: {rc}")
+    return out

@@ -56,6 +56,14 @@ try:
                               if hist[1 + i]})
 except Exception as e:
     print("no node type helper:", e)
+try:
+    buf = ctypes.create_string_buffer(64 << 10)
+    if lib.tf_graph_kernel_histo(ctypes.c_void_p(raw), buf,
+                                 ctypes.c_size_t(len(buf))) == 0:
+        print("kernel histo:")
+        print(buf.value.decode()[:4000])
+except Exception as e:
+    print("no histo helper:", e)
 # replay timing
 for _ in range(3):
     g.replay()
