@@ -149,6 +149,21 @@ def build_tiering():
     return target
 
 
+def build_fused_ops():
+    """Fused normalization kernels (gfx950)."""
+
+    src = os.path.join(NATIVE, "ops", "fused_ops.hip")
+    if not os.path.exists(src):
+        return None
+    os.makedirs(OUT, exist_ok=True)
+    target = os.path.join(OUT, "libtfops.so")
+    if not _newer(target, [src]):
+        return target
+    _run([HIPCC, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
+          "-shared", src, "-o", target])
+    return target
+
+
 ALL = {
     "limiter": build_limiter,
     "mockhip": build_mockhip,
@@ -158,6 +173,7 @@ ALL = {
     "accel_dump": build_accel_dump,
     "remoting": build_remoting,
     "tiering": build_tiering,
+    "fused_ops": build_fused_ops,
 }
 
 

@@ -47,6 +47,9 @@ CONFIGS = {
 }
 
 
+_FUSED_OPS = os.environ.get("TF_FUSED_OPS") == "1"
+
+
 class RMSNorm(nn.Module):
     def __init__(self, dim: int, eps: float):
         super().__init__()
@@ -54,6 +57,12 @@ class RMSNorm(nn.Module):
         self.weight = nn.Parameter(torch.ones(dim))
 
     def forward(self, x):
+        if (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
+                and x.shape[-1] % 8 == 0 and x.is_contiguous()):
+            # one gfx950 kernel instead of ~5 eager kernels (ops/fused.py;
+            # numerics vs the fp32 reference: tests/test_gpu_fused.py)
+            from ..ops import fused
+            return fused.rmsnorm(x, self.weight, self.eps)
         dt = x.dtype
         x = x.float()
         x = x * torch.rsqrt(x.pow(2).mean(-1, keepdim=True) + self.eps)
