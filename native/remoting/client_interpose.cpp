@@ -1422,6 +1422,14 @@ hipError_t hipDeviceGetPCIBusId(char* id, int len, int dev) {
 
 hipError_t hipGetStreamDeviceId(void*) { return Client::cur_device; }
 
+hipError_t hipStreamGetDevice(void*, int* device) {
+  // aotriton gates flash attention on this: falling through to the
+  // GPU-less local runtime made SDPA silently use the math path remotely
+  if (!device) return hipErrorInvalidValue;
+  *device = Client::cur_device;
+  return hipSuccess;
+}
+
 struct hipFuncAttributes_small {
   int binaryVersion, cacheModeCA;
   size_t constSizeBytes, localSizeBytes;
