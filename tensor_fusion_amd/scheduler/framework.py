@@ -244,6 +244,9 @@ class Scheduler:
                 allowed = set(nodes) if allowed is None else allowed & set(nodes)
 
         cand = [n for n in self._nodes() if allowed is None or n in allowed]
+        # kube-scheduler's numFeasibleNodesToFind: at scale, stop after
+        # enough feasible nodes instead of filtering/scoring all of them
+        want = max(100, len(cand) * 8 // 100)
         feasible = []
         reasons: List[str] = []
         for n in cand:
@@ -256,6 +259,8 @@ class Scheduler:
                     break
             if ok:
                 feasible.append(n)
+                if len(feasible) >= want:
+                    break
 
         if not feasible:
             # PostFilter (preemption) — first plugin that succeeds wins.
