@@ -214,3 +214,35 @@ def test_fractional_vgpu_resnet50_config2():
     assert full["img_s"] > 0 and frac["img_s"] > 0
     # the 25% vGPU must be meaningfully slower than the full device
     assert frac["img_s"] < 0.7 * full["img_s"], (full, frac)
+
+
+def test_multi_tenant_oversubscription():
+    """BASELINE config 4 shape (scaled down): several vGPUs whose caps
+    oversubscribe HBM keep running concurrently — over-cap slabs spill to
+    the host tier per worker."""
+
+    import threading
+    results = {}
+
+    def run_tenant(i):
+        env = dict(os.environ)
+        env["LD_PRELOAD"] = LIMITER
+        env.pop("TF_SHM_PATH", None)
+        env.update({"TF_VRAM_LIMIT_BYTES": str(2 << 30),
+                    "TF_VRAM_EXPAND": "1"})
+        out = subprocess.run([sys.executable, "-c", EXPAND_CHILD], env=env,
+                             capture_output=True, text=True, timeout=420)
+        results[i] = (out.returncode, out.stdout, out.stderr[-500:])
+
+    threads = [threading.Thread(target=run_tenant, args=(i,))
+               for i in range(4)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join(timeout=480)
+    assert len(results) == 4
+    for i, (rc, stdout, stderr) in results.items():
+        assert rc == 0, (i, stderr)
+        r = json.loads(stdout.strip().splitlines()[-1])
+        assert r["ok"] and r["ok_after_demote"], (i, r)
+        assert r["expanded"] > 0, (i, r)

@@ -80,3 +80,21 @@ def test_tp_matches_reference_2rank():
         env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
     assert "TP_OK" in out.stdout, out.stdout
+
+
+def test_bench_tp_tool_2rank_cpu():
+    """tools/bench_tp.py runs end-to-end over gloo on CPU (tiny model)."""
+
+    env = dict(os.environ)
+    env["MASTER_ADDR"] = "127.0.0.1"
+    out = subprocess.run(
+        [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
+         "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
+         "--master-port", "29531", "tools/bench_tp.py", "--model", "tiny",
+         "--batch", "2", "--ctx", "8", "--steps", "2", "--warmup", "1",
+         "--device", "cpu", "--backend", "gloo"],
+        env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    import json
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["value"] > 0 and r["parallelism"] == "tp2"
