@@ -149,6 +149,18 @@ def build_tiering():
     return target
 
 
+def build_bw_sweep():
+    src = os.path.join(NATIVE, "tiering", "bw_sweep.hip")
+    if not os.path.exists(src):
+        return None
+    target = os.path.join(OUT, "tf_bw_sweep")
+    if not _newer(target, [src]):
+        return target
+    _run([HIPCC, "--offload-arch=gfx950", "-O3", "-std=c++17", src, "-o",
+          target])
+    return target
+
+
 def build_fused_ops():
     """Fused normalization kernels (gfx950)."""
 
@@ -174,6 +186,7 @@ ALL = {
     "remoting": build_remoting,
     "tiering": build_tiering,
     "fused_ops": build_fused_ops,
+    "bw_sweep": build_bw_sweep,
 }
 
 
