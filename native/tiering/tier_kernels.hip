@@ -40,25 +40,15 @@ namespace {
 // builtins reject; the clang ext_vector maps straight to dwordx4 ops).
 typedef uint32_t u4 __attribute__((ext_vector_type(4)));
 
-// Plain vectorized copy, grid-strided, 4 independent 16-B accesses per
-// iteration: the unroll keeps 4 loads in flight per lane (64 B/lane/iter),
-// which is what closes the gap to the ~6.3 TB/s float4-copy ceiling the
-// microarch sheet measures — a single dependent load/store chain leaves
-// the memory queues underfed.
+// Plain vectorized copy, grid-strided. Swept on hardware
+// (native/tiering/bw_sweep.hip, gpurun_out/bw_sweep.log): the SIMPLE
+// 16 B/lane loop at grid≈1024 (4 waves/CU) is the fastest variant
+// (5.38 TB/s for a 4 GiB copy); 4×-unrolled variants LOSE (4.6–5.2),
+// and 64 B/lane blocked non-temporal collapses to 1.1 TB/s.
 __global__ void copy16_kernel(const u4* __restrict__ src, u4* __restrict__ dst,
                               size_t n16) {
-  size_t stride = (size_t)gridDim.x * blockDim.x;
   size_t i = blockIdx.x * blockDim.x + threadIdx.x;
-  for (; i + 3 * stride < n16; i += 4 * stride) {
-    u4 a = src[i];
-    u4 b = src[i + stride];
-    u4 c = src[i + 2 * stride];
-    u4 d = src[i + 3 * stride];
-    dst[i] = a;
-    dst[i + stride] = b;
-    dst[i + 2 * stride] = c;
-    dst[i + 3 * stride] = d;
-  }
+  size_t stride = (size_t)gridDim.x * blockDim.x;
   for (; i < n16; i += stride) dst[i] = src[i];
 }
 
@@ -66,23 +56,9 @@ __global__ void copy16_kernel(const u4* __restrict__ src, u4* __restrict__ dst,
 // lines as last-use in L2 so tiering traffic does not evict workload data.
 __global__ void copy16_nt_kernel(const u4* __restrict__ src,
                                  u4* __restrict__ dst, size_t n16) {
-  // blocked layout: 4 consecutive u4 per lane = 64 B/lane, so one
-  // wavefront streams 4 KB contiguously per iteration (DRAM-friendly)
+  size_t i = blockIdx.x * blockDim.x + threadIdx.x;
   size_t stride = (size_t)gridDim.x * blockDim.x;
-  size_t t = blockIdx.x * blockDim.x + threadIdx.x;
-  size_t n4 = n16 / 4;
-  for (size_t j = t; j < n4; j += stride) {
-    size_t i = 4 * j;
-    u4 a = __builtin_nontemporal_load(&src[i]);
-    u4 b = __builtin_nontemporal_load(&src[i + 1]);
-    u4 c = __builtin_nontemporal_load(&src[i + 2]);
-    u4 d = __builtin_nontemporal_load(&src[i + 3]);
-    __builtin_nontemporal_store(a, &dst[i]);
-    __builtin_nontemporal_store(b, &dst[i + 1]);
-    __builtin_nontemporal_store(c, &dst[i + 2]);
-    __builtin_nontemporal_store(d, &dst[i + 3]);
-  }
-  for (size_t i = 4 * n4 + t; i < n16; i += stride) {
+  for (; i < n16; i += stride) {
     u4 v = __builtin_nontemporal_load(&src[i]);
     __builtin_nontemporal_store(v, &dst[i]);
   }
@@ -128,10 +104,10 @@ __global__ void scatter_pages_kernel(const uint8_t* __restrict__ staging,
 }
 
 inline int copy_grid(size_t n16, int block = 256) {
-  // ≫256 workgroups to cover 8 XCDs × 32 CUs with several blocks each;
-  // with the 4× unroll each workgroup covers 4·block elements per pass.
-  size_t want = (n16 + 4 * block - 1) / (4 * block);
-  size_t cap = 2048;  // 8 waves per CU over 256 CUs
+  // hardware-swept optimum: 1024 workgroups (4 per CU) saturates HBM;
+  // bigger grids only add scheduling overhead (bw_sweep.hip)
+  size_t want = (n16 + block - 1) / block;
+  size_t cap = 1024;
   return (int)(want < cap ? (want ? want : 1) : cap);
 }
 
