@@ -440,7 +440,8 @@ class NodeClaimReconciler(Reconciler):
 
 def default_controllers(store: Store, allocator=None, provider=None
                         ) -> List[Reconciler]:
-    from .provider import ProviderConfigReconciler, ProviderManager
+    from .provider import (ProviderConfigReconciler, ProviderManager,
+                           SchedulingConfigReconciler)
     return [
         ClusterReconciler(store),
         PoolReconciler(store),
@@ -452,4 +453,5 @@ def default_controllers(store: Store, allocator=None, provider=None
         NodeClaimReconciler(store, provider=provider),
         ProviderConfigReconciler(store,
                                  ProviderManager(allocator=allocator)),
+        SchedulingConfigReconciler(store, allocator=allocator),
     ]

@@ -61,3 +61,33 @@ class ProviderConfigReconciler(Reconciler):
         pc = self.store.get(self.kind, req.name, req.namespace)
         self.manager.apply(pc)
         return 0.0
+
+
+class SchedulingConfigReconciler(Reconciler):
+    """SchedulingConfigTemplate → live scheduler/allocator behavior:
+    placement mode (CompactFirst/LowLoadFirst/default), score weights and
+    ERL gains hot-swap without restarts (reference
+    schedulingconfigtemplate controller + scheduler_config.go)."""
+
+    kind = "SchedulingConfigTemplate"
+
+    def __init__(self, store: Store, allocator=None, erl_params=None):
+        super().__init__(store)
+        self.allocator = allocator
+        self.erl_params = erl_params  # mutated in place for the ERL loop
+
+    def reconcile(self, req: Request):
+        tpl = self.store.get(self.kind, req.name, req.namespace)
+        if self.allocator is not None:
+            from ..allocator.strategy import make_strategy
+            self.allocator.strategy = make_strategy(
+                tpl.placement_mode,
+                vram_weight=tpl.vram_weight,
+                tflops_weight=tpl.tflops_weight)
+        if self.erl_params is not None:
+            src = tpl.erl
+            for f in ("kp", "ki", "kd", "ema_alpha", "deadband_percent",
+                      "slew_up_percent", "slew_down_percent",
+                      "loop_interval_s", "min_rate", "max_rate"):
+                setattr(self.erl_params, f, getattr(src, f))
+        return 0.0

@@ -399,3 +399,22 @@ class TestProviderManager:
         assert len(alloc.partition_templates) == 4
         ids = {t.id for t in alloc.partition_templates}
         assert ids == {"xcd1", "xcd2", "xcd4", "xcd8"}
+
+
+class TestSchedulingConfig:
+    def test_placement_mode_hot_swap(self):
+        from tensor_fusion_amd.api.types import SchedulingConfigTemplate
+        store = Store()
+        alloc = GpuAllocator(store=store)
+        mgr = ControllerManager(store)
+        for ctrl in default_controllers(store, allocator=alloc):
+            mgr.register(ctrl)
+        assert type(alloc.strategy).__name__ == "NodeCompactGPULowLoad"
+        tpl = SchedulingConfigTemplate()
+        tpl.meta.name = "t1"
+        tpl.placement_mode = "LowLoadFirst"
+        tpl.vram_weight = 0.5
+        store.create(tpl)
+        mgr.reconcile_now()
+        assert type(alloc.strategy).__name__ == "LowLoadFirst"
+        assert alloc.strategy.vram_weight == 0.5
