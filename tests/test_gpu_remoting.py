@@ -144,3 +144,18 @@ def test_graph_decode_remoted(worker):
     assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-4000:]
     r = json.loads(out.stdout.strip().splitlines()[-1])
     assert r["tok_s"] > 0
+
+
+def test_resnet50_inference_remoted(worker):
+    """Conv workloads through GPU-over-IP: MIOpen loads its kernels via a
+    different path than rocBLAS/Tensile — this catches interception gaps
+    beyond the Llama surface."""
+
+    env = client_env(worker.socket_path)
+    out = subprocess.run(
+        [sys.executable, "-m", "tensor_fusion_amd.models.resnet",
+         "--batch", "16", "--steps", "8", "--warmup", "2"],
+        capture_output=True, text=True, timeout=900, env=env, cwd=REPO)
+    assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-4000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["img_s"] > 0
