@@ -32,17 +32,31 @@ _QOS_BY_FRACTION = ((0.75, C.QosHigh), (0.25, C.QosMedium), (0.0, C.QosLow))
 
 
 class PodMutator:
-    def __init__(self, store: Store, index_allocator=None, port_allocator=None):
+    def __init__(self, store: Store, index_allocator=None, port_allocator=None,
+                 adoption_percent: int = 100):
         self.store = store
         self.index_allocator = index_allocator
         self.port_allocator = port_allocator
+        # grey release (reference pod_counter.go): only N % of eligible
+        # plain-GPU pods are auto-migrated; explicit opt-ins always pass
+        self.adoption_percent = adoption_percent
+        self.counters = {"seen": 0, "handled": 0, "skipped_grey": 0}
 
     # ------------------------------------------------------------ gating
 
     def should_handle(self, pod: Pod) -> bool:
+        self.counters["seen"] += 1
         if pod.meta.labels.get(C.LabelEnabled) == "true":
             return True
-        return self.should_auto_migrate(pod)
+        if not self.should_auto_migrate(pod):
+            return False
+        if self.adoption_percent < 100:
+            import zlib
+            bucket = zlib.crc32(pod.meta.key.encode()) % 100
+            if bucket >= self.adoption_percent:
+                self.counters["skipped_grey"] += 1
+                return False
+        return True
 
     def should_auto_migrate(self, pod: Pod) -> bool:
         """Adopt pods that request a plain GPU extended resource
@@ -103,6 +117,7 @@ class PodMutator:
 
         if not self.should_handle(pod):
             return pod
+        self.counters["handled"] += 1
         profile = self.parse(pod)
         # persist parsed values so re-admission (and controllers) see them
         # even after plain-GPU resources are stripped below

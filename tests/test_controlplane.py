@@ -418,3 +418,24 @@ class TestSchedulingConfig:
         mgr.reconcile_now()
         assert type(alloc.strategy).__name__ == "LowLoadFirst"
         assert alloc.strategy.vram_weight == 0.5
+
+
+class TestGreyRelease:
+    def test_adoption_percent_gates_plain_gpu_pods(self):
+        store = Store()
+        m = PodMutator(store, adoption_percent=0)
+        pod = Pod()
+        pod.meta.name = "legacy2"
+        pod.meta.namespace = "d"
+        pod.containers = [Container(name="m", resources={"amd.com/gpu": "1"})]
+        assert m.should_handle(pod) is False
+        assert m.counters["skipped_grey"] == 1
+        # explicit opt-in bypasses the grey gate
+        pod.meta.labels[C.LabelEnabled] = "true"
+        assert m.should_handle(pod) is True
+        m100 = PodMutator(store, adoption_percent=100)
+        pod2 = Pod()
+        pod2.meta.name = "legacy3"
+        pod2.meta.namespace = "d"
+        pod2.containers = [Container(name="m", resources={"amd.com/gpu": "1"})]
+        assert m100.should_handle(pod2) is True
