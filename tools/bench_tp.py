@@ -42,7 +42,9 @@ def main():
     rank = dist.get_rank()
     world = dist.get_world_size()
     if args.device == "cuda":
-        torch.cuda.set_device(int(os.environ.get("LOCAL_RANK", rank)))
+        dev = os.environ.get("TF_BENCH_DEVICE")  # test override: shared GPU
+        torch.cuda.set_device(int(dev) if dev is not None
+                              else int(os.environ.get("LOCAL_RANK", rank)))
 
     from tensor_fusion_amd.models.llama import decode_bench
     from tensor_fusion_amd.parallel.tp import build_tp_model
