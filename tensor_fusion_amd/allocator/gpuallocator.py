@@ -644,6 +644,7 @@ class GpuAllocator:
         annotations (reference reconcileAllocationState :2906)."""
 
         with self._mu:
+            self._soa_invalidate()  # bulk reset below bypasses row sync
             # Rebuild from scratch: reset devices to capacity, then reapply
             # every record (the store's synced status already reflects the
             # old allocations — replaying on top would double-subtract).

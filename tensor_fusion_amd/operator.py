@@ -72,6 +72,31 @@ class Operator:
             self.store.update(pod)
         return pod
 
+    def recover(self) -> int:
+        """Operator-restart recovery: rebuild the allocator's committed
+        state from scheduled worker pods' gpu-ids annotations (reference
+        reconcileAllocationState gpuallocator.go:2906)."""
+
+        from .utils.resource import (compose_allocation_request,
+                                     profile_from_annotations)
+        records = []
+        for pod in self.store.list("Pod"):
+            gpus = pod.meta.annotations.get(C.AnnoContainerGpus)
+            if not gpus or not pod.status.node:
+                continue
+            if pod.meta.labels.get(C.LabelComponent) not in (
+                    C.ComponentWorker, None) and \
+                    not pod.meta.annotations.get(C.AnnoIsLocalGpu):
+                continue
+            try:
+                profile = profile_from_annotations(pod)
+                req = compose_allocation_request(pod, profile)
+            except (ValueError, KeyError):
+                continue
+            records.append((req, gpus.split(",")))
+        self.allocator.reconcile_from_allocations(records)
+        return len(records)
+
     # -------------------------------------------------------- lifecycle
 
     def tick(self):

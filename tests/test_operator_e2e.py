@@ -181,3 +181,45 @@ class TestE2E:
         assert hp.status.node, "critical pod should preempt"
         lp = op.store.get("Pod", "low", "default")
         assert lp.status.phase == "Failed" or not lp.status.node
+
+
+class TestRestartRecovery:
+    def test_operator_restart_rebuilds_allocations(self, tmp_path):
+        """a fresh operator over the persisted store rebuilds
+        committed allocations from worker pod annotations."""
+
+        persist = str(tmp_path / "state")
+        op = build_operator(persist_dir=persist)
+        mk_world(op, nodes=1, gpus_per_node=2)
+        pod = client_pod(annotations={
+            C.AnnoTflopsRequest: "600",
+            C.AnnoVramRequest: str(48 << 30)})
+        op.admit(pod)
+        for _ in range(6):
+            op.tick()
+        workers = [p for p in op.store.list("Pod", namespace="default")
+                   if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker]
+        assert workers and workers[0].status.node
+        wkey = workers[0].meta.key
+        gpu_used = workers[0].meta.annotations[C.AnnoContainerGpus]
+
+        # "restart": fresh operator over the same persisted store
+        op2 = build_operator(persist_dir=persist)
+        assert op2.allocator.allocation(wkey) is None
+        n = op2.recover()
+        assert n == 1
+        alloc = op2.allocator.allocation(wkey)
+        assert alloc is not None and alloc.bound
+        g = op2.allocator.gpu(gpu_used)
+        assert g.status.available.tflops <= g.status.capacity.tflops - 600
+        # a new pod still schedules correctly against recovered state
+        pod2 = client_pod(name="app-2", annotations={
+            C.AnnoTflopsRequest: "600",
+            C.AnnoVramRequest: str(48 << 30)})
+        op2.admit(pod2)
+        for _ in range(6):
+            op2.tick()
+        w2 = [p for p in op2.store.list("Pod", namespace="default")
+              if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker
+              and p.meta.labels.get(C.LabelWorkload) == "app-2-wl"]
+        assert w2 and w2[0].status.node
