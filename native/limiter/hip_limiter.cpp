@@ -99,6 +99,12 @@ struct Limiter {
   std::atomic<uint64_t> promoted_bytes{0};
   std::atomic<int> tier_thread_running{0};
 
+  // ---- per-call latency histograms (SURVEY §5.1 tracing) ----
+  // log2-bucketed ns per op class, enabled with TF_LIMITER_TRACE=1:
+  // class 0 = kernel launches, 1 = memcpy/memset, 2 = sync/other.
+  bool trace = false;
+  std::atomic<uint64_t> lat_hist[3][32] = {};
+
   Limiter();  // all init lives in the constructor: the instance is a Meyers
               // singleton so there is no static-init-order hazard between
               // the ELF constructor and this TU's dynamic initializers.
@@ -115,6 +121,26 @@ uint64_t now_ns() {
   clock_gettime(CLOCK_MONOTONIC, &ts);
   return uint64_t(ts.tv_sec) * 1000000000ull + ts.tv_nsec;
 }
+
+void record_latency(int cls, uint64_t t0_ns) {
+  uint64_t dt = now_ns() - t0_ns;
+  int b = 0;
+  while ((1ull << (b + 1)) <= dt && b < 31) ++b;
+  g.lat_hist[cls][b].fetch_add(1, std::memory_order_relaxed);
+}
+
+#define TF_TRACE_CALL(cls, expr)                 \
+  ({                                             \
+    hipError_t _r;                               \
+    if (g.trace) {                               \
+      uint64_t _t0 = now_ns();                   \
+      _r = (expr);                               \
+      record_latency(cls, _t0);                  \
+    } else {                                     \
+      _r = (expr);                               \
+    }                                            \
+    _r;                                          \
+  })
 
 void dbg(const char* fmt, ...) {
   if (!g.debug) return;
@@ -133,6 +159,8 @@ double env_f(const char* k, double d) {
 
 // Thread-local current device (updated by interposed hipSetDevice).
 thread_local int tls_device = 0;
+
+void record_latency(int cls, uint64_t t0_ns);  // fwd (uses now_ns)
 
 TfDeviceEntry* cur_dev() {
   if (!g.shm) return nullptr;
@@ -380,6 +408,7 @@ void ensure_tier_thread() {
 
 Limiter::Limiter() {
   debug = getenv("TF_LIMITER_DEBUG") != nullptr;
+  trace = getenv("TF_LIMITER_TRACE") != nullptr;
   expand_enabled = getenv("TF_VRAM_EXPAND") != nullptr &&
                    atoi(getenv("TF_VRAM_EXPAND")) != 0;
   expand_limit = (uint64_t)env_f("TF_VRAM_EXPAND_LIMIT_BYTES", 0);
@@ -613,7 +642,8 @@ hipError_t hipLaunchKernel(const void* f, dim3u grid, dim3u block, void** args,
   REAL(hipError_t, hipLaunchKernel, const void*, dim3u, dim3u, void**, size_t,
        hipStream_t);
   if (g.enabled) consume_blocking(g.tokens_per_launch);
-  return call_hipLaunchKernel(f, grid, block, args, shmem, stream);
+  return TF_TRACE_CALL(0, call_hipLaunchKernel(f, grid, block, args, shmem,
+                                               stream));
 }
 
 hipError_t hipLaunchKernel_spt(const void* f, dim3u grid, dim3u block,
@@ -690,7 +720,7 @@ hipError_t hipGraphLaunch_spt(void* graphExec, hipStream_t stream) {
 hipError_t hipMemcpy(void* dst, const void* src, size_t n, int kind) {
   REAL(hipError_t, hipMemcpy, void*, const void*, size_t, int);
   if (g.enabled) consume_blocking(g.tokens_per_memcpy);
-  return call_hipMemcpy(dst, src, n, kind);
+  return TF_TRACE_CALL(1, call_hipMemcpy(dst, src, n, kind));
 }
 
 hipError_t hipMemcpyAsync(void* dst, const void* src, size_t n, int kind,
@@ -698,7 +728,7 @@ hipError_t hipMemcpyAsync(void* dst, const void* src, size_t n, int kind,
   REAL(hipError_t, hipMemcpyAsync, void*, const void*, size_t, int,
        hipStream_t);
   if (g.enabled) consume_blocking(g.tokens_per_memcpy);
-  return call_hipMemcpyAsync(dst, src, n, kind, s);
+  return TF_TRACE_CALL(1, call_hipMemcpyAsync(dst, src, n, kind, s));
 }
 
 hipError_t hipMemcpyWithStream(void* dst, const void* src, size_t n, int kind,
@@ -769,6 +799,21 @@ unsigned long long tf_limiter_demote_all() { return tier_migrate(false, 0); }
 // Promote up to budget bytes into HBM; returns bytes moved.
 unsigned long long tf_limiter_promote(unsigned long long budget) {
   return tier_migrate(true, budget);
+}
+
+// Per-call latency histogram readout: cls 0 launch, 1 memcpy, 2 other;
+// out[32] = counts per log2(ns) bucket. Returns total samples.
+unsigned long long tf_limiter_latency_hist(int cls,
+                                           unsigned long long* out32) {
+  if (cls < 0 || cls > 2) return 0;
+  unsigned long long total = 0;
+  for (int b = 0; b < 32; ++b) {
+    unsigned long long v =
+        g.lat_hist[cls][b].load(std::memory_order_relaxed);
+    if (out32) out32[b] = v;
+    total += v;
+  }
+  return total;
 }
 
 void tf_limiter_set_vram_pressure(int on) {

@@ -6,6 +6,7 @@
 //   alloc <n> <bytes>   n hipMallocs of <bytes>; prints admitted/denied
 //   launch <n>          n hipLaunchKernels; prints elapsed seconds
 //   stats               prints tf_limiter_stats of device 0 (if preloaded)
+//   lat <n>             n launches with TF_LIMITER_TRACE; prints histogram totals
 
 #include <dlfcn.h>
 #include <stdio.h>
@@ -34,6 +35,17 @@ static double now_s() {
 int main(int argc, char** argv) {
   if (argc < 2) return 2;
   const char* sc = argv[1];
+  if (!strcmp(sc, "lat")) {
+    int n = argc > 2 ? atoi(argv[2]) : 100;
+    dim3u g{1, 1, 1}, b{64, 1, 1};
+    for (int i = 0; i < n; ++i)
+      hipLaunchKernel((void*)0x1, g, b, nullptr, 0, nullptr);
+    typedef unsigned long long (*hist_fn)(int, unsigned long long*);
+    hist_fn f = (hist_fn)dlsym(RTLD_DEFAULT, "tf_limiter_latency_hist");
+    unsigned long long total = f ? f(0, nullptr) : 0;
+    printf("{\"launch_samples\": %llu}\n", total);
+    return 0;
+  }
   if (!strcmp(sc, "alloc")) {
     int n = argc > 2 ? atoi(argv[2]) : 10;
     size_t bytes = argc > 3 ? strtoull(argv[3], nullptr, 10) : (1u << 20);

@@ -103,3 +103,10 @@ def test_freeze_blocks_and_resumes(native_built, tmp_path):
     r = json.loads(out.strip())
     assert r["launches"] == 10
     w.close()
+
+
+def test_latency_histogram_traces_launches(native_built):
+    r = run_testbed(native_built, "lat", 250,
+                    env={"TF_LIMITER_TRACE": "1",
+                         "TF_UP_LIMIT_PERCENT": "100"})
+    assert r["launch_samples"] == 250
