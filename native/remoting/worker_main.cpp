@@ -1393,8 +1393,11 @@ int main(int argc, char** argv) {
   const char* restore_path = getenv("TF_WORKER_RESTORE_PATH");
   bool want_vmm = !getenv("TF_WORKER_NO_VMM");
   if (want_vmm && !restore_path) {
-    // fresh start: VA-stable heap (restore re-reserves from the snapshot)
-    size_t heap = 192ull << 30;  // VA only; physical commits per mapping
+    // fresh start: VA-stable heap (restore re-reserves from the snapshot).
+    // VA only — physical memory commits per mapping, so size it well past
+    // HBM (fp32-init peaks of a 70B model exceed 192 GB before the bf16
+    // cast frees it).
+    size_t heap = 512ull << 30;
     const char* hb = getenv("TF_WORKER_VMM_BYTES");
     if (hb) heap = strtoull(hb, nullptr, 10);
     if (!W.vmm.init(0, VMM_BASE_HINT, heap))
