@@ -27,7 +27,7 @@ from ..api.types import (GPU, AllocRequest, GPUPartition, PartitionTemplate,
                          Resource)
 from ..quota.quota_store import QuotaExceeded, QuotaStore
 from . import partitioning
-from .filters import FilterRegistry, default_registry, make_same_node_filter
+from .filters import default_registry, make_same_node_filter
 from .strategy import Strategy, make_strategy
 
 

@@ -15,7 +15,7 @@ import threading
 import time
 from dataclasses import asdict, is_dataclass
 from queue import Empty, Queue
-from typing import Callable, Dict, Iterable, List, Optional, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 from .types import TFObject
 

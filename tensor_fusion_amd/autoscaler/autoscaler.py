@@ -14,7 +14,7 @@ import threading
 import time
 import urllib.request
 from dataclasses import dataclass
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 from .. import constants as C
 from ..api.store import Store

@@ -10,7 +10,7 @@ from __future__ import annotations
 import hashlib
 import json
 import time
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from .. import constants as C
 from ..api.store import NotFound, Store

@@ -13,7 +13,7 @@ import threading
 import time
 import traceback
 from dataclasses import dataclass
-from typing import Callable, Dict, List, Optional, Set, Tuple
+from typing import Dict, List, Optional, Tuple
 
 from ..api.store import NotFound, Store
 

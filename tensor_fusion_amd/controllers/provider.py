@@ -12,7 +12,7 @@ from __future__ import annotations
 from typing import Dict, Optional
 
 from ..api.store import Store
-from ..api.types import HardwareModel, PartitionTemplate, ProviderConfig
+from ..api.types import HardwareModel, ProviderConfig
 from .base import Reconciler, Request
 
 

@@ -9,7 +9,7 @@ from the hardware table and topology tiers), GPUNode status patches.
 from __future__ import annotations
 
 import threading
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 from ... import constants as C
 from ...api.store import NotFound, Store

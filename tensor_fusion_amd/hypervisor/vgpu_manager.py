@@ -15,7 +15,6 @@ import time
 from dataclasses import dataclass, field
 from typing import Dict, Optional
 
-from .. import constants as C
 from ..client.runtime import (WorkerHandle, migrate_worker, snapshot_and_stop,
                               start_worker)
 

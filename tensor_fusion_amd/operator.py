@@ -21,7 +21,7 @@ from .allocator.gpuallocator import GpuAllocator
 from .api.store import Store
 from .autoscaler import Autoscaler
 from .cloudprovider import MockProvider
-from .config import ConfigWatcher, GlobalConfig
+from .config import ConfigWatcher
 from .controllers import ControllerManager, default_controllers
 from .controllers.defrag import DefragController
 from .gang.manager import GangManager

@@ -25,8 +25,8 @@ import torch.distributed as dist
 import torch.nn as nn
 import torch.nn.functional as F
 
-from ..models.llama import (CONFIGS, Llama, LlamaConfig, RMSNorm, apply_rope,
-                            precompute_rope)
+from ..models.llama import (CONFIGS, Llama, LlamaConfig, RMSNorm,
+                            apply_rope)
 
 
 def _world():

@@ -9,7 +9,7 @@ annotation patch with rollback), PostBind, Unreserve.
 from __future__ import annotations
 
 import threading
-from typing import Dict, List, Optional, Tuple
+from typing import List, Optional, Tuple
 
 from .. import constants as C
 from ..allocator.gpuallocator import AllocationError, GpuAllocator

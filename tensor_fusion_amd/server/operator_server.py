@@ -21,7 +21,7 @@ from typing import Optional
 from fastapi import FastAPI, Header, Query
 from fastapi.responses import JSONResponse
 
-from ..api.store import NotFound, Store
+from ..api.store import Store
 
 
 def make_token(secret: str, namespace: str, pod: str) -> str:

@@ -10,7 +10,6 @@ mutator, and responds with an RFC-6902 patch.
 from __future__ import annotations
 
 import base64
-import copy
 import json
 from typing import Any, Dict, List
 

@@ -6,8 +6,7 @@ import re
 from typing import Optional
 
 from .. import constants as C
-from ..api.types import (AllocRequest, Pod, Requirements, Resource,
-                         WorkloadProfile)
+from ..api.types import AllocRequest, Pod, WorkloadProfile
 
 _QUANT_RE = re.compile(r"^\s*([0-9.]+)\s*([KMGTPE]i?)?\s*$")
 _SUFFIX = {

@@ -20,7 +20,7 @@ from __future__ import annotations
 from typing import Optional
 
 from .. import constants as C
-from ..api.store import AlreadyExists, NotFound, Store
+from ..api.store import AlreadyExists, Store
 from ..api.types import (Pod, TensorFusionWorkload, WorkloadProfile)
 from ..utils.resource import profile_from_annotations
 
