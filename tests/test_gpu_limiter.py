@@ -246,3 +246,40 @@ def test_multi_tenant_oversubscription():
         r = json.loads(stdout.strip().splitlines()[-1])
         assert r["ok"] and r["ok_after_demote"], (i, r)
         assert r["expanded"] > 0, (i, r)
+
+
+def test_hypervisor_shm_governs_real_workload(tmp_path):
+    """Full hypervisor⇄limiter integration on a real GPU: the hypervisor
+    creates the shm page, the limiter attaches via TF_SHM_PATH, runs
+    throttled at 25 %, the hypervisor observes usage + heartbeat, then
+    raises the rate live (the ERL control channel, reference §3.3)."""
+
+    from tensor_fusion_amd.hypervisor import shm as S
+
+    path = str(tmp_path / "shm")
+    page = S.WorkerShm.create(path)
+    page.set_device(0, "gpu-real-0", up_limit_percent=25,
+                    mem_limit_bytes=32 << 30, refill_rate=200.0,
+                    capacity=20.0)
+
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = LIMITER
+    env["TF_SHM_PATH"] = path
+    out = subprocess.run([sys.executable, "-c", CHILD, "matmul"], env=env,
+                         capture_output=True, text=True, timeout=300)
+    assert out.returncode == 0, out.stderr[-2000:]
+    slow = json.loads(out.stdout.strip().splitlines()[-1])
+    # 400 launches at 200 tok/s ≈ 2 s forced pacing
+    assert slow["elapsed_s"] > 1.5, slow
+    d = page.device(0)
+    assert d.launch_count >= 400
+    assert d.block_ns_total > 0
+    assert page.heartbeat() > 0
+
+    # live rate raise → unthrottled run through the same page
+    page.update_erl(0, refill_rate=1_000_000.0, capacity=100_000.0)
+    out2 = subprocess.run([sys.executable, "-c", CHILD, "matmul"], env=env,
+                          capture_output=True, text=True, timeout=300)
+    assert out2.returncode == 0, out2.stderr[-2000:]
+    fast = json.loads(out2.stdout.strip().splitlines()[-1])
+    assert fast["elapsed_s"] < 0.5 * slow["elapsed_s"], (slow, fast)

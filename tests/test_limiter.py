@@ -110,3 +110,28 @@ def test_latency_histogram_traces_launches(native_built):
                     env={"TF_LIMITER_TRACE": "1",
                          "TF_UP_LIMIT_PERCENT": "100"})
     assert r["launch_samples"] == 250
+
+
+def test_fake_amd_smi_reads_shm(tmp_path):
+    """tools/fake_amd_smi shows the vGPU-scoped view from the limiter shm
+    (reference: fake nvidia-smi bind-mounted into workload pods)."""
+
+    import json as _json
+    import subprocess as _sp
+    import sys as _sys
+
+    from tensor_fusion_amd.hypervisor import shm as S
+    path = str(tmp_path / "shm")
+    page = S.WorkerShm.create(path)
+    page.set_device(0, "vgpu-abc", up_limit_percent=25,
+                    mem_limit_bytes=8 << 30)
+    env = dict(os.environ)
+    env["TF_SHM_PATH"] = path
+    out = _sp.run([_sys.executable, "-m",
+                   "tensor_fusion_amd.tools.fake_amd_smi", "--json"],
+                  env=env, capture_output=True, text=True, timeout=60)
+    assert out.returncode == 0, out.stderr
+    data = _json.loads(out.stdout)
+    assert data["gpus"][0]["uuid"] == "vgpu-abc"
+    assert data["gpus"][0]["up_limit_percent"] == 25
+    assert data["gpus"][0]["mem_limit"] == 8 << 30
