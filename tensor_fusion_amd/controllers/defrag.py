@@ -44,13 +44,20 @@ class DefragController:
     def __init__(self, store: Store, allocator,
                  utilization_threshold: float = 0.3,
                  eviction_ttl_s: float = 60.0,
-                 campaign_cooldown_s: float = 300.0):
+                 campaign_cooldown_s: float = 300.0,
+                 migrate_fn=None):
         self.store = store
         self.allocator = allocator
         self.utilization_threshold = utilization_threshold
         self.eviction_ttl_s = eviction_ttl_s
         self.campaign_cooldown_s = campaign_cooldown_s
+        # migrate_fn(pod_key, target_gpus) → bool: live migration via the
+        # vGPU worker snapshot machinery (hypervisor/vgpu_manager). When
+        # set, defrag MOVES workers instead of evict+reschedule — the
+        # workload never restarts.
+        self.migrate_fn = migrate_fn
         self._last_campaign = 0.0
+        self.migrated: List[str] = []
 
     # ------------------------------------------------------ candidates
 
@@ -135,10 +142,45 @@ class DefragController:
         for k in range(len(cands), 0, -1):
             plan = self.simulate(cands[:k])
             if plan is not None:
-                self._mark_evictions(plan, now)
+                if self.migrate_fn is not None:
+                    self._execute_migrations(plan)
+                else:
+                    self._mark_evictions(plan, now)
                 self._last_campaign = now
                 return plan
         return None
+
+    def _execute_migrations(self, plan: DefragPlan):
+        """Live-migrate each planned worker to its simulated placement;
+        fall back to eviction marking for any that fail."""
+
+        import time as _time
+        failed = []
+        for pod_key in plan.evict_pods:
+            target = plan.placements.get(pod_key, [])
+            try:
+                ok = self.migrate_fn(pod_key, target)
+            except Exception:
+                ok = False
+            if ok:
+                self.migrated.append(pod_key)
+                # move the allocation record onto the new devices
+                alloc = self.allocator.allocation(pod_key)
+                if alloc is not None:
+                    self.allocator.dealloc(pod_key)
+                    try:
+                        self.allocator.assume(alloc.req, target)
+                        self.allocator.commit(pod_key)
+                        self.allocator.notify_bound(pod_key)
+                    except Exception:
+                        pass
+            else:
+                failed.append(pod_key)
+        if failed:
+            sub = DefragPlan(candidate_nodes=plan.candidate_nodes,
+                             evict_pods=failed,
+                             placements=plan.placements)
+            self._mark_evictions(sub, _time.time())
 
     def _mark_evictions(self, plan: DefragPlan, now: float):
         """Eviction markers with TTL (reference :686-897): pods are only

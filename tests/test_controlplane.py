@@ -439,3 +439,49 @@ class TestGreyRelease:
         pod2.meta.namespace = "d"
         pod2.containers = [Container(name="m", resources={"amd.com/gpu": "1"})]
         assert m100.should_handle(pod2) is True
+
+
+class TestDefragMigration:
+    def test_campaign_live_migrates_instead_of_evicting(self):
+        from tensor_fusion_amd.api.types import AllocRequest
+        from tensor_fusion_amd.controllers.defrag import AnnoEvictionMark
+        store = Store()
+        alloc = GpuAllocator(store=store)
+        for node in ("node-a", "node-b"):
+            for i in range(2):
+                store.create(mk_gpu(f"{node}-g{i}", node=node))
+        req_big = AllocRequest(pod_name="big", namespace="d",
+                               request=Resource(2000, 200 << 30, 80),
+                               limit=Resource(2000, 200 << 30, 80))
+        alloc.assume(req_big, ["node-a-g0"])
+        alloc.commit("d/big")
+        req_small = AllocRequest(pod_name="small", namespace="d",
+                                 request=Resource(100, 8 << 30, 5),
+                                 limit=Resource(100, 8 << 30, 5))
+        alloc.assume(req_small, ["node-b-g0"])
+        alloc.commit("d/small")
+        pod = Pod()
+        pod.meta.name = "small"
+        pod.meta.namespace = "d"
+        store.create(pod)
+
+        moves = []
+
+        def migrate(pod_key, target_gpus):
+            moves.append((pod_key, list(target_gpus)))
+            return True
+
+        d = DefragController(store, alloc, utilization_threshold=0.3,
+                             campaign_cooldown_s=0.0, migrate_fn=migrate)
+        plan = d.run_campaign()
+        assert plan is not None
+        assert moves and moves[0][0] == "d/small"
+        # allocation moved to the target node, no eviction mark
+        new_alloc = alloc.allocation("d/small")
+        assert new_alloc is not None
+        assert all(g.startswith("node-a") for g in new_alloc.gpu_names)
+        assert AnnoEvictionMark not in store.get("Pod", "small",
+                                                 "d").meta.annotations
+        # node-b is now empty
+        for g in alloc.gpus(node="node-b"):
+            assert g.status.available.vram == C.MI355X_VRAM_BYTES
