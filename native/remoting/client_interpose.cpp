@@ -25,6 +25,9 @@
 #include <stdlib.h>
 #include <string.h>
 #include <sys/mman.h>
+#include <arpa/inet.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
 #include <sys/socket.h>
 #include <sys/stat.h>
 #include <sys/un.h>
@@ -74,6 +77,7 @@ struct FatBin {
 struct Client {
   bool connected = false;
   bool debug = false;
+  bool tcp_mode = false;    // cross-node session (TF_WORKER_TCP=host:port)
   int sock = -1;
   int seg_fd = -1;          // memfd of the segment, kept for re-handshake
   char sock_path[256] = {0};
@@ -155,6 +159,106 @@ bool send_fd(int sock, int fd) {
   return sendmsg(sock, &msg, 0) == 1;
 }
 
+// ------------------------------------------------ TCP transport pumps
+// Cross-node mode mirrors the local rings over a socket: the cmd pump
+// consumes the cmd ring (as the worker would) and frames records +
+// referenced arena payloads onto the wire; the cpl pump turns incoming
+// frames back into cpl-ring records.
+
+bool tcp_read_full(int fd, void* p, size_t n) {
+  uint8_t* b = (uint8_t*)p;
+  while (n) {
+    ssize_t r = recv(fd, b, n, 0);
+    if (r <= 0) return false;
+    b += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+bool tcp_write_full(int fd, const void* p, size_t n) {
+  const uint8_t* b = (const uint8_t*)p;
+  while (n) {
+    ssize_t r = send(fd, b, n, MSG_NOSIGNAL);
+    if (r <= 0) return false;
+    b += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+void* tcp_cmd_pump(void* arg) {
+  Client* c = (Client*)arg;
+  RingView cmd(&c->hdr->cmd, cmd_buf(c->hdr), CMD_RING_BYTES);
+  uint8_t* arena = c->arena_base;
+  for (;;) {
+    size_t len;
+    uint8_t* p = cmd.try_next(&len);
+    if (!p) {
+      cmd.wait_nonempty();
+      continue;
+    }
+    auto* h = reinterpret_cast<CmdHdr*>(p);
+    uint64_t arena_off = 0;
+    uint32_t extra = 0;
+    if (h->op == OP_MEMCPY_H2D && !(h->flags & F_INLINE_DATA)) {
+      auto* m = reinterpret_cast<MemcpyBody*>(p + sizeof(CmdHdr));
+      arena_off = m->arena_off;
+      extra = (uint32_t)m->size;
+    } else if (h->op == OP_LOAD_MODULE) {
+      struct B {
+        uint64_t image_id, size, arena_off;
+      };
+      auto* b = reinterpret_cast<B*>(p + sizeof(CmdHdr));
+      arena_off = b->arena_off;
+      extra = (uint32_t)b->size;
+    }
+    FrameHdr fh{0, (uint32_t)len, arena_off, extra, 0};
+    bool ok = tcp_write_full(c->sock, &fh, sizeof fh) &&
+              tcp_write_full(c->sock, p, len);
+    if (ok && extra) {
+      size_t off = arena_off % ARENA_BYTES;
+      size_t first = ARENA_BYTES - off;
+      if (extra <= first) {
+        ok = tcp_write_full(c->sock, arena + off, extra);
+      } else {
+        ok = tcp_write_full(c->sock, arena + off, first) &&
+             tcp_write_full(c->sock, arena, extra - first);
+      }
+      // shipped: the local staging chunk is reusable immediately
+      uint64_t end = arena_off + ((extra + 63) & ~uint64_t(63));
+      uint64_t cur = at(&c->hdr->arena_freed)->load();
+      if (end > cur) at(&c->hdr->arena_freed)->store(end);
+    }
+    cmd.pop();
+    if (!ok) {
+      fprintf(stderr, "[tf-client] tcp send failed\n");
+      return nullptr;
+    }
+  }
+}
+
+void* tcp_cpl_pump(void* arg) {
+  Client* c = (Client*)arg;
+  RingView cpl(&c->hdr->cpl, cpl_buf(c->hdr), CPL_RING_BYTES);
+  std::vector<uint8_t> rec;
+  for (;;) {
+    FrameHdr fh;
+    if (!tcp_read_full(c->sock, &fh, sizeof fh)) break;
+    if (fh.kind != 1 || fh.rec_len > CPL_RING_BYTES / 2) break;
+    rec.resize(fh.rec_len);
+    if (!tcp_read_full(c->sock, rec.data(), fh.rec_len)) break;
+    uint8_t* p;
+    while (!(p = cpl.try_reserve(fh.rec_len))) usleep(50);
+    memcpy(p, rec.data(), fh.rec_len);
+    cpl.commit();
+    if (at(&c->hdr->futex_cpl)->exchange(1) == 0)
+      futex_wake(&c->hdr->futex_cpl);
+  }
+  fprintf(stderr, "[tf-client] tcp session closed\n");
+  return nullptr;
+}
+
 void* reconnect_main(void* arg) {
   Client* c = reinterpret_cast<Client*>(arg);
   for (;;) {
@@ -186,7 +290,65 @@ void* reconnect_main(void* arg) {
 
 Client::Client() {
   debug = getenv("TF_CLIENT_DEBUG") != nullptr;
+  const char* tcp = getenv("TF_WORKER_TCP");  // host:port → cross-node
   const char* sock_path = getenv("TF_WORKER_SOCKET");
+  if (tcp && *tcp) {
+    char host[128] = {0};
+    int port = 0;
+    const char* colon = strrchr(tcp, ':');
+    if (!colon || sscanf(colon + 1, "%d", &port) != 1) return;
+    size_t hl = (size_t)(colon - tcp);
+    if (hl >= sizeof host) return;
+    memcpy(host, tcp, hl);
+    void* seg = mmap(nullptr, SEG_BYTES, PROT_READ | PROT_WRITE,
+                     MAP_PRIVATE | MAP_ANONYMOUS, -1, 0);
+    if (seg == MAP_FAILED) return;
+    memset(seg, 0, HDR_BYTES);
+    hdr = reinterpret_cast<Header*>(seg);
+    hdr->magic = MAGIC;
+    hdr->version = VERSION;
+    hdr->total_bytes = SEG_BYTES;
+    sock = socket(AF_INET, SOCK_STREAM, 0);
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_port = htons((uint16_t)port);
+    if (inet_pton(AF_INET, host, &a.sin_addr) != 1) return;
+    bool ok = false;
+    for (int i = 0; i < 100; ++i) {
+      if (connect(sock, (sockaddr*)&a, sizeof a) == 0) {
+        ok = true;
+        break;
+      }
+      usleep(100000);
+    }
+    if (!ok) {
+      fprintf(stderr, "[tf-client] cannot reach worker at %s\n", tcp);
+      return;
+    }
+    int nd = 1;
+    setsockopt(sock, IPPROTO_TCP, TCP_NODELAY, &nd, sizeof nd);
+    uint32_t magic = TCP_MAGIC;
+    if (!tcp_write_full(sock, &magic, 4) ||
+        !tcp_read_full(sock, &magic, 4) || magic != TCP_MAGIC) {
+      fprintf(stderr, "[tf-client] tcp handshake failed\n");
+      return;
+    }
+    cmd = RingView(&hdr->cmd, cmd_buf(hdr), CMD_RING_BYTES);
+    cpl = RingView(&hdr->cpl, cpl_buf(hdr), CPL_RING_BYTES);
+    arena_base = arena(hdr);
+    at(&hdr->worker_ready)->store(1, std::memory_order_release);
+    tcp_mode = true;
+    connected = true;
+    pthread_t t1, t2;
+    pthread_create(&t1, nullptr, tcp_cmd_pump, this);
+    pthread_detach(t1);
+    pthread_create(&t2, nullptr, tcp_cpl_pump, this);
+    pthread_detach(t2);
+    if (debug)
+      fprintf(stderr, "[tf-client %d] connected via tcp %s\n", getpid(),
+              tcp);
+    return;
+  }
   if (!sock_path || !*sock_path) return;
 
   int fd = memfd_create("tfrpc-seg", 0);
@@ -773,6 +935,27 @@ static hipError_t do_memcpy(void* dst, const void* src, size_t n, int kind,
       return hipSuccess;
     }
     case 2: {  // D2H: inherently needs the data — always a round trip
+      if (c.tcp_mode) {
+        // cross-node: data rides inside reply bodies; chunk to fit the
+        // completion ring
+        size_t done = 0;
+        while (done < n) {
+          size_t chunk = n - done;
+          if (chunk > CPL_RING_BYTES / 4) chunk = CPL_RING_BYTES / 4;
+          MemcpyBody b{};
+          b.src = (uint64_t)src + done;
+          b.size = chunk;
+          b.stream = (uint64_t)stream;
+          b.kind = 2;
+          size_t got = 0;
+          hipError_t e = send_sync(OP_MEMCPY_D2H, &b, sizeof b,
+                                   (uint8_t*)dst + done, chunk, &got);
+          if (e != hipSuccess) return e;
+          if (got != chunk) return hipErrorInvalidValue;
+          done += chunk;
+        }
+        return hipSuccess;
+      }
       size_t done = 0;
       while (done < n) {
         size_t chunk = n - done;

@@ -175,4 +175,24 @@ struct MemcpyBody {
   uint32_t sync;       // 1 = synchronous semantics requested
 };
 
+// ---- TCP transport framing (cross-node GPU-over-IP) -------------------
+// Same command stream over a socket: each side runs a pump that mirrors
+// its ring to the wire. Frames:
+//   [FrameHdr][record bytes][extra bytes]
+// kind 0 = cmd record (CmdHdr+body). extra = arena payload the command
+//          references (non-inline H2D staging, module images), written
+//          into the receiver's arena at `arena_off % ARENA_BYTES`.
+// kind 1 = cpl record (CplHdr+body). D2H data rides INSIDE the reply body
+//          in TCP sessions (the worker detects the session type), so
+//          extra_len is 0 for completions.
+struct FrameHdr {
+  uint32_t kind;       // 0 cmd, 1 cpl
+  uint32_t rec_len;    // bytes of the record that follows
+  uint64_t arena_off;  // where extra bytes land (kind 0 only)
+  uint32_t extra_len;
+  uint32_t _pad;
+};
+
+constexpr uint32_t TCP_MAGIC = 0x54465443;  // "TFTC" handshake word
+
 }  // namespace tfrpc

@@ -23,6 +23,8 @@
 #include <stdlib.h>
 #include <string.h>
 #include <sys/mman.h>
+#include <netinet/in.h>
+#include <netinet/tcp.h>
 #include <sys/socket.h>
 #include <sys/un.h>
 #include <unistd.h>
@@ -444,6 +446,7 @@ struct Staging {
 struct Worker {
   tfrpc::Header* hdr = nullptr;
   int cli_fd = -1;  // client socket: EOF ⇒ client exited
+  bool tcp_mode = false;  // cross-node session: D2H data rides in replies
   tfrpc::RingView cmd;  // consumer
   tfrpc::RingView cpl;  // producer
   uint8_t* arena = nullptr;
@@ -631,6 +634,17 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
     case OP_MEMCPY_D2H: {
       auto* m = reinterpret_cast<MemcpyBody*>(body);
       m->stream = xl(m->stream);
+      if (W.tcp_mode) {
+        // cross-node: data travels inside the reply body (the client has
+        // no shared arena); client chunks large reads to fit the ring
+        static std::vector<uint8_t> tmp;
+        tmp.resize(m->size);
+        hipError_t e = hip.MemcpyAsync(tmp.data(), (const void*)m->src,
+                                       m->size, 2, (hipStream_t)m->stream);
+        if (e == 0) e = hip.StreamSynchronize((hipStream_t)m->stream);
+        reply(c->seq, e, tmp.data(), e == 0 ? (uint32_t)m->size : 0);
+        break;
+      }
       void* dst = W.arena + (m->arena_off % ARENA_BYTES);
       hipError_t e = hip.MemcpyAsync(dst, (const void*)m->src, m->size,
                                      2 /*D2H*/, (hipStream_t)m->stream);
@@ -1352,6 +1366,140 @@ int serve(tfrpc::Header* hdr) {
   return 0;
 }
 
+// ---------------------------------------------------- TCP transport
+// Cross-node session: the worker owns a PRIVATE segment; a reader thread
+// writes incoming cmd frames into its cmd ring (+ arena payloads), and a
+// writer thread drains the cpl ring back over the socket. serve() and
+// every handler run unchanged.
+
+bool read_full(int fd, void* p, size_t n) {
+  uint8_t* b = (uint8_t*)p;
+  while (n) {
+    ssize_t r = recv(fd, b, n, 0);
+    if (r <= 0) return false;
+    b += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+bool write_full(int fd, const void* p, size_t n) {
+  const uint8_t* b = (const uint8_t*)p;
+  while (n) {
+    ssize_t r = send(fd, b, n, MSG_NOSIGNAL);
+    if (r <= 0) return false;
+    b += r;
+    n -= (size_t)r;
+  }
+  return true;
+}
+
+struct TcpSession {
+  int fd;
+  tfrpc::Header* hdr;
+  std::atomic<bool> dead{false};
+};
+
+void* tcp_reader_main(void* arg) {
+  auto* s = (TcpSession*)arg;
+  tfrpc::RingView cmd(&s->hdr->cmd, tfrpc::cmd_buf(s->hdr),
+                      tfrpc::CMD_RING_BYTES);
+  uint8_t* arena = tfrpc::arena(s->hdr);
+  std::vector<uint8_t> rec;
+  for (;;) {
+    tfrpc::FrameHdr fh;
+    if (!read_full(s->fd, &fh, sizeof fh)) break;
+    if (fh.kind != 0 || fh.rec_len > tfrpc::CMD_RING_BYTES / 2) break;
+    rec.resize(fh.rec_len);
+    if (!read_full(s->fd, rec.data(), fh.rec_len)) break;
+    if (fh.extra_len) {
+      size_t off = fh.arena_off % tfrpc::ARENA_BYTES;
+      size_t first = tfrpc::ARENA_BYTES - off;
+      if (fh.extra_len <= first) {
+        if (!read_full(s->fd, arena + off, fh.extra_len)) break;
+      } else {
+        if (!read_full(s->fd, arena + off, first)) break;
+        if (!read_full(s->fd, arena, fh.extra_len - first)) break;
+      }
+    }
+    uint8_t* p;
+    while (!(p = cmd.try_reserve(fh.rec_len))) usleep(50);
+    memcpy(p, rec.data(), fh.rec_len);
+    cmd.commit();
+    cmd.wake_consumer();
+  }
+  s->dead.store(true);
+  // unblock serve(): mark shutdown so the loop exits
+  s->hdr->shutdown = 1;
+  tfrpc::at(&s->hdr->cmd.futex_nonempty)->store(1);
+  tfrpc::futex_wake(&s->hdr->cmd.futex_nonempty);
+  return nullptr;
+}
+
+void* tcp_writer_main(void* arg) {
+  auto* s = (TcpSession*)arg;
+  tfrpc::RingView cpl(&s->hdr->cpl, tfrpc::cpl_buf(s->hdr),
+                      tfrpc::CPL_RING_BYTES);
+  while (!s->dead.load()) {
+    size_t len;
+    uint8_t* p = cpl.try_next(&len);
+    if (!p) {
+      // park on the cpl futex the worker's reply() wakes
+      tfrpc::at(&s->hdr->futex_cpl)->exchange(0);
+      p = cpl.try_next(&len);
+      if (!p) {
+        tfrpc::futex_wait(&s->hdr->futex_cpl, 0, 100);
+        continue;
+      }
+    }
+    tfrpc::FrameHdr fh{1, (uint32_t)len, 0, 0, 0};
+    if (!write_full(s->fd, &fh, sizeof fh) || !write_full(s->fd, p, len))
+      break;
+    cpl.pop();
+  }
+  s->dead.store(true);
+  return nullptr;
+}
+
+int serve_tcp(int cli) {
+  uint32_t magic = 0;
+  if (!read_full(cli, &magic, 4) || magic != tfrpc::TCP_MAGIC) {
+    close(cli);
+    return -1;
+  }
+  write_full(cli, &magic, 4);
+  void* seg = mmap(nullptr, tfrpc::SEG_BYTES, PROT_READ | PROT_WRITE,
+                   MAP_PRIVATE | MAP_ANONYMOUS, -1, 0);
+  if (seg == MAP_FAILED) {
+    close(cli);
+    return -1;
+  }
+  auto* hdr = reinterpret_cast<tfrpc::Header*>(seg);
+  memset(hdr, 0, tfrpc::HDR_BYTES);
+  hdr->magic = tfrpc::MAGIC;
+  hdr->version = tfrpc::VERSION;
+  hdr->total_bytes = tfrpc::SEG_BYTES;
+  TcpSession sess{cli, hdr};
+  pthread_t rt, wt;
+  pthread_create(&rt, nullptr, tcp_reader_main, &sess);
+  pthread_create(&wt, nullptr, tcp_writer_main, &sess);
+  W.tcp_mode = true;
+  W.cli_fd = -1;  // EOF detection is the reader's job in TCP mode
+  fprintf(stderr, "[worker] tcp session started\n");
+  serve(hdr);
+  sess.dead.store(true);
+  shutdown(cli, SHUT_RDWR);
+  pthread_join(rt, nullptr);
+  // wake a parked writer
+  tfrpc::at(&hdr->futex_cpl)->store(1);
+  tfrpc::futex_wake(&hdr->futex_cpl);
+  pthread_join(wt, nullptr);
+  close(cli);
+  munmap(seg, tfrpc::SEG_BYTES);
+  W = Worker{};
+  return 0;
+}
+
 int recv_fd(int sock) {
   char buf[1];
   iovec iov{buf, 1};
@@ -1373,8 +1521,9 @@ int recv_fd(int sock) {
 
 int main(int argc, char** argv) {
   const char* sock_path = argc > 1 ? argv[1] : getenv("TF_WORKER_SOCKET");
-  if (!sock_path) {
-    fprintf(stderr, "usage: tf_vgpu_worker <socket-path>\n");
+  if (!sock_path && !getenv("TF_WORKER_TCP_PORT")) {
+    fprintf(stderr,
+            "usage: tf_vgpu_worker <socket-path>  (or TF_WORKER_TCP_PORT=N)\n");
     return 2;
   }
   if (!hip.load()) {
@@ -1403,6 +1552,34 @@ int main(int argc, char** argv) {
     if (!W.vmm.init(0, VMM_BASE_HINT, heap))
       fprintf(stderr, "[worker] VMM unavailable — plain hipMalloc, "
                       "snapshot/migration disabled\n");
+  }
+
+  const char* tcp_port = getenv("TF_WORKER_TCP_PORT");
+  if (tcp_port && *tcp_port) {
+    // cross-node mode: accept GPU-over-IP sessions on TCP
+    int srv = socket(AF_INET, SOCK_STREAM, 0);
+    int one = 1;
+    setsockopt(srv, SOL_SOCKET, SO_REUSEADDR, &one, sizeof one);
+    sockaddr_in a{};
+    a.sin_family = AF_INET;
+    a.sin_addr.s_addr = INADDR_ANY;
+    a.sin_port = htons((uint16_t)atoi(tcp_port));
+    if (bind(srv, (sockaddr*)&a, sizeof a) != 0 || listen(srv, 4) != 0) {
+      fprintf(stderr, "[worker] tcp bind :%s: %s\n", tcp_port,
+              strerror(errno));
+      return 2;
+    }
+    fprintf(stderr, "[worker] listening on tcp :%s (%d devices)\n",
+            tcp_port, n);
+    for (;;) {
+      int cli = accept(srv, nullptr, nullptr);
+      if (cli < 0) continue;
+      int nd = 1;
+      setsockopt(cli, IPPROTO_TCP, TCP_NODELAY, &nd, sizeof nd);
+      serve_tcp(cli);
+      if (getenv("TF_WORKER_ONESHOT")) break;
+    }
+    return 0;
   }
 
   unlink(sock_path);

@@ -57,8 +57,10 @@ class WorkerHandle:
 def start_worker(socket_path: str, device_index: int = 0,
                  env: Optional[Dict[str, str]] = None,
                  oneshot: bool = False, wait_s: float = 30.0,
-                 snapshot_path: Optional[str] = None) -> WorkerHandle:
-    """Spawn tf_vgpu_worker bound to one GPU, wait for its socket."""
+                 snapshot_path: Optional[str] = None,
+                 tcp_port: Optional[int] = None) -> WorkerHandle:
+    """Spawn tf_vgpu_worker bound to one GPU; waits for its unix socket,
+    or for the TCP port when tcp_port is given (cross-node mode)."""
 
     exe = os.path.join(_NATIVE, "tf_vgpu_worker")
     if not os.path.exists(exe):
@@ -70,30 +72,49 @@ def start_worker(socket_path: str, device_index: int = 0,
         e["TF_WORKER_SNAPSHOT_PATH"] = snapshot_path
     if oneshot:
         e["TF_WORKER_ONESHOT"] = "1"
+    if tcp_port is not None:
+        e["TF_WORKER_TCP_PORT"] = str(tcp_port)
     e.update(env or {})
-    os.makedirs(os.path.dirname(socket_path) or ".", exist_ok=True)
-    proc = subprocess.Popen([exe, socket_path], env=e)
+    args = [exe] + ([socket_path] if socket_path else [])
+    if socket_path:
+        os.makedirs(os.path.dirname(socket_path) or ".", exist_ok=True)
+    proc = subprocess.Popen(args, env=e)
     deadline = time.time() + wait_s
     while time.time() < deadline:
-        if os.path.exists(socket_path):
+        if tcp_port is not None:
+            import socket as _socket
+            try:
+                with _socket.create_connection(("127.0.0.1", tcp_port),
+                                               timeout=0.2):
+                    return WorkerHandle(proc=proc,
+                                        socket_path=socket_path or "")
+            except OSError:
+                pass
+        elif os.path.exists(socket_path):
             return WorkerHandle(proc=proc, socket_path=socket_path)
         if proc.poll() is not None:
             raise RuntimeError(
                 f"tf_vgpu_worker exited rc={proc.returncode} before listening")
         time.sleep(0.05)
     proc.kill()
-    raise TimeoutError(f"worker socket {socket_path} never appeared")
+    raise TimeoutError(f"worker {socket_path or tcp_port} never came up")
 
 
 def client_env(socket_path: str, base: Optional[Dict[str, str]] = None,
-               debug: bool = False) -> Dict[str, str]:
-    """Environment for a GPU-less client process driving a remote vGPU."""
+               debug: bool = False,
+               tcp: Optional[str] = None) -> Dict[str, str]:
+    """Environment for a GPU-less client process driving a remote vGPU.
+    `tcp="host:port"` selects the cross-node TCP transport."""
 
     e = dict(base if base is not None else os.environ)
     client_lib = os.path.join(_NATIVE, C.ClientLibName)
     prev = e.get("LD_PRELOAD", "")
     e["LD_PRELOAD"] = client_lib + (" " + prev if prev else "")
-    e["TF_WORKER_SOCKET"] = socket_path
+    if tcp:
+        e["TF_WORKER_TCP"] = tcp
+        e.pop("TF_WORKER_SOCKET", None)
+    else:
+        e["TF_WORKER_SOCKET"] = socket_path
     # the client process must NOT see a GPU: remoting is the only path
     e["HIP_VISIBLE_DEVICES"] = ""
     e["ROCR_VISIBLE_DEVICES"] = ""

@@ -159,3 +159,30 @@ def test_resnet50_inference_remoted(worker):
     assert out.returncode == 0, out.stdout[-1500:] + out.stderr[-4000:]
     r = json.loads(out.stdout.strip().splitlines()[-1])
     assert r["img_s"] > 0
+
+
+def test_tcp_transport_cross_node_shape(tmp_path):
+    """GPU-over-IP over actual IP: the TCP transport carries the same
+    command stream (the reference's cross-node mode, README 'Ethernet/
+    InfiniBand'). Same-host loopback here; the wire format is the
+    cross-node one."""
+
+    from tensor_fusion_amd.client.runtime import client_env, start_worker
+    h = start_worker("", device_index=0, tcp_port=47821)
+    try:
+        env = client_env("", tcp="127.0.0.1:47821")
+        out = subprocess.run([os.path.join(NATIVE, "tf_remote_testapp")],
+                             capture_output=True, text=True, timeout=300,
+                             env=env)
+        assert "TESTAPP_OK" in out.stdout, out.stdout + out.stderr
+        # full PyTorch over TCP
+        out2 = subprocess.run(
+            [sys.executable, "-m", "tensor_fusion_amd.models.llama",
+             "--model", "tiny", "--batch", "2", "--ctx", "16", "--steps",
+             "4", "--warmup", "1"],
+            capture_output=True, text=True, timeout=600, env=env, cwd=REPO)
+        assert out2.returncode == 0, out2.stdout[-1500:] + out2.stderr[-3000:]
+        r = json.loads(out2.stdout.strip().splitlines()[-1])
+        assert r["tok_s"] > 0
+    finally:
+        h.stop()
