@@ -485,3 +485,49 @@ class TestDefragMigration:
         # node-b is now empty
         for g in alloc.gpus(node="node-b"):
             assert g.status.available.vram == C.MI355X_VRAM_BYTES
+
+
+class TestConnectionFailover:
+    def test_worker_loss_reselects_and_updates_url(self):
+        store = Store()
+        mgr = ControllerManager(store)
+        for ctrl in default_controllers(store):
+            mgr.register(ctrl)
+        from tensor_fusion_amd.api.types import (TensorFusionConnection,
+                                                 TensorFusionWorkload)
+        wl = TensorFusionWorkload()
+        wl.meta.name = "wl1"
+        wl.meta.namespace = "d"
+        wl.replicas = 2
+        store.create(wl)
+        mgr.reconcile_now()
+        workers = sorted(
+            (p for p in store.list("Pod", namespace="d")
+             if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker),
+            key=lambda p: p.meta.name)
+        assert len(workers) == 2
+        for i, w in enumerate(workers):
+            def _s(obj, ip=f"10.0.0.{i+1}"):
+                obj.status.phase = "Running"
+                obj.status.pod_ip = ip
+            store.patch("Pod", w.meta.name, "d", _s)
+        conn = TensorFusionConnection()
+        conn.meta.name = "c1"
+        conn.meta.namespace = "d"
+        conn.workload = "wl1"
+        store.create(conn)
+        mgr.reconcile_now()
+        conn = store.get("TensorFusionConnection", "c1", "d")
+        first_worker = conn.status.worker
+        assert first_worker
+
+        # the selected worker dies → controller must fail over
+        def _fail(obj):
+            obj.status.phase = "Failed"
+        store.patch("Pod", first_worker, "d", _fail)
+        mgr.reconcile_now()
+        conn = store.get("TensorFusionConnection", "c1", "d")
+        assert conn.status.worker != first_worker
+        assert conn.status.worker
+        ip = conn.status.connection_url.split("+")[1]
+        assert ip.startswith("10.0.0.")
