@@ -56,6 +56,16 @@ def run_child(mode: str, args, local_rank: int) -> dict:
             worker = start_worker(
                 sock, device_index=int(env["HIP_VISIBLE_DEVICES"] or "0"))
             env = client_env(sock, base=env)
+        elif vgpu_mode == "tcp":
+            # cross-node wire format over loopback TCP
+            sys.path.insert(0, REPO)
+            from tensor_fusion_amd.client.runtime import (client_env,
+                                                          start_worker)
+            port = 47900 + local_rank
+            worker = start_worker(
+                "", device_index=int(env["HIP_VISIBLE_DEVICES"] or "0"),
+                tcp_port=port)
+            env = client_env("", base=env, tcp=f"127.0.0.1:{port}")
     cmd = [sys.executable, "-m", "tensor_fusion_amd.models.llama",
            "--model", args.model, "--batch", str(args.batch),
            "--ctx", str(args.ctx), "--steps", str(args.steps),
