@@ -283,3 +283,71 @@ def test_hypervisor_shm_governs_real_workload(tmp_path):
     assert out2.returncode == 0, out2.stderr[-2000:]
     fast = json.loads(out2.stdout.strip().splitlines()[-1])
     assert fast["elapsed_s"] < 0.5 * slow["elapsed_s"], (slow, fast)
+
+
+def test_erl_pid_converges_to_setpoint(tmp_path):
+    """The full ERL control loop on real hardware: hypervisor PID measures
+    gfx activity via amd-smi and modulates the token rate in shm until the
+    throttled workload sits at its 25 % setpoint (reference §3.3 — the
+    500 ms quota_controller loop)."""
+
+    import threading
+    import time as _time
+
+    from tensor_fusion_amd.hypervisor import shm as S
+    from tensor_fusion_amd.hypervisor.device import (Accelerator,
+                                                     DeviceController)
+    from tensor_fusion_amd.hypervisor.erl import ErlQuotaController
+
+    stop_file = str(tmp_path / "stop")
+    CHILD_LOOP = f"""
+import os, time, torch
+a = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+b = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+while not os.path.exists({stop_file!r}):
+    for _ in range(8):
+        a @ b
+    torch.cuda.synchronize()
+print("CHILD_DONE", flush=True)
+"""
+    path = str(tmp_path / "shm")
+    page = S.WorkerShm.create(path)
+    page.set_device(0, "gpu0", up_limit_percent=25,
+                    mem_limit_bytes=64 << 30, refill_rate=5000.0,
+                    capacity=1000.0)
+
+    accel = Accelerator()
+    devices = DeviceController(accel)
+    erl = ErlQuotaController(devices)
+    erl.attach(page)
+
+    env = dict(os.environ)
+    env["LD_PRELOAD"] = LIMITER
+    env["TF_SHM_PATH"] = path
+    child = subprocess.Popen([sys.executable, "-c", CHILD_LOOP], env=env,
+                             stdout=subprocess.PIPE, stderr=subprocess.PIPE,
+                             text=True)
+    try:
+        _time.sleep(3.0)  # warmup: unthrottled-ish burst
+        samples = []
+        for tick in range(26):
+            erl.tick(dt=0.5)
+            m = devices.metrics(0)
+            samples.append(m.gfx_activity)
+            _time.sleep(0.5)
+        tail = samples[-8:]
+        avg = sum(tail) / len(tail)
+        d = page.device(0)
+        assert d.block_ns_total > 0, "workload was never throttled"
+        # PID should hold activity near the 25% setpoint (wide band: the
+        # activity counter is coarse and the plant gain is kernel-shaped)
+        assert 5.0 <= avg <= 60.0, (samples, d)
+        # and meaningfully below the unthrottled ~100%
+        assert avg < 80.0, samples
+    finally:
+        with open(stop_file, "w") as f:
+            f.write("x")
+        try:
+            child.wait(timeout=60)
+        except subprocess.TimeoutExpired:
+            child.kill()
