@@ -180,11 +180,38 @@ class Llama(nn.Module):
         if not hasattr(self, "_cos") or self._cos.device != device:
             self._cos, self._sin = precompute_rope(self.cfg, device)
         x = self.embed(tokens)
+        if (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
+                and x.shape[-1] % 8 == 0):
+            return self._forward_fused(x, pos, caches, pos_end, mask)
         for i, blk in enumerate(self.blocks):
             x = blk(x, self._cos, self._sin, pos,
                     caches[i] if caches is not None else None, pos_end,
                     mask=mask)
         return self.lm_head(self.norm(x))
+
+    def _forward_fused(self, x, pos, caches, pos_end, mask):
+        """Residual-carry layout: the residual stream lives in one buffer
+        and every add+RMSNorm pair is ONE gfx950 kernel (ops/fused.py) —
+        the vLLM-style fusion of the decode hot path."""
+
+        from ..ops import fused
+        res = x.contiguous()
+        normed = fused.rmsnorm(res, self.blocks[0].ln1.weight,
+                               self.blocks[0].ln1.eps)
+        n = len(self.blocks)
+        for i, blk in enumerate(self.blocks):
+            a = blk.attn(normed, self._cos, self._sin, pos,
+                         caches[i] if caches is not None else None,
+                         pos_end, mask=mask)
+            normed = fused.add_rmsnorm(a.contiguous(), res, blk.ln2.weight,
+                                       blk.ln2.eps)
+            m = blk.mlp(normed)
+            nxt_w = (self.blocks[i + 1].ln1.weight if i + 1 < n
+                     else self.norm.weight)
+            nxt_eps = (self.blocks[i + 1].ln1.eps if i + 1 < n
+                       else self.norm.eps)
+            normed = fused.add_rmsnorm(m.contiguous(), res, nxt_w, nxt_eps)
+        return self.lm_head(normed)
 
     def make_kv_cache(self, batch: int, max_seq: int, device, dtype):
         cfg = self.cfg
