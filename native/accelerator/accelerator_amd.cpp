@@ -303,10 +303,10 @@ int tf_accel_get_processes(int device, TfAccelProc* out, int max_procs,
 // ---- isolation helpers -----------------------------------------------
 
 static int mask_env_from_ranges(const char* ranges, char* out_env, int out_len) {
-  // ROCr consumes HSA_CU_MASK as a per-queue hex bitmask list; the simpler
-  // documented form "start-end" CU id ranges is what our limiter/worker
-  // parses; both are emitted: HSA_CU_MASK for ROCr, TF_CU_RANGES for ours.
-  int w = snprintf(out_env, out_len, "TF_CU_RANGES=%s", ranges);
+  // ROCr syntax: HSA_CU_MASK=<queue-list>:<cu-list>. Cover queues 0-15 so
+  // every compute queue a workload creates is confined (validated on
+  // MI355X: tests/test_gpu_accelerator.py partition test).
+  int w = snprintf(out_env, out_len, "HSA_CU_MASK=0-15:%s", ranges);
   return (w > 0 && w < out_len) ? TF_ACCEL_OK : TF_ACCEL_ERR;
 }
 

@@ -81,13 +81,13 @@ class AllocationController:
         if spec.isolation == C.IsolationHard:
             pct = spec.compute_percent_limit or up_limit
             mask, _ = cu_mask_for_percent(max(pct, 0.5))
-            # ROCr applies HSA_CU_MASK per queue: "<queue>:<ranges>"; queue 0
-            # covers the default compute queues the workload creates.
-            env[C.EnvCuMask] = f"0:{mask}"
+            # ROCr syntax "<queue-list>:<cu-list>"; queues 0-15 cover every
+            # compute queue the workload creates.
+            env[C.EnvCuMask] = f"0-15:{mask}"
             env["TF_CU_RANGES"] = mask
         elif spec.isolation == C.IsolationPartitioned and spec.partition_xcds:
             mask = cu_mask_for_xcds(spec.partition_xcds)
-            env[C.EnvCuMask] = f"0:{mask}"
+            env[C.EnvCuMask] = f"0-15:{mask}"
             env["TF_CU_RANGES"] = mask
 
         alloc = WorkerAllocation(
