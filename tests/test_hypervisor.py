@@ -62,7 +62,7 @@ def test_allocation_env_hard_and_partitioned(hyp):
     devices, alloc, *_ = hyp
     a = alloc.allocate(make_spec(name="h", isolation=C.IsolationHard,
                                  compute_percent_limit=25.0))
-    assert a.env[C.EnvCuMask] == "0:0-63"
+    assert a.env[C.EnvCuMask] == "0-15:0-63"
     b = alloc.allocate(make_spec(name="p", isolation=C.IsolationPartitioned,
                                  partition_xcds=[2, 3]))
     assert b.env[C.EnvCuMask] == "0:64-127"
