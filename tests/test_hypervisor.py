@@ -65,7 +65,7 @@ def test_allocation_env_hard_and_partitioned(hyp):
     assert a.env[C.EnvCuMask] == "0-15:0-63"
     b = alloc.allocate(make_spec(name="p", isolation=C.IsolationPartitioned,
                                  partition_xcds=[2, 3]))
-    assert b.env[C.EnvCuMask] == "0:64-127"
+    assert b.env[C.EnvCuMask] == "0-15:64-127"
 
 
 def test_worker_shm_lifecycle_and_orphan_sweep(hyp, tmp_path):
