@@ -164,15 +164,17 @@ def build_bw_sweep():
 def build_fused_ops():
     """Fused normalization kernels (gfx950)."""
 
-    src = os.path.join(NATIVE, "ops", "fused_ops.hip")
-    if not os.path.exists(src):
+    srcs = [os.path.join(NATIVE, "ops", "fused_ops.hip"),
+            os.path.join(NATIVE, "ops", "skinny_gemm.hip")]
+    srcs = [s0 for s0 in srcs if os.path.exists(s0)]
+    if not srcs:
         return None
     os.makedirs(OUT, exist_ok=True)
     target = os.path.join(OUT, "libtfops.so")
-    if not _newer(target, [src]):
+    if not _newer(target, srcs):
         return target
     _run([HIPCC, "--offload-arch=gfx950", "-O3", "-std=c++17", "-fPIC",
-          "-shared", src, "-o", target])
+          "-shared"] + srcs + ["-o", target])
     return target
 
 

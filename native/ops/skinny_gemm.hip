@@ -1,0 +1,104 @@
+// Skinny-GEMM (decode GEMV) on gfx950 matrix cores.
+//
+//   y[M,N] = x[M,K] @ W[N,K]^T        bf16 in, fp32 accumulate, bf16 out
+//   M ≤ 16 (decode batch), K ∈ {4096, 14336, ...}, N up to 128256.
+//
+// Shape analysis (guide §3): at M ≤ 16 the GEMM is pure weight streaming
+// (2·M flops per loaded W element); one v_mfma_f32_16x16x32_bf16 consumes a
+// 16(col)×32(k) B tile = 1 KiB of W per wave per ~5-cycle instruction, so
+// the matrix pipe is never the bound — HBM is. The kernel therefore
+// optimizes the W access pattern: each wave owns one 16-column tile and
+// streams its 16 rows sequentially in 64 B/row chunks (4 k-groups × 16 B)
+// with non-temporal hints (weights are read once per token).
+//
+// Fragment mapping (cdna4: 2×K extension of the classic CDNA layout,
+// validated by tests/test_gpu_fused.py numerics vs fp32 torch):
+//   A (x):  lane l, elem j → A[i = l&15][k = (l>>4)*8 + j]
+//   B (Wᵀ): lane l, elem j → B[k = (l>>4)*8 + j][n = l&15]  = W[n][k]
+//   C/D:    lane l, reg r  → D[row = (l>>4)*4 + r][col = l&15]
+//
+// Launch: one wave per 16-column tile, 4 waves per workgroup,
+// grid.x = ceil(N / 64). x is small (≤16×K×2 B ≤ 448 KiB) and shared by
+// every workgroup — its loads hit L2 after the first tile.
+
+#include <hip/hip_bf16.h>
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+
+namespace {
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+typedef uint32_t u4 __attribute__((ext_vector_type(4)));
+
+union B16x8 {
+  u4 raw;
+  bf16x8 v;
+};
+
+__global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
+                                   const __hip_bfloat16* __restrict__ w,
+                                   __hip_bfloat16* __restrict__ y,
+                                   int M, int N, int K) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = (blockIdx.x * 4 + wave) * 16;
+  if (n0 >= N) return;
+
+  const int col = lane & 15;      // A row i / B col n / D col
+  const int kgrp = lane >> 4;     // 0..3 → k-subgroup of 8
+  const int n = n0 + col;
+  const bool ncol_ok = n < N;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+
+  // row pointer for this lane's W stream (8 bf16 = 16 B per step)
+  const __hip_bfloat16* wrow = w + (size_t)(ncol_ok ? n : 0) * K;
+  const __hip_bfloat16* xrow = x + (size_t)(col < M ? col : 0) * K;
+
+  for (int k0 = 0; k0 < K; k0 += 32) {
+    int k = k0 + kgrp * 8;
+    B16x8 a, b;
+    // x fragment: A[i=col][k..k+7]; rows ≥ M contribute zeros
+    if (col < M) {
+      a.raw = __builtin_nontemporal_load(
+          reinterpret_cast<const u4*>(xrow + k));
+    } else {
+      a.raw = u4{0, 0, 0, 0};
+    }
+    // W fragment: B[k][n] = W[n][k..k+7], streamed once → non-temporal
+    if (ncol_ok) {
+      b.raw = __builtin_nontemporal_load(
+          reinterpret_cast<const u4*>(wrow + k));
+    } else {
+      b.raw = u4{0, 0, 0, 0};
+    }
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = kgrp * 4 + r;
+    if (row < M && ncol_ok)
+      y[(size_t)row * N + n] = __float2bfloat16(acc[r]);
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// x[M,K], w[N,K] row-major bf16; y[M,N] bf16. K % 32 == 0, M <= 16.
+int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,
+                   int K, void* stream) {
+  if (M < 1 || M > 16 || (K & 31)) return 1;
+  dim3 block(256);  // 4 waves
+  dim3 grid((N + 63) / 64);
+  hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0,
+                     (hipStream_t)stream, (const __hip_bfloat16*)x,
+                     (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N, K);
+  return (int)hipGetLastError();
+}
+
+}  // extern "C"

@@ -84,3 +84,28 @@ def add_rmsnorm(x, residual, weight, eps: float):
     if rc != 0:
         raise RuntimeError(f"tf_add_rmsnorm failed: {rc}")
     return out
+
+
+def skinny_gemm(x, weight):
+    """y = x @ weight.T via the gfx950 MFMA decode-GEMV kernel
+    (native/ops/skinny_gemm.hip). x[M,K] bf16 with M ≤ 16; weight[N,K]
+    bf16 row-major (nn.Linear layout)."""
+
+    import torch
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    assert weight.dtype == torch.bfloat16 and weight.is_contiguous()
+    M, K = x.shape[-2], x.shape[-1]
+    N = weight.shape[0]
+    lb = lib()
+    if not hasattr(lb, "_skinny_ready"):
+        lb.tf_skinny_gemm.restype = ctypes.c_int
+        lb.tf_skinny_gemm.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+            ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
+        lb._skinny_ready = True
+    y = torch.empty(*x.shape[:-1], N, device=x.device, dtype=torch.bfloat16)
+    rc = lb.tf_skinny_gemm(x.data_ptr(), weight.data_ptr(), y.data_ptr(),
+                           M, N, K, _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_skinny_gemm failed: {rc}")
+    return y
