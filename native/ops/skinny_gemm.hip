@@ -61,31 +61,32 @@ __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
   // 4× unrolled K loop: 4 independent (a,b) load pairs in flight per lane
   // before their MFMAs — a single load→mfma chain left the memory queues
   // underfed (measured 3.2 TB/s; hipBLASLt streams the same shape at 5.7)
+  // NOTE x loads are CACHED (not nt): every workgroup re-reads the same
+  // small x, so last-use-marking it in L2 would re-fetch it from DRAM
+  // N/64 times. W is streamed once → nt.
   int k0 = 0;
-  for (; k0 + 128 <= K; k0 += 128) {
-    B16x8 a[4], b[4];
+  constexpr int UNROLL = 8;  // 8 W-loads (128 B/lane) in flight
+  for (; k0 + 32 * UNROLL <= K; k0 += 32 * UNROLL) {
+    B16x8 a[UNROLL], b[UNROLL];
 #pragma unroll
-    for (int u = 0; u < 4; ++u) {
+    for (int u = 0; u < UNROLL; ++u) {
       int k = k0 + u * 32 + kgrp * 8;
-      a[u].raw = arow_ok
-                     ? __builtin_nontemporal_load(
-                           reinterpret_cast<const u4*>(xrow + k))
-                     : u4{0, 0, 0, 0};
+      a[u].raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                         : u4{0, 0, 0, 0};
       b[u].raw = ncol_ok
                      ? __builtin_nontemporal_load(
                            reinterpret_cast<const u4*>(wrow + k))
                      : u4{0, 0, 0, 0};
     }
 #pragma unroll
-    for (int u = 0; u < 4; ++u)
+    for (int u = 0; u < UNROLL; ++u)
       acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u].v, b[u].v, acc, 0,
                                                     0, 0);
   }
   for (; k0 < K; k0 += 32) {
     int k = k0 + kgrp * 8;
     B16x8 a, b;
-    a.raw = arow_ok ? __builtin_nontemporal_load(
-                          reinterpret_cast<const u4*>(xrow + k))
+    a.raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
                     : u4{0, 0, 0, 0};
     b.raw = ncol_ok ? __builtin_nontemporal_load(
                           reinterpret_cast<const u4*>(wrow + k))
