@@ -25,6 +25,7 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <cstdlib>
 
 namespace {
 
@@ -102,6 +103,86 @@ __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
   }
 }
 
+// ---- v2: LDS-DMA weight streaming (guide T3 minimum 2-phase) ----------
+// Each wave stages its 16 W rows through a double-buffered LDS tile with
+// global_load_lds: the 64 lanes of one instruction read 1 KiB of ONE row
+// contiguously (fully coalesced DRAM), and the MFMA B fragments are
+// ds_read from LDS. Row stride padded by 4 elements so the 16 fragment
+// rows land on distinct bank groups.
+
+constexpr int KT = 512;            // K elements per tile (1 KiB rows)
+constexpr int ROW_PAD = 4;         // 8 B pad → row stride 258 words
+constexpr int ROW_ELEMS = KT + ROW_PAD;
+
+__global__ void __launch_bounds__(256)
+skinny_gemm_lds_kernel(const __hip_bfloat16* __restrict__ x,
+                       const __hip_bfloat16* __restrict__ w,
+                       __hip_bfloat16* __restrict__ y,
+                       int M, int N, int K) {
+  __shared__ __hip_bfloat16 tile[4][2][16][ROW_ELEMS];
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int n0 = (blockIdx.x * 4 + wave) * 16;
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n = n0 + col;
+  const bool ncol_ok = n < N && n0 < N;
+  const bool arow_ok = col < M;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const __hip_bfloat16* xrow = x + (size_t)(arow_ok ? col : 0) * K;
+
+  auto stage = [&](int buf, int kt) {
+    // 16 glds: instruction r streams W[n0+r][kt..kt+KT) → tile row r
+#pragma unroll 4
+    for (int r = 0; r < 16; ++r) {
+      int row = n0 + r < N ? n0 + r : N - 1;
+      const void* src = w + (size_t)row * K + kt + lane * 8;
+      __builtin_amdgcn_global_load_lds(
+          (const __attribute__((address_space(1))) void*)src,
+          (__attribute__((address_space(3))) void*)&tile[wave][buf][r][0],
+          16, 0, 0);
+    }
+  };
+
+  const int ntiles = K / KT;
+  int cur = 0;
+  stage(cur, 0);
+  __syncthreads();  // vmcnt(0)+lgkmcnt(0)+barrier: tile 0 resident
+  for (int t = 0; t < ntiles; ++t) {
+    if (t + 1 < ntiles) stage(cur ^ 1, (t + 1) * KT);
+#pragma unroll
+    for (int k0 = 0; k0 < KT; k0 += 32) {
+      int kk = k0 + kgrp * 8;
+      B16x8 a, b;
+      a.raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + t * KT + kk)
+                      : u4{0, 0, 0, 0};
+      b.raw = *reinterpret_cast<const u4*>(&tile[wave][cur][col][kk]);
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+    }
+    __syncthreads();  // drains the prefetch glds; next tile resident
+    cur ^= 1;
+  }
+  // K remainder (K % KT) via direct loads
+  for (int k0 = ntiles * KT; k0 < K; k0 += 32) {
+    int k = k0 + kgrp * 8;
+    B16x8 a, b;
+    a.raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                    : u4{0, 0, 0, 0};
+    b.raw = ncol_ok ? __builtin_nontemporal_load(
+                          reinterpret_cast<const u4*>(w + (size_t)n * K + k))
+                    : u4{0, 0, 0, 0};
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = kgrp * 4 + r;
+    if (row < M && ncol_ok)
+      y[(size_t)row * N + n] = __float2bfloat16(acc[r]);
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -112,9 +193,20 @@ int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,
   if (M < 1 || M > 16 || (K & 31)) return 1;
   dim3 block(256);  // 4 waves
   dim3 grid((N + 63) / 64);
-  hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0,
-                     (hipStream_t)stream, (const __hip_bfloat16*)x,
-                     (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N, K);
+  static int use_lds = [] {
+    const char* v = getenv("TF_SKINNY_LDS");
+    return v && atoi(v) != 0;
+  }();
+  if (use_lds && K >= KT)
+    hipLaunchKernelGGL(skinny_gemm_lds_kernel, grid, block, 0,
+                       (hipStream_t)stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
+                       K);
+  else
+    hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0,
+                       (hipStream_t)stream, (const __hip_bfloat16*)x,
+                       (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
+                       K);
   return (int)hipGetLastError();
 }
 
