@@ -108,6 +108,14 @@ class StoreBackend:
         if event == "DELETED" or pod.status.phase in ("Failed", "Succeeded"):
             if self.workers.get(key):
                 self.workers.remove_worker(key)
+            self._sync_external_usage()
+            return
+        if self._is_external_gpu_pod(pod):
+            # coexistence with a plain device plugin (reference kubelet
+            # checkpoint detector, external_dp/kubelet_checkpoint.go):
+            # devices claimed outside tensor-fusion are marked so the
+            # allocator's phase filter skips them (progressive migration)
+            self._sync_external_usage()
             return
         if not self._is_my_worker(pod):
             return
@@ -116,6 +124,47 @@ class StoreBackend:
         spec = self.worker_spec_from_pod(pod)
         if spec:
             self.workers.add_worker(spec)
+
+    def _is_external_gpu_pod(self, pod: Pod) -> bool:
+        if pod.status.node != self.node_name:
+            return False
+        if pod.meta.labels.get(C.LabelManaged) == "tensor-fusion":
+            return False
+        for c in pod.containers:
+            if any(r in c.resources for r in ("amd.com/gpu",
+                                              "nvidia.com/gpu")):
+                return True
+        return False
+
+    def _sync_external_usage(self):
+        """Mark GPUs consumed by non-tensor-fusion device-plugin pods
+        (via their gpu-ids annotation or device index env) as
+        used_by=external-device-plugin, and release them when freed."""
+
+        external_idx = set()
+        for pod in self.store.list("Pod"):
+            if not self._is_external_gpu_pod(pod):
+                continue
+            if pod.status.phase in ("Failed", "Succeeded"):
+                continue
+            for c in pod.containers:
+                vis = c.env.get(C.EnvVisibleDevices)
+                if vis:
+                    for tok in vis.split(","):
+                        try:
+                            external_idx.add(int(tok))
+                        except ValueError:
+                            continue
+        for d in self.devices.devices():
+            name = self.gpu_name(d)
+            want = ("external-device-plugin" if d.index in external_idx
+                    else "tensor-fusion")
+            try:
+                def _p(obj, want=want):
+                    obj.status.used_by = want
+                self.store.patch("GPU", name, "", _p)
+            except NotFound:
+                continue
 
     def worker_spec_from_pod(self, pod: Pod) -> Optional[WorkerSpec]:
         a = pod.meta.annotations

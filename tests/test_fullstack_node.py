@@ -72,3 +72,53 @@ def test_node_onboarding_and_worker_flow(tmp_path):
     erl.tick(dt=0.5)
     d = st.shm.device(0)
     assert d.erl_refill_rate > 0
+
+
+def test_external_device_plugin_coexistence(tmp_path):
+    """Devices claimed by a non-tensor-fusion pod (plain device plugin)
+    are marked used_by=external-device-plugin and skipped by the
+    allocator — the reference's kubelet-checkpoint coexistence path."""
+
+    op = build_operator()
+    pool = GPUPool()
+    pool.meta.name = "pool-a"
+    op.store.create(pool)
+    node = Node()
+    node.meta.name = "node-0"
+    op.store.create(node)
+    devices, workers, erl, backend = build_hypervisor(
+        node="node-0", mock_devices=2, shm_root=str(tmp_path / "shm"),
+        store=op.store, pool="pool-a")
+    backend.publish_devices()
+    backend.start()
+    op.tick()
+
+    ext = Pod()
+    ext.meta.name = "legacy-dp-pod"
+    ext.meta.namespace = "other"
+    ext.status.node = "node-0"
+    ext.status.phase = "Running"
+    ext.containers = [Container(name="m",
+                                resources={"amd.com/gpu": "1"},
+                                env={C.EnvVisibleDevices: "0"})]
+    op.store.create(ext)
+    op.tick()
+    g0 = op.store.get("GPU", "node-0-gpu-0")
+    g1 = op.store.get("GPU", "node-0-gpu-1")
+    assert g0.status.used_by == "external-device-plugin"
+    assert g1.status.used_by == "tensor-fusion"
+    # the allocator must refuse device 0 now: asking for EVERY device on
+    # the node cannot be satisfied while one is externally held
+    # (the mock accel's device count is process-sticky, so derive it)
+    n_dev = len(devices.devices())
+    from tensor_fusion_amd.api.types import AllocRequest, Resource
+    req = AllocRequest(pod_name="p", namespace="d", gpu_count=n_dev,
+                       request=Resource(1, 1 << 30, 1),
+                       limit=Resource(1, 1 << 30, 1))
+    scores, reasons = op.allocator.check_quota_and_filter(req)
+    assert scores == {}, scores
+    # external pod exits → device returns to the pool
+    op.store.delete("Pod", "legacy-dp-pod", "other")
+    op.tick()
+    g0 = op.store.get("GPU", "node-0-gpu-0")
+    assert g0.status.used_by == "tensor-fusion"
