@@ -94,10 +94,15 @@ class Accelerator:
             lib_path = os.path.join(
                 os.path.dirname(os.path.dirname(os.path.abspath(__file__))),
                 "_native", C.AcceleratorLibName)
+        prev_mock = os.environ.get("TF_ACCEL_MOCK")
         if mock_devices:
             os.environ["TF_ACCEL_MOCK"] = str(mock_devices)
         self._lib = ctypes.CDLL(lib_path)
         self._lib.tf_accel_init.restype = ctypes.c_int
+        # the C state is process-global; a changed mock count (tests) needs
+        # a shutdown→init cycle so the env is re-read
+        if mock_devices and prev_mock != str(mock_devices):
+            self._lib.tf_accel_shutdown()
         rc = self._lib.tf_accel_init()
         if rc != 0:
             raise RuntimeError(f"tf_accel_init failed: {rc}")
