@@ -1,3 +1,5 @@
-from .evaluator import AlertEvaluator, AlertRule, AlertState, default_rules
+from .evaluator import (AlertEvaluator, AlertRule, AlertState, default_rules,
+                        rules_from_config)
 
-__all__ = ["AlertEvaluator", "AlertRule", "AlertState", "default_rules"]
+__all__ = ["AlertEvaluator", "AlertRule", "AlertState", "default_rules",
+           "rules_from_config"]

@@ -43,6 +43,26 @@ _OPS = {
 }
 
 
+def rules_from_config(raw_rules) -> List[AlertRule]:
+    """Parse dynamic-config alert rules (config.py GlobalConfig
+    .alert_rules — the reference's hot-reloaded alert config)."""
+
+    out = []
+    for r in raw_rules or []:
+        try:
+            out.append(AlertRule(
+                name=r["name"],
+                query=r["query"],
+                threshold=float(r["threshold"]),
+                op=r.get("op", ">"),
+                interval_s=float(r.get("intervalSeconds", 60.0)),
+                severity=r.get("severity", "warning"),
+                summary=r.get("summary", "")))
+        except (KeyError, TypeError, ValueError):
+            continue
+    return out
+
+
 def default_rules() -> List[AlertRule]:
     """The reference ships similar defaults in its alert config."""
 

@@ -184,6 +184,14 @@ def build_operator(persist_dir: Optional[str] = None,
     autoscaler = Autoscaler(store, tsdb=tsdb, allocator=allocator)
     defrag = DefragController(store, allocator)
     cfg = ConfigWatcher(config_path) if config_path else None
+    if cfg is not None:
+        from .alert import default_rules, rules_from_config
+
+        def _apply_alert_rules(conf):
+            extra = rules_from_config(conf.alert_rules)
+            alerts.rules = default_rules() + extra
+        _apply_alert_rules(cfg.config)
+        cfg.on_change(_apply_alert_rules)
     return Operator(
         store=store, allocator=allocator, quota=quota, gang=gang,
         scheduler=scheduler, mutator=mutator, controllers=mgr,

@@ -220,3 +220,19 @@ class TestWebhookServer:
         resp = r.json()["response"]
         assert resp["allowed"] is True
         assert "patch" not in resp
+
+
+class TestConfigAlertRules:
+    def test_rules_hot_reload_into_evaluator(self, tmp_path):
+        from tensor_fusion_amd.operator import build_operator
+        p = tmp_path / "config.yaml"
+        p.write_text(
+            "alertRules:\n"
+            "- name: CustomHighUtil\n"
+            "  query: \"SELECT max(value) FROM points WHERE field='u'\"\n"
+            "  threshold: 90\n"
+            "  severity: critical\n")
+        op = build_operator(config_path=str(p))
+        names = {r.name for r in op.alerts.rules}
+        assert "CustomHighUtil" in names
+        assert "PoolVramSaturation" in names  # defaults retained
