@@ -191,18 +191,19 @@ extern "C" {
 int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,
                    int K, void* stream) {
   if (M < 1 || M > 16 || (K & 31)) return 1;
-  dim3 block(256);  // 4 waves
-  dim3 grid((N + 63) / 64);
+  dim3 block(256);  // 4 waves × 32 cols
+  dim3 grid((N + 127) / 128);
   static int use_lds = [] {
     const char* v = getenv("TF_SKINNY_LDS");
     return v && atoi(v) != 0;
   }();
-  if (use_lds && K >= KT)
-    hipLaunchKernelGGL(skinny_gemm_lds_kernel, grid, block, 0,
+  if (use_lds && K >= KT) {
+    dim3 grid_lds((N + 63) / 64);
+    hipLaunchKernelGGL(skinny_gemm_lds_kernel, grid_lds, block, 0,
                        (hipStream_t)stream, (const __hip_bfloat16*)x,
                        (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
                        K);
-  else
+  } else
     hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0,
                        (hipStream_t)stream, (const __hip_bfloat16*)x,
                        (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
