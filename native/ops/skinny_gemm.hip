@@ -191,8 +191,8 @@ extern "C" {
 int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,
                    int K, void* stream) {
   if (M < 1 || M > 16 || (K & 31)) return 1;
-  dim3 block(256);  // 4 waves × 32 cols
-  dim3 grid((N + 127) / 128);
+  dim3 block(256);  // 4 waves × 16 cols each
+  dim3 grid((N + 63) / 64);
   static int use_lds = [] {
     const char* v = getenv("TF_SKINNY_LDS");
     return v && atoi(v) != 0;
