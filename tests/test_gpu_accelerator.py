@@ -72,13 +72,13 @@ def test_partition_cu_mask_on_hardware(accel):
     child = r"""
 import json, sys, time
 import torch
-a = torch.randn(2048, 2048, device="cuda", dtype=torch.bfloat16)
-b = torch.randn(2048, 2048, device="cuda", dtype=torch.bfloat16)
+a = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
+b = torch.randn(4096, 4096, device="cuda", dtype=torch.bfloat16)
 for _ in range(10):
     a @ b
 torch.cuda.synchronize()
 t0 = time.perf_counter()
-for _ in range(200):
+for _ in range(100):
     a @ b
 torch.cuda.synchronize()
 print(json.dumps({"s": time.perf_counter() - t0}))
@@ -93,6 +93,7 @@ print(json.dumps({"s": time.perf_counter() - t0}))
         assert out.returncode == 0, out.stderr[-1500:]
         return json.loads(out.stdout.strip().splitlines()[-1])["s"]
 
-    full = run({})
-    part = run({"HSA_CU_MASK": mask_val})
+    # best-of-2 per config: absorbs one-off clock ramps / autotune noise
+    full = min(run({}) for _ in range(2))
+    part = min(run({"HSA_CU_MASK": mask_val}) for _ in range(2))
     assert part > 1.3 * full, (full, part)
