@@ -6,9 +6,14 @@ Usage (on a GPU box):
   rocprofv3 --pmc MfmaUtil VALUBusy --stats -d $GRAFT_REPO_ROOT/gpurun_out/pmc \
       -- python $GRAFT_REPO_ROOT/tools/pmc_target.py
 """
-import torch
+import os
+import sys
 
-from tensor_fusion_amd.ops import fused
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import torch  # noqa: E402
+
+from tensor_fusion_amd.ops import fused  # noqa: E402
 
 REPS = 50
 
