@@ -38,6 +38,7 @@ union B16x8 {
   bf16x8 v;
 };
 
+template <int UNROLL>
 __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
                                    const __hip_bfloat16* __restrict__ w,
                                    __hip_bfloat16* __restrict__ y,
@@ -66,7 +67,6 @@ __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
   // small x, so last-use-marking it in L2 would re-fetch it from DRAM
   // N/64 times. W is streamed once → nt.
   int k0 = 0;
-  constexpr int UNROLL = 8;  // 8 W-loads (128 B/lane) in flight
   for (; k0 + 32 * UNROLL <= K; k0 += 32 * UNROLL) {
     B16x8 a[UNROLL], b[UNROLL];
 #pragma unroll
@@ -110,15 +110,19 @@ __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
 // ds_read from LDS. Row stride padded by 4 elements so the 16 fragment
 // rows land on distinct bank groups.
 
-constexpr int KT = 512;            // K elements per tile (1 KiB rows)
-constexpr int ROW_PAD = 4;         // 8 B pad → row stride 258 words
-constexpr int ROW_ELEMS = KT + ROW_PAD;
+// KT (K elements per tile) is a template knob: 512 → 1 KiB rows staged
+// with 16 B/lane glds but 132 KiB LDS caps occupancy at 1 WG/CU;
+// 128 → 256 B rows (4 B/lane glds) at 33.8 KiB LDS → 4 WG/CU, letting
+// other workgroups cover each one's __syncthreads() drains.
+constexpr int ROW_PAD = 4;         // 8 B pad → fragment rows on distinct banks
 
+template <int KT>
 __global__ void __launch_bounds__(256)
 skinny_gemm_lds_kernel(const __hip_bfloat16* __restrict__ x,
                        const __hip_bfloat16* __restrict__ w,
                        __hip_bfloat16* __restrict__ y,
                        int M, int N, int K) {
+  constexpr int ROW_ELEMS = KT + ROW_PAD;
   __shared__ __hip_bfloat16 tile[4][2][16][ROW_ELEMS];
   const int lane = threadIdx.x & 63;
   const int wave = threadIdx.x >> 6;
@@ -137,11 +141,14 @@ skinny_gemm_lds_kernel(const __hip_bfloat16* __restrict__ x,
 #pragma unroll 4
     for (int r = 0; r < 16; ++r) {
       int row = n0 + r < N ? n0 + r : N - 1;
-      const void* src = w + (size_t)row * K + kt + lane * 8;
-      __builtin_amdgcn_global_load_lds(
-          (const __attribute__((address_space(1))) void*)src,
-          (__attribute__((address_space(3))) void*)&tile[wave][buf][r][0],
-          16, 0, 0);
+      const void* src = w + (size_t)row * K + kt + lane * (KT / 64);
+      auto* gsrc = (const __attribute__((address_space(1))) void*)src;
+      auto* ldst =
+          (__attribute__((address_space(3))) void*)&tile[wave][buf][r][0];
+      if constexpr (KT == 512)
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 16, 0, 0);
+      else
+        __builtin_amdgcn_global_load_lds(gsrc, ldst, 4, 0, 0);
     }
   };
 
@@ -194,20 +201,41 @@ int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,
   dim3 block(256);  // 4 waves × 16 cols each
   dim3 grid((N + 63) / 64);
   static int use_lds = [] {
-    const char* v = getenv("TF_SKINNY_LDS");
-    return v && atoi(v) != 0;
+    const char* v = getenv("TF_SKINNY_LDS");  // 1 → KT=512, 2 → KT=128
+    return v ? atoi(v) : 0;
   }();
-  if (use_lds && K >= KT) {
-    dim3 grid_lds((N + 63) / 64);
-    hipLaunchKernelGGL(skinny_gemm_lds_kernel, grid_lds, block, 0,
+  if (use_lds == 2 && K >= 128) {
+    hipLaunchKernelGGL(skinny_gemm_lds_kernel<128>, grid, block, 0,
                        (hipStream_t)stream, (const __hip_bfloat16*)x,
                        (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
                        K);
-  } else
-    hipLaunchKernelGGL(skinny_gemm_kernel, grid, block, 0,
+  } else if (use_lds && K >= 512) {
+    hipLaunchKernelGGL(skinny_gemm_lds_kernel<512>, grid, block, 0,
                        (hipStream_t)stream, (const __hip_bfloat16*)x,
                        (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
                        K);
+  } else {
+    static int unroll = [] {
+      const char* v = getenv("TF_SKINNY_UNROLL");  // W-loads in flight/lane
+      int u = v ? atoi(v) : 8;
+      return (u == 12 || u == 16) ? u : 8;
+    }();
+    if (unroll == 16)
+      hipLaunchKernelGGL(skinny_gemm_kernel<16>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
+                         K);
+    else if (unroll == 12)
+      hipLaunchKernelGGL(skinny_gemm_kernel<12>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
+                         K);
+    else
+      hipLaunchKernelGGL(skinny_gemm_kernel<8>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)w, (__hip_bfloat16*)y, M, N,
+                         K);
+  }
   return (int)hipGetLastError();
 }
 
