@@ -156,3 +156,37 @@ def test_skinny_gemm_bandwidth():
     tbs_blas = 2.0 * N * K * reps / dt_blas / 1e12
     print(f"skinny_gemm lm_head: {tbs:.2f} TB/s (hipBLASLt {tbs_blas:.2f})")
     assert tbs > 1.0
+
+
+def test_skinny_gemm_env_variants_numerics():
+    """The env-selected kernel variants (LDS KT=512/KT=128, unroll 12/16)
+    must all match the fp32 reference — run in subprocesses because the
+    selector env is latched at the library's first call."""
+
+    import subprocess
+    import sys
+
+    child = r"""
+import torch
+from tensor_fusion_amd.ops import fused
+torch.manual_seed(0)
+for (M, N, K) in [(8, 4096, 4096), (8, 1000, 14336), (16, 4096, 1024)]:
+    x = torch.randn(M, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16) * 0.1
+    got = fused.skinny_gemm(x, w).float()
+    want = x.float() @ w.float().T
+    err = (got - want).abs().max().item()
+    scale = want.abs().max().item()
+    assert err < 0.02 * scale + 0.05, (M, N, K, err)
+print("variant ok")
+"""
+    import os
+    for env_extra in ({"TF_SKINNY_LDS": "1"}, {"TF_SKINNY_LDS": "2"},
+                      {"TF_SKINNY_UNROLL": "12"},
+                      {"TF_SKINNY_UNROLL": "16"}):
+        env = dict(os.environ)
+        env.update(env_extra)
+        out = subprocess.run([sys.executable, "-c", child], env=env,
+                             capture_output=True, text=True, timeout=300)
+        assert out.returncode == 0, (env_extra, out.stderr[-1200:])
+        assert "variant ok" in out.stdout, env_extra
