@@ -47,11 +47,19 @@ class WorkerAllocation:
     up_limit_percent: int
 
 
+def _default_limiter_lib() -> str:
+    return os.path.join(os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))), "_native", "libtfhip_limiter.so")
+
+
 class AllocationController:
     def __init__(self, devices: DeviceController,
                  device_nodes: Optional[List[str]] = None,
-                 shm_root: str = C.ShmRoot):
+                 shm_root: str = C.ShmRoot,
+                 limiter_lib: Optional[str] = None):
         self.devices = devices
+        self.limiter_lib = limiter_lib if limiter_lib is not None \
+            else _default_limiter_lib()
         self.device_nodes = device_nodes or ["/dev/kfd", "/dev/dri"]
         self.shm_root = shm_root
         self._allocations: Dict[str, WorkerAllocation] = {}
@@ -77,6 +85,12 @@ class AllocationController:
         if spec.vram_limit:
             env[C.EnvVramLimit] = str(spec.vram_limit)
         env[C.EnvUpLimitPercent] = str(up_limit)
+
+        if spec.isolation == C.IsolationSoft and self.limiter_lib and \
+                os.path.exists(self.limiter_lib):
+            # the reference injects the soft limiter via an init container
+            # copying libXXX_limiter.so + LD_PRELOAD (compose.go:1576-1612)
+            env["LD_PRELOAD"] = self.limiter_lib
 
         if spec.isolation == C.IsolationHard:
             pct = spec.compute_percent_limit or up_limit

@@ -56,6 +56,10 @@ def test_allocation_env_soft(hyp):
     assert a.env[C.EnvVramLimit] == str(8 << 30)
     assert a.up_limit_percent == 25  # 625/2500
     assert C.EnvCuMask not in a.env
+    # soft isolation attaches the LD_PRELOAD limiter (compose.go:1576-1612)
+    import os
+    if os.path.exists(alloc.limiter_lib):
+        assert a.env["LD_PRELOAD"] == alloc.limiter_lib
 
 
 def test_allocation_env_hard_and_partitioned(hyp):
