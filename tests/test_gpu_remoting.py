@@ -186,3 +186,34 @@ def test_tcp_transport_cross_node_shape(tmp_path):
         assert r["tok_s"] > 0
     finally:
         h.stop()
+
+
+def test_two_workers_one_gpu_concurrent(tmp_path):
+    """Multi-tenant GPU-over-IP: two vGPU workers share one physical
+    MI355X, each serving its own GPU-less client concurrently (the
+    pooling deployment shape — oversubscribed device, isolated command
+    streams)."""
+
+    import threading
+    w1 = start_worker(str(tmp_path / "a.sock"), device_index=0)
+    w2 = start_worker(str(tmp_path / "b.sock"), device_index=0)
+    results = {}
+
+    def client(name, h):
+        env = client_env(h.socket_path)
+        out = subprocess.run([os.path.join(NATIVE, "tf_remote_testapp")],
+                             capture_output=True, text=True, timeout=240,
+                             env=env)
+        results[name] = out
+    try:
+        t1 = threading.Thread(target=client, args=("a", w1))
+        t2 = threading.Thread(target=client, args=("b", w2))
+        t1.start(); t2.start()
+        t1.join(timeout=260); t2.join(timeout=260)
+    finally:
+        w1.stop()
+        w2.stop()
+    for name in ("a", "b"):
+        out = results.get(name)
+        assert out is not None, f"client {name} did not finish"
+        assert "TESTAPP_OK" in out.stdout, (name, out.stdout, out.stderr)
