@@ -236,3 +236,29 @@ class TestConfigAlertRules:
         names = {r.name for r in op.alerts.rules}
         assert "CustomHighUtil" in names
         assert "PoolVramSaturation" in names  # defaults retained
+
+
+def test_tui_renders_device_and_worker_frame():
+    """TUI frame rendering from hypervisor API payloads (reference
+    pkg/hypervisor/tui device/worker views) — no HTTP, pure render."""
+
+    from rich.console import Console
+
+    from tensor_fusion_amd.tui.app import build_frame
+
+    devices = [{"index": 0, "uuid": "GPU-abc123", "vram_used": 24 << 30,
+                "vram_total": 288 << 30, "busy_percent": 42.0,
+                "compute_units": 256, "worker_count": 2}]
+    workers = [{"namespace": "default", "pod": "w-0", "qos": "medium",
+                "isolation": "soft",
+                "limits": {"vram": 8 << 30, "compute_percent": 25},
+                "usage": {"vram": 2 << 30, "erl_rate": 120.5,
+                          "block_ns": 3_000_000},
+                "heartbeat_ts": 0}]
+    console = Console(record=True, width=120)
+    console.print(build_frame(devices, workers))
+    text = console.export_text()
+    assert "GPU-abc123"[:10] in text
+    assert "288" in text and "24.0" in text
+    assert "default/w-0" in text
+    assert "medium" in text
