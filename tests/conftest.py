@@ -13,6 +13,8 @@ sys.path.insert(0, REPO)
 def pytest_configure(config):
     config.addinivalue_line(
         "markers", "gpu: test requires a real MI355X GPU (run via gpurun)")
+    config.addinivalue_line(
+        "markers", "slow: long-running CPU benchmark test")
 
 
 @pytest.fixture(scope="session")
