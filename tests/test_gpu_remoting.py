@@ -217,3 +217,39 @@ def test_two_workers_one_gpu_concurrent(tmp_path):
         out = results.get(name)
         assert out is not None, f"client {name} did not finish"
         assert "TESTAPP_OK" in out.stdout, (name, out.stdout, out.stderr)
+
+
+def test_remote_worker_under_erl_quota(tmp_path):
+    """Composition of the two isolation layers: the vGPU worker process
+    itself runs under the LD_PRELOAD ERL limiter (the worker-pod shape —
+    hypervisor/allocation.py attaches LD_PRELOAD for soft isolation), so
+    a GPU-less remote client is throttled by its vGPU's quota."""
+
+    import time
+
+    def run(env_extra, tag):
+        h = start_worker(str(tmp_path / f"{tag}.sock"), device_index=0,
+                         env=env_extra)
+        try:
+            env = client_env(h.socket_path)
+            t0 = time.perf_counter()
+            out = subprocess.run(
+                [sys.executable, "-m", "tensor_fusion_amd.models.llama",
+                 "--model", "tiny", "--batch", "2", "--ctx", "16",
+                 "--steps", "12", "--warmup", "2"],
+                capture_output=True, text=True, timeout=600, env=env,
+                cwd=REPO)
+            dt = time.perf_counter() - t0
+            assert out.returncode == 0, out.stdout[-1000:] + out.stderr[-3000:]
+            return json.loads(out.stdout.strip().splitlines()[-1]), dt
+        finally:
+            h.stop()
+
+    limiter = os.path.join(NATIVE, "libtfhip_limiter.so")
+    free, _ = run({}, "free")
+    capped, _ = run({"LD_PRELOAD": limiter,
+                     "TF_UP_LIMIT_PERCENT": "25",
+                     "TF_ERL_RATE": "300", "TF_ERL_CAPACITY": "30"}, "cap")
+    # a decode step is dozens of launches; at 300 launches/s the capped
+    # worker must be far below the free worker's token rate
+    assert capped["tok_s"] < 0.5 * free["tok_s"], (free, capped)
