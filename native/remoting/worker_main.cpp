@@ -119,11 +119,15 @@ struct Hip {
       if (h) break;
     }
     if (!h) return false;
-#define R(f, sym)                                      \
-  f = reinterpret_cast<decltype(f)>(dlsym(h, sym));    \
-  if (!(f)) {                                          \
-    fprintf(stderr, "[worker] missing %s\n", sym);     \
-    return false;                                      \
+// Resolve through the global scope FIRST so an LD_PRELOADed limiter
+// (soft isolation on the worker pod) interposes the worker's own HIP
+// calls; fall back to the libamdhip64 handle.
+#define R(f, sym)                                              \
+  f = reinterpret_cast<decltype(f)>(dlsym(RTLD_DEFAULT, sym)); \
+  if (!(f)) f = reinterpret_cast<decltype(f)>(dlsym(h, sym));  \
+  if (!(f)) {                                                  \
+    fprintf(stderr, "[worker] missing %s\n", sym);             \
+    return false;                                              \
   }
     R(GetDeviceCount, "hipGetDeviceCount")
     R(SetDevice, "hipSetDevice")
