@@ -60,9 +60,10 @@ __global__ void skinny_gemm_kernel(const __hip_bfloat16* __restrict__ x,
   const __hip_bfloat16* xrow = x + (size_t)(col < M ? col : 0) * K;
   const bool arow_ok = col < M;
 
-  // 4× unrolled K loop: 4 independent (a,b) load pairs in flight per lane
+  // UNROLL-deep K loop: independent (a,b) load pairs in flight per lane
   // before their MFMAs — a single load→mfma chain left the memory queues
-  // underfed (measured 3.2 TB/s; hipBLASLt streams the same shape at 5.7)
+  // underfed (3.2 TB/s vs 4.4 at depth 8; flat 8→16, see
+  // profiles/skinny_gemm_mfma_r01.md; hipBLASLt streams the shape at 5.7)
   // NOTE x loads are CACHED (not nt): every workgroup re-reads the same
   // small x, so last-use-marking it in L2 would re-fetch it from DRAM
   // N/64 times. W is streamed once → nt.
