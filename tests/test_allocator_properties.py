@@ -1,0 +1,126 @@
+"""Property-based allocator invariants (hypothesis): arbitrary
+interleavings of the two-phase lifecycle (filter → assume → commit/
+rollback → notify_bound → dealloc) must conserve capacity exactly and
+never oversubscribe any GPU — the reference defends this with dedicated
+race tests and idempotent unique-allocation maps
+(gpuallocator.go:300-316); here the whole operation space is fuzzed."""
+import pytest
+from hypothesis import given, settings
+from hypothesis import strategies as st
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.allocator.gpuallocator import (AllocationError,
+                                                      GpuAllocator)
+from tensor_fusion_amd.api.types import GPU, AllocRequest, Resource
+
+N_GPUS = 8
+CAP_T, CAP_V = 2500.0, C.MI355X_VRAM_BYTES
+
+
+def fleet():
+    a = GpuAllocator(store=None)
+    for i in range(N_GPUS):
+        g = GPU()
+        g.meta.name = f"n0-g{i}"
+        g.status.uuid = g.meta.name
+        g.status.node = "n0"
+        g.status.capacity = Resource(CAP_T, CAP_V, 100.0)
+        g.status.available = Resource(CAP_T, CAP_V, 100.0)
+        a.upsert_gpu_for_testing(g)
+    return a
+
+
+# ops: (kind, pod_id, tflops_frac_steps, vram_frac_steps, gpu_count)
+op = st.tuples(
+    st.sampled_from(["alloc", "commit", "rollback", "bind", "dealloc"]),
+    st.integers(min_value=0, max_value=11),
+    st.integers(min_value=1, max_value=10),
+    st.integers(min_value=1, max_value=10),
+    st.integers(min_value=1, max_value=2),
+)
+
+
+@settings(max_examples=120, deadline=None)
+@given(st.lists(op, min_size=1, max_size=60))
+def test_lifecycle_interleavings_conserve_capacity(ops):
+    a = fleet()
+    state = {}  # pod_key -> phase ("assumed" | "committed" | "bound")
+    for kind, pid, tf, vf, count in ops:
+        key = f"prop/p{pid}"
+        if kind == "alloc" and key not in state:
+            req = AllocRequest(
+                pod_name=f"p{pid}", namespace="prop", gpu_count=count,
+                request=Resource(CAP_T * tf / 10, int(CAP_V * vf / 10),
+                                 10.0 * tf),
+                limit=Resource(CAP_T, CAP_V, 100.0))
+            try:
+                scores, _ = a.check_quota_and_filter(req)
+                if not scores:
+                    continue
+                node = max(scores, key=lambda n: scores[n].score)
+                picked = a.pick_gpus(req, node)
+                a.assume(req, picked)
+                state[key] = "assumed"
+            except (AllocationError, KeyError):
+                continue
+        elif kind == "commit" and state.get(key) == "assumed":
+            a.commit(key)
+            state[key] = "committed"
+        elif kind == "rollback" and state.get(key) == "assumed":
+            a.rollback(key)
+            del state[key]
+        elif kind == "bind" and state.get(key) == "committed":
+            a.notify_bound(key)
+            state[key] = "bound"
+        elif kind == "dealloc" and state.get(key) in ("committed", "bound"):
+            a.dealloc(key)
+            del state[key]
+
+        # invariant 1: no GPU is ever oversubscribed or over-returned
+        for g in a.gpus():
+            assert -1e-6 <= g.status.available.tflops <= CAP_T + 1e-6, \
+                g.meta.name
+            assert 0 <= g.status.available.vram <= CAP_V, g.meta.name
+
+    # invariant 2: after releasing everything, capacity returns exactly
+    for key, phase in list(state.items()):
+        if phase == "assumed":
+            a.rollback(key)
+        else:
+            a.dealloc(key)
+    for g in a.gpus():
+        assert abs(g.status.available.tflops - CAP_T) < 1e-6
+        assert g.status.available.vram == CAP_V
+        assert abs(g.status.available.compute_percent - 100.0) < 1e-6
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(st.integers(min_value=1, max_value=10), min_size=1,
+                max_size=24))
+def test_greedy_fill_never_exceeds_device_capacity(fracs):
+    """Greedily admit pods of arbitrary VRAM fractions until filters say
+    no; total admitted per GPU must fit its capacity."""
+
+    a = fleet()
+    admitted = []
+    for i, f in enumerate(fracs):
+        req = AllocRequest(
+            pod_name=f"g{i}", namespace="prop",
+            request=Resource(10.0, int(CAP_V * f / 10), 1.0),
+            limit=Resource(CAP_T, CAP_V, 100.0))
+        scores, _ = a.check_quota_and_filter(req)
+        if not scores:
+            continue
+        node = max(scores, key=lambda n: scores[n].score)
+        try:
+            picked = a.pick_gpus(req, node)
+            a.assume(req, picked)
+            a.commit(req.pod_key)
+            admitted.append((req.pod_key, picked[0], int(CAP_V * f / 10)))
+        except (AllocationError, KeyError):
+            continue
+    per_gpu = {}
+    for _, gpu, vram in admitted:
+        per_gpu[gpu] = per_gpu.get(gpu, 0) + vram
+    for gpu, total in per_gpu.items():
+        assert total <= CAP_V, (gpu, total)
