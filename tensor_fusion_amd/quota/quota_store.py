@@ -41,8 +41,15 @@ class QuotaStore:
     def _quota(self, ns: str) -> Optional[GPUResourceQuota]:
         if not self._store:
             return None
-        return self._store.try_get("GPUResourceQuota", ns, ns) or \
+        q = self._store.try_get("GPUResourceQuota", ns, ns) or \
             self._store.try_get("GPUResourceQuota", "default", ns)
+        if q is not None:
+            return q
+        # the reference keys quotas by NAMESPACE (informer, any name):
+        # fall back to the first quota CR living in the namespace so an
+        # unconventionally-named object still applies
+        lst = self._store.list("GPUResourceQuota", namespace=ns)
+        return lst[0] if lst else None
 
     def _on_quota_event(self, event: str, obj):
         # spec changes need no local recompute: usage is tracked locally,

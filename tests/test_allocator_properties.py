@@ -124,3 +124,65 @@ def test_greedy_fill_never_exceeds_device_capacity(fracs):
         per_gpu[gpu] = per_gpu.get(gpu, 0) + vram
     for gpu, total in per_gpu.items():
         assert total <= CAP_V, (gpu, total)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.lists(
+    st.tuples(st.sampled_from(["check_assume", "commit", "forget",
+                               "release"]),
+              st.integers(min_value=0, max_value=7),
+              st.integers(min_value=1, max_value=6)),
+    min_size=1, max_size=50))
+def test_quota_store_never_exceeds_namespace_total(ops):
+    """Fuzzed two-phase quota lifecycle: committed+assumed never exceeds
+    the namespace total when every admission goes through check(), and
+    full release drains usage to zero (reference quota_store.go:400-456
+    assumed-usage overlay semantics)."""
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.api.types import GPUResourceQuota
+    from tensor_fusion_amd.quota.quota_store import (QuotaExceeded,
+                                                     QuotaStore)
+
+    store = Store()
+    q = GPUResourceQuota()
+    q.meta.name = "quota"
+    q.meta.namespace = "fuzz"
+    q.spec.total = Resource(1000.0, 100 << 30, 0.0)
+    q.spec.max_workers = 6
+    store.create(q)
+    qs = QuotaStore(store=store)
+
+    live = {}  # pid -> ("assumed"|"committed", req)
+    for kind, pid, size in ops:
+        if kind == "check_assume" and pid not in live:
+            req = AllocRequest(pod_name=f"q{pid}", namespace="fuzz",
+                               request=Resource(size * 100.0,
+                                                size * (10 << 30), 0.0))
+            try:
+                qs.check(req)
+            except QuotaExceeded:
+                continue
+            qs.assume(req)
+            live[pid] = ("assumed", req)
+        elif kind == "commit" and live.get(pid, ("",))[0] == "assumed":
+            qs.commit(live[pid][1])
+            live[pid] = ("committed", live[pid][1])
+        elif kind == "forget" and live.get(pid, ("",))[0] == "assumed":
+            qs.forget(live[pid][1])
+            del live[pid]
+        elif kind == "release" and live.get(pid, ("",))[0] == "committed":
+            qs.release(live[pid][1])
+            del live[pid]
+
+        u = qs.usage("fuzz")
+        # the admitted set never exceeds the namespace total
+        total = sum(r.request.tflops for _, r in live.values())
+        assert total <= 1000.0 + 1e-9
+        assert len(live) <= 6
+        assert u.tflops <= 1000.0 + 1e-9
+
+    for pid, (phase, req) in list(live.items()):
+        qs.forget(req) if phase == "assumed" else qs.release(req)
+    u = qs.usage("fuzz")
+    assert u.tflops == 0 and u.vram == 0
