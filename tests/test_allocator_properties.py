@@ -186,3 +186,77 @@ def test_quota_store_never_exceeds_namespace_total(ops):
         qs.forget(req) if phase == "assumed" else qs.release(req)
     u = qs.usage("fuzz")
     assert u.tflops == 0 and u.vram == 0
+
+
+@settings(max_examples=80, deadline=None)
+@given(st.lists(
+    st.tuples(st.sampled_from(["cluster", "node", "release"]),
+              st.integers(min_value=0, max_value=30),
+              st.integers(min_value=0, max_value=3)),
+    min_size=1, max_size=120))
+def test_port_allocator_never_double_assigns(ops):
+    """Fuzzed assign/release: a port is never held by two pods at once
+    and always comes from its documented range (reference
+    portallocator.go:36-307 bitmaps)."""
+
+    from tensor_fusion_amd.portallocator import PortAllocator, PortExhausted
+
+    pa = PortAllocator()
+    held = {}  # pod_key -> (kind, node, port)
+    for kind, pid, nd in ops:
+        key = f"ns/p{pid}"
+        if kind == "release":
+            pa.release(key)
+            held.pop(key, None)
+            continue
+        if key in held:
+            continue
+        try:
+            if kind == "cluster":
+                port = pa.assign_cluster_port(key)
+                assert 42000 <= port < 62000, port
+            else:
+                port = pa.assign_node_port(f"node-{nd}", key)
+                assert 40000 <= port < 42000, port
+        except PortExhausted:
+            continue
+        held[key] = (kind, nd, port)
+        # no two pods hold the same port in the same scope
+        seen = set()
+        for k, (kk, n, p) in held.items():
+            scope = ("cluster",) if kk == "cluster" else ("node", n)
+            assert (scope, p) not in seen, (k, scope, p)
+            seen.add((scope, p))
+    for key in list(held):
+        pa.release(key)
+    assert pa.in_use() == 0
+
+
+@settings(max_examples=200, deadline=None)
+@given(st.lists(st.integers(min_value=0, max_value=7), min_size=1,
+                max_size=8, unique=True),
+       st.floats(min_value=0.1, max_value=100.0))
+def test_cu_masks_cover_exactly_requested_cus(xcds, percent):
+    """CU-mask composition invariants: XCD masks cover exactly 32 CUs per
+    XCD with correct ids; percent masks round to the documented 1-CU
+    granularity and never exceed 256."""
+
+    from tensor_fusion_amd.allocator.partitioning import (cu_mask_for_percent,
+                                                          cu_mask_for_xcds)
+
+    mask = cu_mask_for_xcds(xcds)
+    cus = set()
+    for part in mask.split(","):
+        lo, hi = (int(v) for v in part.split("-"))
+        assert 0 <= lo <= hi < 256
+        cus.update(range(lo, hi + 1))
+    expect = set()
+    for x in xcds:
+        expect.update(range(32 * x, 32 * x + 32))
+    assert cus == expect
+
+    m2, count = cu_mask_for_percent(percent)
+    assert 1 <= count <= 256
+    assert abs(count - 256 * percent / 100.0) <= 1.0
+    lo, hi = (int(v) for v in m2.split("-"))
+    assert hi - lo + 1 == count
