@@ -260,3 +260,65 @@ def test_cu_masks_cover_exactly_requested_cus(xcds, percent):
     assert abs(count - 256 * percent / 100.0) <= 1.0
     lo, hi = (int(v) for v in m2.split("-"))
     assert hi - lo + 1 == count
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.lists(
+    st.tuples(st.integers(min_value=0, max_value=19),   # pod id
+              st.integers(min_value=1, max_value=7),    # vram tenths
+              st.integers(min_value=0, max_value=2)),   # preferred node
+    min_size=1, max_size=18))
+def test_defrag_simulation_is_sound(placed):
+    """Fuzzed fleets: when the defrag planner claims a candidate node can
+    be drained, every evicted allocation must (a) actually live on a
+    candidate and (b) be re-placed only on surviving nodes — and the
+    plan must cover ALL allocations on the candidates (the reference's
+    joint-placement simulation contract, gpupool_defrag.go:1095)."""
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.controllers.defrag import DefragController
+
+    a = GpuAllocator(store=None)
+    for n in range(3):
+        for i in range(4):
+            g = GPU()
+            g.meta.name = f"n{n}-g{i}"
+            g.status.uuid = g.meta.name
+            g.status.node = f"n{n}"
+            g.status.capacity = Resource(CAP_T, CAP_V, 100.0)
+            g.status.available = Resource(CAP_T, CAP_V, 100.0)
+            a.upsert_gpu_for_testing(g)
+
+    gpu_to_node = {g.meta.name: g.status.node for g in a.gpus()}
+    for pid, vf, node_pref in placed:
+        key = f"d/p{pid}"
+        if any(k == key for k, _ in a.allocations_on(
+                [f"n{n}" for n in range(3)])):
+            continue
+        req = AllocRequest(pod_name=f"p{pid}", namespace="d",
+                           request=Resource(50.0, int(CAP_V * vf / 10), 5.0),
+                           limit=Resource(CAP_T, CAP_V, 100.0))
+        scores, _ = a.check_quota_and_filter(req)
+        if not scores:
+            continue
+        node = f"n{node_pref}" if f"n{node_pref}" in scores \
+            else max(scores, key=lambda n: scores[n].score)
+        try:
+            picked = a.pick_gpus(req, node)
+            a.assume(req, picked)
+            a.commit(req.pod_key)
+            a.notify_bound(req.pod_key)
+        except (AllocationError, KeyError):
+            continue
+
+    d = DefragController(store=Store(), allocator=a)
+    for cand in (["n0"], ["n1"], ["n2"], ["n0", "n1"]):
+        on_cand = {k for k, _ in a.allocations_on(cand)}
+        plan = d.simulate(cand)
+        if plan is None:
+            continue
+        assert set(plan.evict_pods) == on_cand
+        for pod_key, gpus in plan.placements.items():
+            assert pod_key in on_cand
+            for g in gpus:
+                assert gpu_to_node[g] not in cand, (pod_key, g)
