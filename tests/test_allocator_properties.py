@@ -322,3 +322,54 @@ def test_defrag_simulation_is_sound(placed):
             assert pod_key in on_cand
             for g in gpus:
                 assert gpu_to_node[g] not in cand, (pod_key, g)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.integers(min_value=2, max_value=6),
+       st.lists(st.tuples(
+           st.sampled_from(["arrive", "permit", "schedule", "reject"]),
+           st.integers(min_value=0, max_value=7)),
+           min_size=1, max_size=40))
+def test_gang_all_or_nothing_invariants(min_members, ops):
+    """Fuzzed gang lifecycle: Permit never releases a pod before quorum
+    (waiting+scheduled ≥ min_members), and a rejection empties the
+    waiting set atomically (reference gang/manager.go strict
+    all-or-nothing :262/:1099)."""
+
+    from tensor_fusion_amd.api.types import Pod
+    from tensor_fusion_amd.gang.manager import GangManager
+
+    gm = GangManager()
+    pods = {}
+
+    def mk(i):
+        p = Pod()
+        p.meta.name = f"g{i}"
+        p.meta.namespace = "gang"
+        p.meta.annotations[C.AnnoGangEnabled] = "true"
+        p.meta.annotations[C.AnnoGangGroupKey] = "grp"
+        p.meta.annotations[C.AnnoGangMinMembers] = str(min_members)
+        return p
+
+    for kind, i in ops:
+        if kind == "arrive":
+            pods[i] = mk(i)
+            gm.register_pod(pods[i])
+        elif kind == "permit" and i in pods:
+            wait = gm.permit(pods[i])
+            g = gm.groups.get("gang/grp")
+            if wait is None:
+                # released immediately → quorum must actually be met
+                assert g.quorum_now >= min_members, \
+                    (g.quorum_now, min_members)
+            else:
+                assert wait > 0
+        elif kind == "schedule" and i in pods:
+            gm.mark_scheduled(pods[i])
+        elif kind == "reject":
+            gm.reject_group("gang/grp")
+            g = gm.groups.get("gang/grp")
+            if g is not None:
+                assert not g.waiting  # atomic clear
+                # a rejected group backs off: nothing admits until expiry
+                assert gm.pre_enqueue(mk(99)) is not None
