@@ -373,3 +373,51 @@ def test_gang_all_or_nothing_invariants(min_members, ops):
                 assert not g.waiting  # atomic clear
                 # a rejected group backs off: nothing admits until expiry
                 assert gm.pre_enqueue(mk(99)) is not None
+
+
+_ann_text = st.text(
+    alphabet=st.characters(min_codepoint=32, max_codepoint=126), max_size=24)
+
+
+@settings(max_examples=100, deadline=None)
+@given(st.dictionaries(
+    st.sampled_from([C.AnnoTflopsRequest, C.AnnoTflopsLimit,
+                     C.AnnoVramRequest, C.AnnoVramLimit, C.AnnoGpuCount,
+                     C.AnnoComputePercentRequest, C.AnnoQos,
+                     C.AnnoIsolation, C.AnnoGangMinMembers,
+                     C.AnnoGpuIndices, C.AnnoHostPort, C.AnnoGpuModel]),
+    _ann_text, max_size=8),
+    st.booleans())
+def test_admission_boundary_never_500s_on_garbage(annotations, enabled):
+    """Admission is the untrusted-input boundary: arbitrary annotation
+    values (garbage numbers, empty strings, punctuation) must never
+    crash the webhook endpoint — a malformed TF pod gets a clean
+    allowed:False denial, a non-TF pod passes through untouched
+    (reference pod_webhook.go Handle error responses)."""
+
+    fastapi = pytest.importorskip("fastapi")
+    from fastapi.testclient import TestClient
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.server.webhook_server import create_webhook_app
+    from tensor_fusion_amd.webhook import PodMutator
+
+    labels = {C.LabelEnabled: "true"} if enabled else {}
+    review = {"apiVersion": "admission.k8s.io/v1", "kind": "AdmissionReview",
+              "request": {"uid": "u", "namespace": "default", "object": {
+                  "metadata": {"name": "fz", "namespace": "default",
+                               "labels": labels,
+                               "annotations": dict(annotations)},
+                  "spec": {"containers": [{"name": "main"}]}}}}
+    client = TestClient(create_webhook_app(PodMutator(Store())))
+    r = client.post("/mutate-v1-pod", json=review)
+    assert r.status_code == 200  # never a 500
+    resp = r.json()["response"]
+    if not enabled:
+        assert resp["allowed"] is True and not resp.get("patch")
+    else:
+        # either a clean denial (bad values) or a successful mutation
+        if resp["allowed"]:
+            assert resp.get("patch") or True
+        else:
+            assert "mutation failed" in resp["status"]["message"]
