@@ -217,7 +217,8 @@ def main():
     from .server import create_operator_app
     app = create_operator_app(op.store, allocator=op.allocator,
                               port_allocator=op.port_allocator,
-                              index_allocator=op.index_allocator)
+                              index_allocator=op.index_allocator,
+                              expander=getattr(op, "expander", None))
     uvicorn.run(app, host="0.0.0.0", port=args.http_port, log_level="warning")
 
 

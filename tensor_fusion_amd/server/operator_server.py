@@ -21,6 +21,7 @@ from typing import Optional
 from fastapi import FastAPI, Header, Query
 from fastapi.responses import JSONResponse
 
+from .. import constants as C
 from ..api.store import Store
 
 
@@ -44,7 +45,8 @@ def parse_token(secret: str, token: str):
 
 def create_operator_app(store: Store, allocator=None, port_allocator=None,
                         index_allocator=None, secret: str = "tf-dev-secret",
-                        long_poll_s: float = 10.0) -> FastAPI:
+                        long_poll_s: float = 10.0,
+                        expander=None) -> FastAPI:
     app = FastAPI(title="tensor-fusion-operator")
 
     def _auth(authorization: Optional[str]):
@@ -117,6 +119,25 @@ def create_operator_app(store: Store, allocator=None, port_allocator=None,
                                for p in s.allocated_partitions],
             })
         return {"gpus": out}
+
+    @app.get("/node-scaler-info")
+    def node_scaler_info():
+        """Autoscaler-integration debug state (reference router
+        node_scaler_info.go): in-flight provisioned node claims and
+        pending unschedulable demand."""
+
+        claims = []
+        if expander is not None:
+            for c in expander.in_flight_claims():
+                claims.append({"name": c.meta.name,
+                               "instanceType": c.instance_type,
+                               "pool": c.pool,
+                               "phase": c.status.phase})
+        pending = []
+        for pod in store.list("Pod"):
+            if pod.meta.annotations.get(f"{C.Domain}/unschedulable"):
+                pending.append(pod.meta.key)
+        return {"inFlightClaims": claims, "pendingPods": pending}
 
     @app.get("/healthz")
     def healthz():

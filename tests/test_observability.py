@@ -262,3 +262,25 @@ def test_tui_renders_device_and_worker_frame():
     assert "288" in text and "24.0" in text
     assert "default/w-0" in text
     assert "medium" in text
+
+
+def test_node_scaler_info_route():
+    """GET /node-scaler-info exposes in-flight provisioning claims
+    (reference router/node_scaler_info.go)."""
+
+    from tensor_fusion_amd.api.types import GPUNodeClaim
+    from tensor_fusion_amd.scheduler.expander import NodeExpander
+
+    store = Store()
+    c = GPUNodeClaim()
+    c.meta.name = "claim-1"
+    c.pool = "pool-a"
+    c.instance_type = "mi355x.8x"
+    store.create(c)
+    app = create_operator_app(store, expander=NodeExpander(store))
+    from fastapi.testclient import TestClient
+    r = TestClient(app).get("/node-scaler-info")
+    assert r.status_code == 200
+    data = r.json()
+    assert data["inFlightClaims"][0]["name"] == "claim-1"
+    assert data["inFlightClaims"][0]["instanceType"] == "mi355x.8x"
