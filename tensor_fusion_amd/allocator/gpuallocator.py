@@ -558,6 +558,7 @@ class GpuAllocator:
                     continue
                 g.status.available = g.status.available.sub(delta)
                 self._dirty.add(n)
+                self._soa_sync(n)
 
     # ---------------------------------------------------- preemption sim
 

@@ -272,3 +272,41 @@ def test_dirty_sync_to_store_and_reconcile():
     # rebuild resets then replays: no double subtraction
     g = a2.gpu(a.allocation(req.pod_key).gpu_names[0])
     assert g.status.available.vram == C.MI355X_VRAM_BYTES - (10 << 30)
+
+
+def test_adjust_allocation_updates_soa_filter_view():
+    """Regression: the numpy SoA fast-filter mirror must see vertical
+    scaling immediately — a scale-up that fills the only GPU has to make
+    the next filter pass come back empty, with no other state change in
+    between to invalidate the mirror."""
+
+    from tensor_fusion_amd.api.types import GPU, AllocRequest
+
+    a = GpuAllocator(store=None)
+    g = GPU()
+    g.meta.name = "solo-g0"
+    g.status.uuid = g.meta.name
+    g.status.node = "solo"
+    g.status.capacity = Resource(2500.0, 1 << 40, 100.0)
+    g.status.available = Resource(2500.0, 1 << 40, 100.0)
+    a.upsert_gpu_for_testing(g)
+    cap = g.status.capacity
+
+    req = AllocRequest(pod_name="soa", namespace="d",
+                       request=Resource(10.0, 1 << 30, 1.0),
+                       limit=Resource(cap.tflops, cap.vram, 100.0))
+    scores, _ = a.check_quota_and_filter(req)  # builds the SoA mirror
+    assert scores
+    picked = a.pick_gpus(req, "solo")
+    a.assume(req, picked)
+    a.commit(req.pod_key)
+
+    # scale the pod up to (almost) the whole GPU's VRAM
+    a.adjust_allocation(req.pod_key,
+                        Resource(10.0, cap.vram - (1 << 20), 1.0))
+    # a half-GPU pod must now be rejected — stale mirror would admit it
+    req2 = AllocRequest(pod_name="soa2", namespace="d",
+                        request=Resource(10.0, cap.vram // 2, 1.0),
+                        limit=Resource(cap.tflops, cap.vram, 100.0))
+    scores2, _ = a.check_quota_and_filter(req2)
+    assert scores2 == {}, scores2
