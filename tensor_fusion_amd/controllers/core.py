@@ -101,9 +101,32 @@ class PoolReconciler(Reconciler):
 
 class NodeReconciler(Reconciler):
     """k8s Node → GPUNode CR for nodes matching a pool's selector
-    (reference node_controller.go:220 generateGPUNode)."""
+    (reference node_controller.go:220 generateGPUNode); node deletion
+    tears down the GPUNode and its GPU inventory (node failure path,
+    gpunode_controller.go:376)."""
 
     kind = "Node"
+
+    def __init__(self, store: Store):
+        super().__init__(store)
+        store.on_change("Node", self._on_event)
+
+    def _on_event(self, event: str, obj):
+        if event != "DELETED":
+            return
+        name = obj.meta.name
+        # deleting the GPU CRs evicts them from the allocator through its
+        # own informer (gpuallocator _evict on DELETE)
+        for g in list(self.store.list("GPU")):
+            if g.status.node == name:
+                try:
+                    self.store.delete("GPU", g.meta.name)
+                except Exception:
+                    pass
+        try:
+            self.store.delete("GPUNode", name)
+        except Exception:
+            pass
 
     def reconcile(self, req: Request):
         node = self.store.get(self.kind, req.name)

@@ -531,3 +531,39 @@ class TestConnectionFailover:
         assert conn.status.worker
         ip = conn.status.connection_url.split("+")[1]
         assert ip.startswith("10.0.0.")
+
+
+def test_node_deletion_tears_down_gpunode_and_inventory():
+    """Node failure path (reference gpunode_controller.go:376): deleting
+    the k8s Node removes its GPUNode and GPU CRs, and the allocator
+    stops offering that node's capacity."""
+
+    from tensor_fusion_amd.operator import build_operator
+
+    op = build_operator()
+    pool = GPUPool()
+    pool.meta.name = "pool-a"
+    op.store.create(pool)
+    node = Node()
+    node.meta.name = "dying-node"
+    op.store.create(node)
+    op.tick()
+    # publish two GPUs on the node (as the hypervisor backend would)
+    for i in range(2):
+        g = GPU()
+        g.meta.name = f"dying-node-gpu-{i}"
+        g.status.uuid = g.meta.name
+        g.status.node = "dying-node"
+        g.status.pool = "pool-a"
+        g.status.capacity = Resource(2500.0, 288 << 30, 100.0)
+        g.status.available = Resource(2500.0, 288 << 30, 100.0)
+        op.store.create(g)
+    op.tick()
+    assert len(op.allocator.gpus(node="dying-node")) == 2
+
+    op.store.delete("Node", "dying-node")
+    op.tick()
+    assert op.store.try_get("GPUNode", "dying-node") is None
+    assert [g for g in op.store.list("GPU")
+            if g.status.node == "dying-node"] == []
+    assert op.allocator.gpus(node="dying-node") == []
