@@ -144,6 +144,20 @@ class Store:
                 raise NotFound(f"{kind} {key}")
             self._persist(kind)
             self._notify("DELETED", obj)
+        # owner-reference GC (kube garbage collector semantics): objects
+        # whose meta.owner points at the deleted object go with it
+        ref = f"{kind}/{namespace}/{name}" if namespace else f"{kind}//{name}"
+        owned = []
+        with self._lock:
+            for k2, b2 in self._objs.items():
+                for o in b2.values():
+                    if getattr(o.meta, "owner", "") == ref:
+                        owned.append((k2, o.meta.name, o.meta.namespace))
+        for k2, n2, ns2 in owned:
+            try:
+                self.delete(k2, n2, ns2)
+            except NotFound:
+                pass
 
     def list(self, kind: str, namespace: Optional[str] = None,
              labels: Optional[Dict[str, str]] = None) -> List[TFObject]:

@@ -567,3 +567,30 @@ def test_node_deletion_tears_down_gpunode_and_inventory():
     assert [g for g in op.store.list("GPU")
             if g.status.node == "dying-node"] == []
     assert op.allocator.gpus(node="dying-node") == []
+
+
+def test_owner_reference_gc_cascades():
+    """Store-level owner GC (kube garbage collector semantics): deleting
+    an owner deletes transitively-owned objects — Node → GPUNode →
+    hypervisor Pod in one cascade."""
+
+    from tensor_fusion_amd.operator import build_operator
+
+    op = build_operator()
+    pool = GPUPool()
+    pool.meta.name = "pool-a"
+    op.store.create(pool)
+    node = Node()
+    node.meta.name = "gc-node"
+    op.store.create(node)
+    op.tick()
+    gn = op.store.try_get("GPUNode", "gc-node")
+    assert gn is not None and gn.meta.owner == "Node//gc-node"
+    op.tick()
+    hyp = op.store.try_get("Pod", "hypervisor-gc-node", "tensor-fusion-sys")
+    assert hyp is not None and hyp.meta.owner == "GPUNode//gc-node"
+
+    op.store.delete("Node", "gc-node")
+    assert op.store.try_get("GPUNode", "gc-node") is None
+    assert op.store.try_get("Pod", "hypervisor-gc-node",
+                            "tensor-fusion-sys") is None
