@@ -223,3 +223,30 @@ class TestRestartRecovery:
               if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker
               and p.meta.labels.get(C.LabelWorkload) == "app-2-wl"]
         assert w2 and w2[0].status.node
+
+
+def test_workload_deletion_releases_capacity():
+    """Deleting a TensorFusionWorkload cascades to its worker pods
+    (owner GC) and the allocator returns their capacity (§3.5 dealloc
+    path, end to end)."""
+
+    op = build_operator()
+    mk_world(op)
+    pod = client_pod(annotations={C.AnnoTflopsRequest: "100",
+                                  C.AnnoVramRequest: str(16 << 30)})
+    op.admit(pod)
+    for _ in range(6):
+        op.tick()
+    workers = [p for p in op.store.list("Pod", namespace="default")
+               if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker]
+    assert workers
+    before = sum(g.status.available.vram for g in op.allocator.gpus())
+    wl_name = workers[0].meta.labels[C.LabelWorkload]
+    op.store.delete("TensorFusionWorkload", wl_name, "default")
+    op.tick()
+    left = [p for p in op.store.list("Pod", namespace="default")
+            if p.meta.labels.get(C.LabelComponent) == C.ComponentWorker
+            and p.meta.labels.get(C.LabelWorkload) == wl_name]
+    assert left == []
+    after = sum(g.status.available.vram for g in op.allocator.gpus())
+    assert after > before  # the workers' VRAM came back
