@@ -121,19 +121,20 @@ class Store:
 
     def patch(self, kind: str, name: str, namespace: str,
               fn: Callable[[TFObject], None], retries: int = 8) -> TFObject:
-        """Read-modify-write with conflict retry (the controllers' idiom)."""
+        """Atomic read-modify-write (the controllers' idiom). The store is
+        in-process, so the whole RMW runs under the store lock — unlike a
+        remote apiserver there is no window for a conflicting writer, and
+        concurrent patches can never exhaust retries and drop updates
+        (found by tests/test_allocator_properties.py fuzzing the old
+        bounded-retry loop under thread contention)."""
 
-        for _ in range(retries):
+        with self._lock:
             obj = self.get(kind, name, namespace)
             before = _to_dict(obj)
             fn(obj)
             if _to_dict(obj) == before:
                 return obj  # no-op patch: no rv bump, no event storm
-            try:
-                return self.update(obj)
-            except Conflict:
-                time.sleep(0.001)
-        raise Conflict(f"{kind} {namespace}/{name}: retries exhausted")
+            return self.update(obj)
 
     def delete(self, kind: str, name: str, namespace: str = "") -> None:
         key = f"{namespace}/{name}" if namespace else name

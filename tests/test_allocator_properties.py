@@ -459,3 +459,38 @@ def test_erl_pid_rate_always_bounded_and_slew_limited(setpoint, utils, dt):
         assert abs(final_util - setpoint) <= \
             max(p.deadband_percent * 2, 0.12 * setpoint), \
             (setpoint, final_util)
+
+
+@settings(max_examples=60, deadline=None)
+@given(st.integers(min_value=2, max_value=8),
+       st.integers(min_value=5, max_value=40))
+def test_store_concurrent_patches_lose_no_updates(n_threads, n_incr):
+    """The store's patch() retry loop under real thread contention:
+    N threads × M increments on one object must all land (no lost
+    updates through the RV-conflict path) and every event fires."""
+
+    import threading
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.api.types import GPUPool
+
+    store = Store()
+    p = GPUPool()
+    p.meta.name = "ctr"
+    store.create(p)
+    events = []
+    store.on_change("GPUPool", lambda e, o: events.append(e))
+
+    def worker():
+        for _ in range(n_incr):
+            def _p(obj):
+                obj.status.gpu_count += 1
+            store.patch("GPUPool", "ctr", "", _p)
+
+    threads = [threading.Thread(target=worker) for _ in range(n_threads)]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    got = store.get("GPUPool", "ctr")
+    assert got.status.gpu_count == n_threads * n_incr
