@@ -37,8 +37,11 @@ def _rank_main(rank: int, world: int, port: int, q):
 
 
 def test_bench_two_rank_gloo_aggregation():
+    import socket
     world = 2
-    port = 29771
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
     procs = [ctx.Process(target=_rank_main, args=(r, world, port, q))
