@@ -9,6 +9,13 @@ import pytest
 
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
+
+def _free_port() -> str:
+    import socket
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        return str(sk.getsockname()[1])
+
 WORKER = r"""
 import os
 import torch
@@ -76,7 +83,7 @@ def test_tp_matches_reference_2rank():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29517", "-m", "tensor_fusion_amd.parallel._tp_test_worker"],
+         "--master-port", _free_port(), "-m", "tensor_fusion_amd.parallel._tp_test_worker"],
         env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
     assert out.returncode == 0, out.stdout[-3000:] + out.stderr[-3000:]
     assert "TP_OK" in out.stdout, out.stdout
@@ -90,7 +97,7 @@ def test_bench_tp_tool_2rank_cpu():
     out = subprocess.run(
         [sys.executable, "-m", "torch.distributed.run", "--nnodes=1",
          "--nproc-per-node", "2", "--master-addr", "127.0.0.1",
-         "--master-port", "29531", "tools/bench_tp.py", "--model", "tiny",
+         "--master-port", _free_port(), "tools/bench_tp.py", "--model", "tiny",
          "--batch", "2", "--ctx", "8", "--steps", "2", "--warmup", "1",
          "--device", "cpu", "--backend", "gloo"],
         env=env, capture_output=True, text=True, timeout=600, cwd=REPO)
