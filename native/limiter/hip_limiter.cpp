@@ -73,6 +73,9 @@ struct AllocMap {
 struct ExpandedRange {
   size_t bytes = 0;
   bool device_resident = false;  // last prefetch target was HBM
+  int device = 0;  // owning device at expand_alloc time — the tier thread
+                   // never calls hipSetDevice, so its tls_device is always
+                   // 0 and must not be used as a prefetch/advise target
 };
 
 struct Limiter {
@@ -334,7 +337,7 @@ hipError_t expand_alloc(void** p, size_t sz) {
   eh.MemAdvise(*p, sz, kAdviseSetAccessedBy, tls_device);
   {
     std::lock_guard<std::mutex> l(g.expand_mu);
-    g.expanded[*p] = ExpandedRange{sz, false};
+    g.expanded[*p] = ExpandedRange{sz, false, tls_device};
   }
   g.expanded_bytes.fetch_add(sz, std::memory_order_relaxed);
   dbg("expand_alloc %zu B -> host tier (total expanded %lu)", sz,
@@ -352,7 +355,7 @@ uint64_t tier_migrate(bool to_device, uint64_t budget) {
   for (auto& [ptr, r] : g.expanded) {
     if (to_device == r.device_resident) continue;
     if (to_device && moved + r.bytes > budget) continue;
-    int dst = to_device ? tls_device : kHipCpuDeviceId;
+    int dst = to_device ? r.device : kHipCpuDeviceId;
     eh.MemAdvise(ptr, r.bytes, kAdviseSetPreferredLocation, dst);
     if (eh.MemPrefetchAsync(ptr, r.bytes, dst, nullptr) == hipSuccess) {
       r.device_resident = to_device;

@@ -75,10 +75,19 @@ class RingView {
   void wake_consumer() {
     // load-before-RMW: during bursts the flag is already 1 and the plain
     // load keeps the line shared instead of ping-ponging an exchange
-    // between the producer and consumer cores on every record
-    if (at(&r_->futex_nonempty)->load(std::memory_order_relaxed) == 1)
+    // between the producer and consumer cores on every record.
+    // The fence makes this safe: without it the head.store(release) in
+    // commit() can sit in the producer's store buffer while the flag load
+    // executes, so the producer could read flag==1 even though the consumer
+    // has already cleared the flag and re-checked head — both sides would
+    // then skip the wake and the consumer parks (lost wakeup). seq_cst
+    // fence + seq_cst flag ops on both sides restore a total order between
+    // {head publish, flag load} here and {flag clear, head re-check} in
+    // wait_nonempty.
+    std::atomic_thread_fence(std::memory_order_seq_cst);
+    if (at(&r_->futex_nonempty)->load(std::memory_order_seq_cst) == 1)
       return;
-    if (at(&r_->futex_nonempty)->exchange(1, std::memory_order_release) == 0)
+    if (at(&r_->futex_nonempty)->exchange(1, std::memory_order_seq_cst) == 0)
       futex_wake(&r_->futex_nonempty);
   }
 
@@ -122,10 +131,11 @@ class RingView {
       __builtin_ia32_pause();
 #endif
     }
-    // exchange (full barrier), not a plain store: pairs with the
-    // producer's cheap load in wake_consumer — the barrier guarantees the
-    // head re-check below observes any commit whose wake was skipped
-    at(&r_->futex_nonempty)->exchange(0, std::memory_order_acq_rel);
+    // seq_cst exchange, not a plain store: pairs with the seq_cst fence +
+    // flag load in wake_consumer — the total order guarantees either the
+    // head re-check below observes the producer's commit, or the producer's
+    // flag load observes our clear and takes the exchange+wake path
+    at(&r_->futex_nonempty)->exchange(0, std::memory_order_seq_cst);
     uint64_t tail = at(&r_->tail)->load(std::memory_order_relaxed);
     uint64_t head = at(&r_->head)->load(std::memory_order_acquire);
     if (head > tail) return;

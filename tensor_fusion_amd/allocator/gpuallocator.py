@@ -226,8 +226,15 @@ class GpuAllocator:
                 & (soa["av_v"] >= r.vram)
                 & (soa["av_c"] >= r.compute_percent - 1e-9))
         empty = (soa["n_apps"] == 0) & (soa["n_parts"] == 0)
-        iso_ok = empty | ((soa["iso"] == self._ISO_CODES.get(
-            req.isolation_mode, 1)) & (soa["n_parts"] == 0))
+        if req.isolation_mode == C.IsolationShared:
+            # shared = whole-GPU: mirror shared_whole_gpu_filter exactly —
+            # only fully-free devices (no apps, all VRAM available) qualify,
+            # else the fast path ranks infeasible nodes that Reserve then
+            # rejects (retry churn / unschedulable shared pods)
+            iso_ok = empty & (soa["av_v"] >= soa["cap_v"] - 0.5)
+        else:
+            iso_ok = empty | ((soa["iso"] == self._ISO_CODES.get(
+                req.isolation_mode, 1)) & (soa["n_parts"] == 0))
         ok = soa["ready"] & fits & iso_ok
         if not ok.any():
             reasons = {}

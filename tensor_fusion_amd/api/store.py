@@ -67,14 +67,55 @@ class Store:
         self._watches: Dict[str, List[Queue]] = {}
         self._handlers: Dict[str, List[Callable[[str, TFObject], None]]] = {}
         self._persist_dir = persist_dir
+        # per-thread re-entrancy depth + deferred handler events: handlers
+        # must run OUTSIDE the store lock (informer contract) or a handler
+        # that takes its own lock can deadlock against a thread holding
+        # that lock while calling into the store
+        self._tls = threading.local()
         if persist_dir:
             os.makedirs(persist_dir, exist_ok=True)
             self._load()
 
+    class _Mutate:
+        """Context manager: store lock + deferred handler dispatch.
+
+        Watch-queue puts happen inline (Queue.put never calls back into
+        user code); on_change handlers collected during the mutation fire
+        after the OUTERMOST lock release, preserving event order."""
+
+        def __init__(self, store: "Store"):
+            self.store = store
+
+        def __enter__(self):
+            s = self.store
+            s._lock.acquire()
+            s._tls.depth = getattr(s._tls, "depth", 0) + 1
+            if s._tls.depth == 1:
+                s._tls.pending = []
+            return self
+
+        def __exit__(self, *exc):
+            s = self.store
+            s._tls.depth -= 1
+            outermost = s._tls.depth == 0
+            pending = s._tls.pending if outermost else None
+            if outermost:
+                s._tls.pending = []
+            s._lock.release()
+            if pending:
+                for event, obj, handlers in pending:
+                    for h in handlers:
+                        try:
+                            h(event, copy.deepcopy(obj))
+                        except Exception:  # handlers must not break callers
+                            import traceback
+                            traceback.print_exc()
+            return False
+
     # ------------------------------------------------------------- CRUD
 
     def create(self, obj: TFObject) -> TFObject:
-        with self._lock:
+        with self._Mutate(self):
             kind = obj.kind
             key = obj.meta.key
             bucket = self._objs.setdefault(kind, {})
@@ -102,7 +143,7 @@ class Store:
             return None
 
     def update(self, obj: TFObject, check_rv: bool = True) -> TFObject:
-        with self._lock:
+        with self._Mutate(self):
             bucket = self._objs.setdefault(obj.kind, {})
             key = obj.meta.key
             cur = bucket.get(key)
@@ -128,7 +169,7 @@ class Store:
         (found by tests/test_allocator_properties.py fuzzing the old
         bounded-retry loop under thread contention)."""
 
-        with self._lock:
+        with self._Mutate(self):
             obj = self.get(kind, name, namespace)
             before = _to_dict(obj)
             fn(obj)
@@ -138,7 +179,7 @@ class Store:
 
     def delete(self, kind: str, name: str, namespace: str = "") -> None:
         key = f"{namespace}/{name}" if namespace else name
-        with self._lock:
+        with self._Mutate(self):
             bucket = self._objs.get(kind, {})
             obj = bucket.pop(key, None)
             if obj is None:
@@ -181,7 +222,9 @@ class Store:
         return Watch(self, kind, q)
 
     def on_change(self, kind: str, handler: Callable[[str, TFObject], None]):
-        """Informer-style synchronous handler (called under no lock)."""
+        """Informer-style synchronous handler. Handlers run AFTER the store
+        lock is released (outermost mutation exit), so they may freely take
+        their own locks and call back into the store."""
 
         with self._lock:
             self._handlers.setdefault(kind, []).append(handler)
@@ -196,12 +239,19 @@ class Store:
     def _notify(self, event: str, obj: TFObject):
         for q in self._watches.get(obj.kind, []):
             q.put((event, copy.deepcopy(obj)))
-        for h in list(self._handlers.get(obj.kind, [])):
-            try:
-                h(event, copy.deepcopy(obj))
-            except Exception:  # handlers must not break the store
-                import traceback
-                traceback.print_exc()
+        handlers = list(self._handlers.get(obj.kind, []))
+        if not handlers:
+            return
+        if getattr(self._tls, "depth", 0) > 0:
+            # defer to the outermost _Mutate exit (fires outside the lock)
+            self._tls.pending.append((event, copy.deepcopy(obj), handlers))
+        else:
+            for h in handlers:
+                try:
+                    h(event, copy.deepcopy(obj))
+                except Exception:  # handlers must not break the store
+                    import traceback
+                    traceback.print_exc()
 
     # ---------------------------------------------------------- persist
 

@@ -167,6 +167,7 @@ class Scheduler:
         self._bind_fn = bind_fn or self._default_bind
         self._unsched_backoff: Dict[str, float] = {}
         self.results: Dict[str, ScheduleResult] = {}
+        self._next_start = 0  # rotating feasible-node scan offset
         # event-maintained ready-node cache: store.list deepcopies every
         # object, which dominated the scheduling hot path at 1k nodes
         self._node_mu = threading.Lock()
@@ -245,7 +246,15 @@ class Scheduler:
 
         cand = [n for n in self._nodes() if allowed is None or n in allowed]
         # kube-scheduler's numFeasibleNodesToFind: at scale, stop after
-        # enough feasible nodes instead of filtering/scoring all of them
+        # enough feasible nodes instead of filtering/scoring all of them.
+        # Rotate the scan start across cycles (kube-scheduler's
+        # nextStartNodeIndex) so truncation doesn't always consider the
+        # same prefix of the fleet — without this, nodes past `want` are
+        # never scored and placement hotspots on the first nodes.
+        if len(cand) > 1:
+            start = self._next_start % len(cand)
+            self._next_start += 1
+            cand = cand[start:] + cand[:start]
         want = max(100, len(cand) * 8 // 100)
         feasible = []
         reasons: List[str] = []
