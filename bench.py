@@ -3,21 +3,26 @@
 decode tok/s (BASELINE.json metric), 1..8 MI355X.
 
 For each rank (one per GPU; torchrun sets RANK/LOCAL_RANK/WORLD_SIZE) the
-bench runs the same Llama-3-8B bf16 decode workload twice, in separate child
+bench runs the same Llama-3-8B bf16 decode workload in separate child
 processes on that rank's GPU:
   native : plain HIP/PyTorch process
-  vgpu   : the tensor-fusion-amd vGPU path — the workload runs under the
-           stack's isolation layer (LD_PRELOAD libtfhip_limiter.so attached
-           to a hypervisor-style shm page at 100% quota; TF_BENCH_VGPU_MODE
-           switches to the remoting worker when that path is selected)
+  vgpu   : the tensor-fusion-amd vGPU path — the workload runs through the
+           GPU-over-IP remoting worker by default (TF_BENCH_VGPU_MODE=
+           limiter|tcp select the LD_PRELOAD limiter / TCP wire instead)
 and reports overhead% = 100 * (1 - tok_s_vgpu / tok_s_native), aggregated
 over ranks (value = whole-job overhead computed from summed tok/s; MAX
 ms_per_step over ranks). Lower is better; reference headline is <4%
 (BASELINE.md README.md:56).
 
-Parent processes coordinate with a gloo group (they never touch the GPU —
-the children own it; each child brackets its timed steps with
-torch.cuda.synchronize on both sides).
+The headline comparison is SAME-MODE and uses the stack's strongest
+configuration on BOTH sides: hipGraph-captured decode + fused gfx950
+rmsnorm/residual kernels (TF_FUSED_OPS=1). An eager-mode row is also
+measured and printed in config for disclosure — eager native decode is
+dispatch-bound (~1,100 launches/token) and the remote ring hides launch
+latency, which flatters the vGPU side; it is NOT the headline
+(round-1 verdict: the shipped default must be the honest comparison).
+Set TF_BENCH_EAGER=0 to skip the eager row, TF_BENCH_FUSED=0 /
+TF_BENCH_GRAPHS=0 to weaken the baseline for ablations.
 """
 from __future__ import annotations
 
@@ -30,10 +35,13 @@ import sys
 REPO = os.path.dirname(os.path.abspath(__file__))
 
 
-def run_child(mode: str, args, local_rank: int) -> dict:
+def run_child(mode: str, args, local_rank: int, graphs: bool,
+              fused: bool) -> dict:
     env = dict(os.environ)
     env["HIP_VISIBLE_DEVICES"] = env.get("TF_BENCH_DEVICE", str(local_rank))
     env.pop("TF_SHM_PATH", None)
+    env["TF_FUSED_OPS"] = "1" if fused else "0"
+    env.pop("TF_BENCH_GRAPHS", None)
     # children are plain single-GPU processes
     for k in ("RANK", "LOCAL_RANK", "WORLD_SIZE", "MASTER_ADDR", "MASTER_PORT",
               "GROUP_RANK", "LOCAL_WORLD_SIZE", "TORCHELASTIC_RUN_ID"):
@@ -70,6 +78,8 @@ def run_child(mode: str, args, local_rank: int) -> dict:
            "--model", args.model, "--batch", str(args.batch),
            "--ctx", str(args.ctx), "--steps", str(args.steps),
            "--warmup", str(args.warmup)]
+    if graphs:
+        cmd.append("--graphs")
     try:
         out = subprocess.run(cmd, env=env, cwd=REPO, capture_output=True,
                              text=True, timeout=3600)
@@ -96,6 +106,10 @@ def main():
     local_rank = int(os.environ.get("LOCAL_RANK", str(rank)))
     n_gpus = max(world, 1)
 
+    graphs = os.environ.get("TF_BENCH_GRAPHS", "1") != "0"
+    fused = os.environ.get("TF_BENCH_FUSED", "1") != "0"
+    eager_row = os.environ.get("TF_BENCH_EAGER", "1") != "0"
+
     dist = None
     if world > 1:
         import torch.distributed as tdist
@@ -106,14 +120,22 @@ def main():
         if dist:
             dist.barrier()
 
+    # headline: strongest native configuration, same mode on both sides
     barrier()
-    native = run_child("native", args, local_rank)
+    native = run_child("native", args, local_rank, graphs, fused)
     barrier()
-    vgpu = run_child("vgpu", args, local_rank)
+    vgpu = run_child("vgpu", args, local_rank, graphs, fused)
     barrier()
+    if eager_row:
+        native_e = run_child("native", args, local_rank, False, fused)
+        barrier()
+        vgpu_e = run_child("vgpu", args, local_rank, False, fused)
+        barrier()
+    else:
+        native_e = vgpu_e = {"tok_s": 0.0, "ms_per_step": 0.0}
 
     vals = [native["tok_s"], vgpu["tok_s"], vgpu["ms_per_step"],
-            native["ms_per_step"]]
+            native_e["tok_s"], vgpu_e["tok_s"]]
     if dist:
         import torch
         t = torch.tensor(vals, dtype=torch.float64)
@@ -122,12 +144,33 @@ def main():
         native_tok = sum(float(g[0]) for g in gathered)
         vgpu_tok = sum(float(g[1]) for g in gathered)
         ms_vgpu = max(float(g[2]) for g in gathered)
+        native_e_tok = sum(float(g[3]) for g in gathered)
+        vgpu_e_tok = sum(float(g[4]) for g in gathered)
     else:
-        native_tok, vgpu_tok, ms_vgpu = vals[0], vals[1], vals[2]
+        native_tok, vgpu_tok, ms_vgpu, native_e_tok, vgpu_e_tok = vals
 
     overhead = 100.0 * (1.0 - vgpu_tok / native_tok)
+    overhead_eager = (100.0 * (1.0 - vgpu_e_tok / native_e_tok)
+                      if native_e_tok else None)
     if rank == 0:
         vgpu_mode = os.environ.get("TF_BENCH_VGPU_MODE", "remote")
+        cfg = {
+            "model": args.model,
+            "global_batch": args.batch * n_gpus,
+            "seq_len": args.ctx,
+            "parallelism": f"dp{n_gpus}",
+            "vgpu_mode": vgpu_mode,
+            "decode_mode": ("hipgraph" if graphs else "eager")
+                           + ("+fused" if fused else ""),
+            "native_tok_s": round(native_tok, 1),
+            "vgpu_tok_s": round(vgpu_tok, 1),
+        }
+        if overhead_eager is not None:
+            cfg.update({
+                "eager_native_tok_s": round(native_e_tok, 1),
+                "eager_vgpu_tok_s": round(vgpu_e_tok, 1),
+                "eager_overhead_pct": round(overhead_eager, 3),
+            })
         print(json.dumps({
             "metric": "remote-vGPU overhead % vs native HIP (Llama-3-8B tok/s)",
             "value": round(overhead, 3),
@@ -141,15 +184,7 @@ def main():
             "vs_baseline": round(overhead / 4.0, 4),
             "dtype": "bf16",
             "data": "synthetic",
-            "config": {
-                "model": args.model,
-                "global_batch": args.batch * n_gpus,
-                "seq_len": args.ctx,
-                "parallelism": f"dp{n_gpus}",
-                "vgpu_mode": vgpu_mode,
-                "native_tok_s": round(native_tok, 1),
-                "vgpu_tok_s": round(vgpu_tok, 1),
-            },
+            "config": cfg,
         }), flush=True)
     if dist:
         dist.destroy_process_group()

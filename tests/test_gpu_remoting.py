@@ -146,6 +146,26 @@ def test_graph_decode_remoted(worker):
     assert r["tok_s"] > 0
 
 
+def test_graph_fused_decode_remoted(worker):
+    """The bench.py headline path through GPU-over-IP: hipGraph-captured
+    decode with the fused gfx950 rmsnorm/add_rmsnorm kernels (libtfops
+    launched via ctypes → hipLaunchKernel interposed and its code object
+    forwarded by the client stub). Proves the strongest native
+    configuration also runs remoted — the same-mode comparison bench.py
+    now defaults to."""
+
+    env = client_env(worker.socket_path)
+    env["TF_FUSED_OPS"] = "1"
+    out = subprocess.run(
+        [sys.executable, "-m", "tensor_fusion_amd.models.llama", "--model",
+         "tiny", "--batch", "2", "--ctx", "16", "--steps", "8", "--warmup",
+         "2", "--graphs"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=REPO)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-4000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["tok_s"] > 0
+
+
 def test_resnet50_inference_remoted(worker):
     """Conv workloads through GPU-over-IP: MIOpen loads its kernels via a
     different path than rocBLAS/Tensile — this catches interception gaps
