@@ -158,8 +158,13 @@ class Operator:
 def build_operator(persist_dir: Optional[str] = None,
                    metrics_dir: str = "",
                    config_path: Optional[str] = None,
-                   provider=None) -> Operator:
-    store = Store(persist_dir=persist_dir)
+                   provider=None, store: Optional[Store] = None) -> Operator:
+    """Assemble the control plane. `store` swaps the backing state plane:
+    the embedded Store (default) or a k8s.bridge.K8sStore, in which case
+    the whole operator runs against a real kube-apiserver (reference
+    cmd/main.go:131-297 manager wiring)."""
+
+    store = store or Store(persist_dir=persist_dir)
     quota = QuotaStore(store)
     allocator = GpuAllocator(store=store, quota=quota)
     gang = GangManager(store)
@@ -206,10 +211,43 @@ def main():
     ap.add_argument("--metrics-dir", default="")
     ap.add_argument("--config", default="")
     ap.add_argument("--http-port", type=int, default=C.OperatorHTTPPort)
+    # --- Kubernetes mode (reference cmd/main.go): the operator runs
+    # against a real apiserver instead of the embedded store ---
+    ap.add_argument("--k8s", action="store_true",
+                    help="back the control plane with a kube-apiserver "
+                         "(kubeconfig / in-cluster / TF_K8S_URL)")
+    ap.add_argument("--kubeconfig", default="")
+    ap.add_argument("--namespace", default="default",
+                    help="namespace watched for namespaced CRDs/pods")
+    ap.add_argument("--install-crds", action="store_true",
+                    help="apply the generated CRD manifests on startup")
+    ap.add_argument("--webhook-port", type=int, default=9443)
+    ap.add_argument("--webhook-certs", default="",
+                    help="dir with tls.crt/tls.key for the webhook server "
+                         "(generated with tensor_fusion_amd.k8s.certs)")
     args = ap.parse_args()
+
+    store = None
+    if args.k8s:
+        from .k8s.bridge import K8sStore
+        from .k8s.client import K8sClient
+        cli = (K8sClient.from_kubeconfig(args.kubeconfig)
+               if args.kubeconfig else K8sClient.auto())
+        if args.install_crds:
+            from .k8s.client import ApiError
+            from .k8s.crdgen import all_crds
+            for crd in all_crds().values():
+                try:
+                    cli.create(crd)
+                except ApiError as e:
+                    if not e.conflict:
+                        raise
+        store = K8sStore(cli, namespace=args.namespace).start()
+
     op = build_operator(persist_dir=args.persist_dir or None,
                         metrics_dir=args.metrics_dir,
-                        config_path=args.config or None)
+                        config_path=args.config or None,
+                        store=store)
     op.start()
 
     import uvicorn
@@ -219,6 +257,21 @@ def main():
                               port_allocator=op.port_allocator,
                               index_allocator=op.index_allocator,
                               expander=getattr(op, "expander", None))
+    if args.k8s:
+        # serve the AdmissionReview webhook the apiserver calls (TLS when
+        # certs are provided; MutatingWebhookConfiguration in deploy/)
+        from .server.webhook_server import create_webhook_app
+        whapp = create_webhook_app(op.mutator)
+        kw = {}
+        if args.webhook_certs:
+            import os
+            kw = {"ssl_certfile": os.path.join(args.webhook_certs, "tls.crt"),
+                  "ssl_keyfile": os.path.join(args.webhook_certs, "tls.key")}
+        wh = uvicorn.Server(uvicorn.Config(
+            whapp, host="0.0.0.0", port=args.webhook_port,
+            log_level="warning", **kw))
+        threading.Thread(target=wh.run, daemon=True,
+                         name="webhook-server").start()
     uvicorn.run(app, host="0.0.0.0", port=args.http_port, log_level="warning")
 
 

@@ -46,6 +46,15 @@ class PodMutator:
 
     def should_handle(self, pod: Pod) -> bool:
         self.counters["seen"] += 1
+        # never re-mutate the stack's own pods: on a real cluster worker/
+        # hypervisor pods pass through admission too (the controllers
+        # create them via the apiserver) and a worker inheriting the
+        # enabled label must not be turned into a client pod
+        # (reference pod_webhook.go skips component-labelled pods)
+        if pod.meta.labels.get(C.LabelComponent) in (
+                C.ComponentWorker, C.ComponentHypervisor,
+                C.ComponentOperator):
+            return False
         if pod.meta.labels.get(C.LabelEnabled) == "true":
             return True
         if not self.should_auto_migrate(pod):

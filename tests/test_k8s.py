@@ -1,0 +1,233 @@
+"""Kubernetes integration layer tests.
+
+BASELINE config 1 names "operator + gpuallocator on kind cluster with 8
+fake-GPU nodes"; this environment has no network so kind cannot run —
+the e2e here drives the same wire contracts (CRDs, AdmissionReview over
+HTTP, watch streams, binding subresource) against the wire-faithful fake
+apiserver (tensor_fusion_amd/k8s/fake_apiserver.py). K8sClient and the
+manifests in deploy/ are written for a real apiserver.
+"""
+import threading
+import time
+
+import pytest
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.api import types as T
+from tensor_fusion_amd.api.store import AlreadyExists, NotFound
+from tensor_fusion_amd.k8s import serde
+from tensor_fusion_amd.k8s.bridge import K8sStore
+from tensor_fusion_amd.k8s.client import ApiError, K8sClient
+from tensor_fusion_amd.k8s.crdgen import all_crds
+from tensor_fusion_amd.k8s.fake_apiserver import FakeApiServer, serve_in_thread
+
+
+@pytest.fixture(scope="module")
+def api():
+    srv, base, us = serve_in_thread()
+    cli = K8sClient(base)
+    # install the generated CRD manifests like a real deployment would
+    for crd in all_crds().values():
+        cli.create(crd)
+    yield srv, base
+    us.should_exit = True
+
+
+@pytest.fixture()
+def cli(api):
+    return K8sClient(api[1])
+
+
+def mk_gpu(name, node, pool="pool-a"):
+    g = T.GPU()
+    g.meta.name = name
+    g.status.uuid = f"uuid-{name}"
+    g.status.node = node
+    g.status.pool = pool
+    g.status.capacity = T.Resource(C.MI355X_BF16_TFLOPS,
+                                   C.MI355X_VRAM_BYTES, 100.0)
+    g.status.available = T.Resource(C.MI355X_BF16_TFLOPS,
+                                    C.MI355X_VRAM_BYTES, 100.0)
+    return g
+
+
+# ----------------------------------------------------------------- serde
+
+
+class TestSerde:
+    def test_gpu_round_trip(self):
+        g = mk_gpu("n0-g0", "n0")
+        g.status.topology = {"uuid-x": 0}
+        w = serde.to_k8s(g)
+        assert w["apiVersion"] == "tensor-fusion.ai/v1"
+        assert w["status"]["capacity"]["vram"] == C.MI355X_VRAM_BYTES
+        # data dict keys untouched by case conversion
+        assert "uuid-x" in w["status"]["topology"]
+        back = serde.from_k8s(w)
+        assert back.status.capacity.tflops == g.status.capacity.tflops
+        assert back.status.topology == {"uuid-x": 0}
+
+    def test_workload_nested_round_trip(self):
+        wl = T.TensorFusionWorkload()
+        wl.meta.name = "wl"
+        wl.meta.namespace = "default"
+        wl.profile.resources.requests.tflops = 600.0
+        wl.profile.gang.enabled = True
+        wl.profile.gang.min_members = 4
+        wl.replicas = 4
+        w = serde.to_k8s(wl)
+        assert w["spec"]["profile"]["gang"]["minMembers"] == 4
+        back = serde.from_k8s(w)
+        assert back.profile.gang.min_members == 4
+        assert back.profile.resources.requests.tflops == 600.0
+
+    def test_pod_round_trip_env_and_annotations(self):
+        p = T.Pod()
+        p.meta.name = "w"
+        p.meta.namespace = "default"
+        p.meta.annotations[C.AnnoGpuIds] = "uuid-a,uuid-b"
+        p.containers.append(T.Container(name="main",
+                                        env={"TF_SHM_PATH": "/x"}))
+        w = serde.to_k8s(p)
+        assert w["spec"]["containers"][0]["env"] == [
+            {"name": "TF_SHM_PATH", "value": "/x"}]
+        back = serde.from_k8s(w)
+        assert back.meta.annotations[C.AnnoGpuIds] == "uuid-a,uuid-b"
+        assert back.containers[0].env["TF_SHM_PATH"] == "/x"
+
+    def test_owner_reference_round_trip(self):
+        conn = T.TensorFusionConnection()
+        conn.meta.name = "c"
+        conn.meta.namespace = "default"
+        conn.meta.owner = "TensorFusionWorkload/default/wl"
+        w = serde.to_k8s(conn)
+        ref = w["metadata"]["ownerReferences"][0]
+        assert ref["kind"] == "TensorFusionWorkload"
+        back = serde.from_k8s(w)
+        assert back.meta.owner == "TensorFusionWorkload/default/wl"
+
+    def test_crd_manifests_shape(self):
+        crds = all_crds()
+        assert len(crds) == 12
+        gpu = crds["GPU"]
+        assert gpu["spec"]["scope"] == "Cluster"
+        v = gpu["spec"]["versions"][0]
+        assert v["subresources"] == {"status": {}}
+        props = v["schema"]["openAPIV3Schema"]["properties"]
+        assert props["status"]["properties"]["capacity"]["properties"][
+            "vram"]["type"] == "integer"
+        wl = crds["TensorFusionWorkload"]
+        assert wl["spec"]["scope"] == "Namespaced"
+
+
+# ---------------------------------------------------------------- client
+
+
+class TestClient:
+    def test_crud_and_conflict(self, cli):
+        g = mk_gpu("crud-g0", "crud-n0")
+        cli.create(serde.to_k8s(g))
+        got = cli.get("GPU", "crud-g0")
+        assert got["status"]["node"] == "crud-n0"
+        with pytest.raises(ApiError) as ei:
+            cli.create(serde.to_k8s(g))
+        assert ei.value.conflict
+        # stale-RV update conflicts
+        stale = dict(got)
+        stale["metadata"] = dict(got["metadata"],
+                                 resourceVersion="1")
+        with pytest.raises(ApiError) as ei:
+            cli.update(stale)
+        assert ei.value.conflict
+        cli.delete("GPU", "crud-g0")
+        assert cli.try_get("GPU", "crud-g0") is None
+
+    def test_label_selector_list(self, cli):
+        for i in range(3):
+            p = T.Pod()
+            p.meta.name = f"sel-{i}"
+            p.meta.namespace = "selns"
+            p.meta.labels["grp"] = "a" if i < 2 else "b"
+            cli.create(serde.to_k8s(p))
+        items = cli.list_items("Pod", "selns", label_selector="grp=a")
+        assert {i["metadata"]["name"] for i in items} == {"sel-0", "sel-1"}
+
+    def test_watch_replay_and_live(self, cli):
+        base_rv = cli.list("GPU")["metadata"]["resourceVersion"]
+        g = mk_gpu("watch-g0", "watch-n0")
+        cli.create(serde.to_k8s(g))
+        seen = []
+
+        def run():
+            for typ, obj in cli.watch("GPU", resource_version=base_rv,
+                                      timeout_s=15):
+                seen.append((typ, obj["metadata"]["name"]))
+                if len(seen) >= 2:
+                    return
+        t = threading.Thread(target=run, daemon=True)
+        t.start()
+        time.sleep(0.3)
+        cli.patch("GPU", "watch-g0", {"status": {"phase": "Migrating"}},
+                  subresource="status")
+        t.join(timeout=10)
+        assert ("ADDED", "watch-g0") in seen
+        assert ("MODIFIED", "watch-g0") in seen
+
+    def test_lease_leader_election(self, cli):
+        assert cli.acquire_lease("op-leader", "selns", "op-a") is True
+        assert cli.acquire_lease("op-leader", "selns", "op-b") is False
+        assert cli.acquire_lease("op-leader", "selns", "op-a") is True
+
+
+# ---------------------------------------------------------------- bridge
+
+
+class TestBridge:
+    def test_write_through_and_informer_merge(self, api):
+        cli = K8sClient(api[1])
+        store = K8sStore(cli, kinds=["GPU", "Pod"],
+                         namespace="bridge").start()
+        try:
+            store.create(mk_gpu("br-g0", "br-n0"))
+            # on the wire
+            assert cli.get("GPU", "br-g0")["status"]["node"] == "br-n0"
+            # read-after-write locally
+            assert store.get("GPU", "br-g0").status.node == "br-n0"
+            # duplicate create surfaces as AlreadyExists
+            with pytest.raises(AlreadyExists):
+                store.create(mk_gpu("br-g0", "br-n0"))
+            # external write flows in through the informer
+            cli.patch("GPU", "br-g0", {"status": {"phase": "Migrating"}},
+                      subresource="status")
+            for _ in range(100):
+                if store.get("GPU", "br-g0").status.phase == "Migrating":
+                    break
+                time.sleep(0.05)
+            assert store.get("GPU", "br-g0").status.phase == "Migrating"
+            # RMW patch pushes status subresource
+            store.patch("GPU", "br-g0", "",
+                        lambda o: setattr(o.status, "phase", "Ready"))
+            assert cli.get("GPU", "br-g0")["status"]["phase"] == "Ready"
+            # delete propagates and NotFound maps
+            store.delete("GPU", "br-g0")
+            with pytest.raises(NotFound):
+                store.delete("GPU", "br-g0")
+        finally:
+            store.stop()
+
+    def test_pod_bind_uses_binding_subresource(self, api):
+        cli = K8sClient(api[1])
+        store = K8sStore(cli, kinds=["Pod"], namespace="bridge").start()
+        try:
+            p = T.Pod()
+            p.meta.name = "bindme"
+            p.meta.namespace = "bridge"
+            store.create(p)
+            cur = store.get("Pod", "bindme", "bridge")
+            cur.status.node = "some-node"
+            cur.status.phase = "Scheduled"
+            store.update(cur)
+            wire = cli.get("Pod", "bindme", "bridge")
+            assert wire["spec"]["nodeName"] == "some-node"
+        finally:
+            store.stop()

@@ -1,0 +1,203 @@
+"""BASELINE config 1 — operator + gpuallocator on a (fake) cluster with
+8 fake-GPU nodes, end to end over the Kubernetes wire:
+
+    kubectl-equivalent client ──HTTP──▶ apiserver
+        pod CREATE ──AdmissionReview HTTP──▶ operator webhook (mutates)
+        watch streams ──▶ K8sStore informers ──▶ controllers/scheduler
+        scheduler binds via the pods/binding subresource
+        connection URL lands in the TensorFusionConnection CR status
+
+Everything the operator does here goes through REST + watch — no direct
+store access from the "user" side. (kind itself cannot run in this
+offline environment; the fake apiserver speaks the same wire, see
+tensor_fusion_amd/k8s/fake_apiserver.py.)
+
+Reference: cmd/main.go:131-297, webhook pod_webhook.go:84, scheduler
+PreBind gpuresources.go:882, connection controller :136.
+"""
+import socket
+import threading
+import time
+
+import pytest
+import uvicorn
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.api import types as T
+from tensor_fusion_amd.k8s import serde
+from tensor_fusion_amd.k8s.bridge import K8sStore
+from tensor_fusion_amd.k8s.client import K8sClient
+from tensor_fusion_amd.k8s.crdgen import all_crds
+from tensor_fusion_amd.k8s.fake_apiserver import serve_in_thread
+from tensor_fusion_amd.operator import build_operator
+from tensor_fusion_amd.server.webhook_server import create_webhook_app
+
+NODES = 8
+GPUS_PER_NODE = 8
+
+
+def _free_port() -> int:
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        return sk.getsockname()[1]
+
+
+@pytest.fixture(scope="module")
+def cluster():
+    """Fake apiserver + CRDs + 8 fake-GPU nodes + kube-backed operator
+    with its AdmissionReview webhook registered and served over HTTP."""
+
+    srv, base, us = serve_in_thread()
+    kubectl = K8sClient(base)
+    for crd in all_crds().values():
+        kubectl.create(crd)
+
+    # the "cluster": 8 nodes × 8 MI355X GPUs, published like the node
+    # plane would (Node + GPU CRs through the API)
+    pool = T.GPUPool()
+    pool.meta.name = "pool-a"
+    kubectl.create(serde.to_k8s(pool))
+    for n in range(NODES):
+        node = T.Node()
+        node.meta.name = f"node-{n}"
+        kubectl.create(serde.to_k8s(node))
+        for i in range(GPUS_PER_NODE):
+            g = T.GPU()
+            g.meta.name = f"node-{n}-g{i}"
+            g.status.uuid = f"uuid-{g.meta.name}"
+            g.status.node = f"node-{n}"
+            g.status.pool = "pool-a"
+            g.status.capacity = T.Resource(C.MI355X_BF16_TFLOPS,
+                                           C.MI355X_VRAM_BYTES, 100.0)
+            g.status.available = T.Resource(C.MI355X_BF16_TFLOPS,
+                                            C.MI355X_VRAM_BYTES, 100.0)
+            wire = serde.to_k8s(g)
+            created = kubectl.create(wire)
+            created["status"] = wire["status"]
+            kubectl.update_status(created)
+
+    # kube-backed operator
+    store = K8sStore(K8sClient(base), namespace="default").start()
+    op = build_operator(store=store)
+
+    # serve the AdmissionReview webhook and register it (url clientConfig,
+    # like a kind setup with host networking)
+    whport = _free_port()
+    whapp = create_webhook_app(op.mutator)
+    whsrv = uvicorn.Server(uvicorn.Config(whapp, host="127.0.0.1",
+                                          port=whport, log_level="error"))
+    threading.Thread(target=whsrv.run, daemon=True).start()
+    import requests
+    for _ in range(100):
+        try:
+            if requests.get(f"http://127.0.0.1:{whport}/healthz",
+                            timeout=1).ok:
+                break
+        except Exception:
+            time.sleep(0.05)
+    kubectl.create({
+        "apiVersion": "admissionregistration.k8s.io/v1",
+        "kind": "MutatingWebhookConfiguration",
+        "metadata": {"name": "tensor-fusion-mutating-webhook"},
+        "webhooks": [{
+            "name": "mpod.tensor-fusion.ai",
+            "clientConfig": {
+                "url": f"http://127.0.0.1:{whport}/mutate-v1-pod"},
+            "objectSelector": {
+                "matchLabels": {C.LabelEnabled: "true"}},
+        }],
+    })
+
+    yield kubectl, op
+    store.stop()
+    whsrv.should_exit = True
+    us.should_exit = True
+
+
+def _wait(fn, timeout=20.0, op=None):
+    deadline = time.time() + timeout
+    while time.time() < deadline:
+        if op is not None:
+            op.tick()
+        r = fn()
+        if r:
+            return r
+        time.sleep(0.05)
+    return fn()
+
+
+class TestKindStyleE2E:
+    def test_fractional_vgpu_pod_schedules_over_the_wire(self, cluster):
+        kubectl, op = cluster
+        # the user's pod: only labels + annotations, created via the API
+        pod = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {
+                "name": "app-1", "namespace": "default",
+                "labels": {C.LabelEnabled: "true"},
+                "annotations": {
+                    C.AnnoTflopsRequest: "600",
+                    C.AnnoVramRequest: str(48 << 30),
+                },
+            },
+            "spec": {"containers": [{"name": "main", "image": "app:1"}]},
+        }
+        kubectl.create(pod)
+
+        # webhook ran during admission: the stored pod is already mutated
+        # (client env injected; worker pods get the custom schedulerName)
+        stored = kubectl.get("Pod", "app-1", "default")
+        env = {e["name"]: e.get("value") for e in
+               stored["spec"]["containers"][0].get("env", [])}
+        assert C.EnvConnectionName in env
+        assert env[C.EnvConnectionName] == "app-1-conn"
+
+        # operator loops: workload → worker pod → schedule → bind
+        worker = _wait(lambda: next(
+            (p for p in kubectl.list_items(
+                "Pod", "default",
+                label_selector=f"{C.LabelComponent}={C.ComponentWorker}")
+             if p.get("spec", {}).get("nodeName")), None), op=op)
+        assert worker, "worker pod should be created and bound on the wire"
+        assert worker["spec"]["nodeName"].startswith("node-")
+        annos = worker["metadata"]["annotations"]
+        assert C.AnnoGpuIds in annos and annos[C.AnnoGpuIds]
+
+        # connection URL published in CR status on the wire
+        conn = _wait(lambda: (kubectl.try_get(
+            "TensorFusionConnection", "app-1-conn", "default") or {}
+        ).get("status", {}).get("connectionUrl"), op=op)
+        assert conn and conn.startswith("native+")
+
+        # allocator accounted the device on the wire (GPU CR status sync)
+        def allocated():
+            for i in range(GPUS_PER_NODE):
+                g = kubectl.get("GPU", f"{worker['spec']['nodeName']}-g{i}")
+                avail = g.get("status", {}).get("available", {})
+                if avail.get("tflops", C.MI355X_BF16_TFLOPS) <= \
+                        C.MI355X_BF16_TFLOPS - 600:
+                    return g
+            return None
+        g = _wait(allocated, op=op)
+        assert g is not None, "GPU CR status should show the allocation"
+
+    def test_workload_scale_down_releases_on_the_wire(self, cluster):
+        kubectl, op = cluster
+        wl = _wait(lambda: kubectl.try_get(
+            "TensorFusionWorkload", "app-1-wl", "default"), op=op)
+        assert wl
+        kubectl.patch("TensorFusionWorkload", "app-1-wl",
+                      {"spec": {"replicas": 0}}, namespace="default")
+        gone = _wait(lambda: not [
+            p for p in kubectl.list_items(
+                "Pod", "default",
+                label_selector=f"{C.LabelComponent}={C.ComponentWorker}")
+            if p["metadata"].get("labels", {}).get(C.LabelWorkload)
+            == "app-1-wl"], op=op)
+        assert gone, "workers should be deleted after scale to 0"
+
+    def test_eight_nodes_visible_to_allocator(self, cluster):
+        kubectl, op = cluster
+        assert len(op.allocator.gpus()) == NODES * GPUS_PER_NODE
+        nodes = {g.status.node for g in op.allocator.gpus()}
+        assert len(nodes) == NODES
