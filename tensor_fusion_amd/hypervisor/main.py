@@ -31,21 +31,70 @@ def build_hypervisor(node: str = "node-0", mock_devices: int = 0,
     return devices, workers, erl, backend
 
 
+def make_dp_resolver(store, node: str, workers):
+    """kubelet device-plugin Allocate() resolver: pod index on this node →
+    the composed worker environment (reference deviceplugin.go:250-338:
+    Allocate keys off the index resource to return the per-pod env +
+    device-node set)."""
+
+    from ..k8s.deviceplugin import default_device_nodes
+
+    def resolver(index: int):
+        for pod in store.list("Pod"):
+            if pod.meta.annotations.get(C.AnnoPodIndex) != str(index):
+                continue
+            if pod.status.node != node:
+                continue
+            alloc = workers.allocation_of(pod.meta.key) \
+                if hasattr(workers, "allocation_of") else None
+            if alloc is None:
+                return None
+            devices = []
+            for dn in alloc.device_nodes:
+                if dn == "/dev/dri":
+                    devices.extend(d for d in default_device_nodes()
+                                   if d["host_path"] != "/dev/kfd")
+                else:
+                    devices.append({"host_path": dn, "container_path": dn,
+                                    "permissions": "rw"})
+            return {
+                "env": dict(alloc.env),
+                "devices": devices,
+                "annotations": {C.AnnoPodIndex: str(index)},
+                "mounts": [{"host_path": C.DataRoot,
+                            "container_path": C.DataRoot}],
+            }
+        return None
+    return resolver
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--node", default="node-0")
-    ap.add_argument("--backend", default="single", choices=["single", "store"])
+    ap.add_argument("--backend", default="single",
+                    choices=["single", "store", "kubernetes"])
     ap.add_argument("--http-port", type=int, default=C.HypervisorHTTPPort)
     ap.add_argument("--mock-devices", type=int, default=0)
     ap.add_argument("--shm-root", default=C.ShmRoot)
     ap.add_argument("--store-dir", default="",
                     help="persist dir for store backend")
+    ap.add_argument("--device-plugin", action="store_true",
+                    help="register tensor-fusion.ai/index-N device "
+                         "plugins with kubelet (kubernetes backend)")
+    ap.add_argument("--kubelet-socket",
+                    default="/var/lib/kubelet/device-plugins/kubelet.sock")
     args = ap.parse_args()
 
     store = None
     if args.backend == "store":
         from ..api.store import Store
         store = Store(persist_dir=args.store_dir or None)
+    elif args.backend == "kubernetes":
+        # same StoreBackend logic, state plane = the kube-apiserver
+        # (reference pkg/hypervisor/backend/kubernetes)
+        from ..k8s.bridge import K8sStore
+        from ..k8s.client import K8sClient
+        store = K8sStore(K8sClient.auto()).start()
     devices, workers, erl, backend = build_hypervisor(
         node=args.node, mock_devices=args.mock_devices,
         shm_root=args.shm_root, store=store)
@@ -53,6 +102,20 @@ def main():
     erl.start()
     if backend:
         backend.start()
+
+    dp_mgr = None
+    ckpt = None
+    if args.backend == "kubernetes":
+        if args.device_plugin:
+            import os
+            from ..k8s.deviceplugin import DevicePluginManager
+            dp_mgr = DevicePluginManager(
+                make_dp_resolver(store, args.node, workers),
+                socket_dir=os.path.dirname(args.kubelet_socket),
+                kubelet_socket=args.kubelet_socket)
+            dp_mgr.start()
+        from ..k8s.kubelet_checkpoint import CheckpointDetector
+        ckpt = CheckpointDetector(store, node=args.node).start()
 
     # periodic shm sync / orphan sweep
     stop = threading.Event()

@@ -73,6 +73,13 @@ class WorkerController:
         with self._mu:
             return self.workers.get(key)
 
+    def allocation_of(self, key: str) -> Optional[WorkerAllocation]:
+        """Composed env/devices for a worker (device-plugin Allocate)."""
+
+        with self._mu:
+            st = self.workers.get(key)
+            return st.allocation if st else None
+
     def list(self) -> List[WorkerStatus]:
         with self._mu:
             return list(self.workers.values())
