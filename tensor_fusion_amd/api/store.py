@@ -123,10 +123,13 @@ class Store:
                 raise AlreadyExists(f"{kind} {key}")
             self._rv += 1
             obj.meta.resource_version = self._rv
+            # one deepcopy, not two: the stored copy is private to the
+            # store; the caller keeps its own object (same isolation,
+            # half the cost — admission throughput, tools/bench_webhook)
             bucket[key] = copy.deepcopy(obj)
             self._persist(kind)
             self._notify("ADDED", bucket[key])
-            return copy.deepcopy(bucket[key])
+            return obj
 
     def get(self, kind: str, name: str, namespace: str = "") -> TFObject:
         key = f"{namespace}/{name}" if namespace else name
@@ -153,12 +156,11 @@ class Store:
                 raise Conflict(f"{obj.kind} {key}: rv {obj.meta.resource_version} "
                                f"!= {cur.meta.resource_version}")
             self._rv += 1
-            obj = copy.deepcopy(obj)
             obj.meta.resource_version = self._rv
-            bucket[key] = obj
+            bucket[key] = copy.deepcopy(obj)
             self._persist(obj.kind)
-            self._notify("MODIFIED", obj)
-            return copy.deepcopy(obj)
+            self._notify("MODIFIED", bucket[key])
+            return obj
 
     def patch(self, kind: str, name: str, namespace: str,
               fn: Callable[[TFObject], None], retries: int = 8) -> TFObject:
@@ -171,10 +173,10 @@ class Store:
 
         with self._Mutate(self):
             obj = self.get(kind, name, namespace)
-            before = _to_dict(obj)
+            before = copy.deepcopy(obj)
             fn(obj)
-            if _to_dict(obj) == before:
-                return obj  # no-op patch: no rv bump, no event storm
+            if obj == before:  # dataclass field-wise eq, cheaper than
+                return obj     # asdict x2; no-op: no rv bump, no events
             return self.update(obj)
 
     def delete(self, kind: str, name: str, namespace: str = "") -> None:
