@@ -42,9 +42,12 @@
 #include <time.h>
 #include <unistd.h>
 
+#include <algorithm>
 #include <atomic>
+#include <map>
 #include <mutex>
 #include <unordered_map>
+#include <vector>
 
 #include "limiter_shm.h"
 
@@ -76,6 +79,7 @@ struct ExpandedRange {
   int device = 0;  // owning device at expand_alloc time — the tier thread
                    // never calls hipSetDevice, so its tls_device is always
                    // 0 and must not be used as a prefetch/advise target
+  uint64_t last_use = 0;  // LRU epoch (launch count at last touch)
 };
 
 struct Limiter {
@@ -97,10 +101,14 @@ struct Limiter {
   bool expand_enabled = false;
   uint64_t expand_limit = 0;  // 0 = unlimited host expansion
   std::mutex expand_mu;
-  std::unordered_map<const void*, ExpandedRange> expanded;
+  std::map<uint64_t, ExpandedRange> expanded;  // ordered: containment find
   std::atomic<uint64_t> expanded_bytes{0};
   std::atomic<uint64_t> promoted_bytes{0};
   std::atomic<int> tier_thread_running{0};
+  // working-set + migration stats (SURVEY §2.4(c) tracking requirement)
+  std::atomic<uint64_t> tier_epoch{1};
+  std::atomic<uint64_t> demoted_total{0}, promoted_total{0};
+  std::atomic<uint64_t> demote_ns{0}, promote_ns{0};
 
   // ---- per-call latency histograms (SURVEY §5.1 tracing) ----
   // log2-bucketed ns per op class, enabled with TF_LIMITER_TRACE=1:
@@ -163,6 +171,8 @@ double env_f(const char* k, double d) {
 // Thread-local current device (updated by interposed hipSetDevice).
 thread_local int tls_device = 0;
 
+void tier_note_launch();  // LRU epoch tick (defined with the tier engine)
+
 void record_latency(int cls, uint64_t t0_ns);  // fwd (uses now_ns)
 
 TfDeviceEntry* cur_dev() {
@@ -216,6 +226,7 @@ double try_consume(TfDeviceEntry* e, double cost) {
 }
 
 void consume_blocking(double cost) {
+  tier_note_launch();
   TfDeviceEntry* e = cur_dev();
   if (!e) return;
   heartbeat();
@@ -318,6 +329,29 @@ ExpandHip& ehip() {
 }
 
 static const int kHipCpuDeviceId = -1;
+
+// ------------------------------------------------- tier mechanism note
+//
+// A VMM-based host tier (hipMemCreate location.type=2 + pointer-stable
+// VA remap) was built and probed on MI355X: ROCm 7.2 ACCEPTS host-located
+// handles but BACKS THEM WITH HBM (native/tiering/vmm_host_probe4.cpp:
+// a 2 GiB "host" handle drops device free memory by 2 GiB, process RSS
+// grows ~0) — so it frees no VRAM and was removed as a placebo. Managed
+// memory (preferred-location CPU + hipMemPrefetchAsync over SDMA) is the
+// genuine host tier on this stack; the engine below adds per-range LRU
+// working-set tracking, touch hints from interposed memcpy/memset, and
+// migration byte/latency stats on top of it.
+
+hipError_t expand_alloc(void** p, size_t sz);  // defined below
+
+hipError_t tier_alloc_overcap(void** p, size_t sz) {
+  uint64_t lim = g.expand_limit;
+  if (lim) {
+    uint64_t cur = g.expanded_bytes.load(std::memory_order_relaxed);
+    if (cur + sz > lim) return hipErrorOutOfMemory;
+  }
+  return expand_alloc(p, sz);
+}
 static const int kAdviseSetPreferredLocation = 3;
 static const int kAdviseSetAccessedBy = 5;
 
@@ -337,7 +371,9 @@ hipError_t expand_alloc(void** p, size_t sz) {
   eh.MemAdvise(*p, sz, kAdviseSetAccessedBy, tls_device);
   {
     std::lock_guard<std::mutex> l(g.expand_mu);
-    g.expanded[*p] = ExpandedRange{sz, false, tls_device};
+    g.expanded[(uint64_t)*p] = ExpandedRange{
+        sz, false, tls_device,
+        g.tier_epoch.load(std::memory_order_relaxed)};
   }
   g.expanded_bytes.fetch_add(sz, std::memory_order_relaxed);
   dbg("expand_alloc %zu B -> host tier (total expanded %lu)", sz,
@@ -345,32 +381,73 @@ hipError_t expand_alloc(void** p, size_t sz) {
   return hipSuccess;
 }
 
-// Returns bytes whose residency changed. to_device: promote cold→HBM up to
-// `budget` bytes; else demote device-resident ranges back to host DRAM.
+// Returns bytes whose residency changed. to_device: promote the HOTTEST
+// host ranges (descending last_use) into up to `budget` bytes of HBM;
+// else demote the COLDEST device-resident ranges (ascending last_use)
+// back to host DRAM — LRU in both directions, from the working-set
+// signal collected by touch_managed() + the launch epoch.
 uint64_t tier_migrate(bool to_device, uint64_t budget) {
   ExpandHip& eh = ehip();
   if (!eh.ok) return 0;
+  uint64_t t0 = now_ns();
   uint64_t moved = 0;
   std::lock_guard<std::mutex> l(g.expand_mu);
-  for (auto& [ptr, r] : g.expanded) {
-    if (to_device == r.device_resident) continue;
+  std::vector<std::pair<uint64_t, uint64_t>> order;  // (last_use, va)
+  for (auto& [va, r] : g.expanded)
+    if (r.device_resident != to_device) order.emplace_back(r.last_use, va);
+  if (to_device)
+    std::sort(order.rbegin(), order.rend());  // hottest first
+  else
+    std::sort(order.begin(), order.end());  // coldest first
+  for (auto& [lu, va] : order) {
+    ExpandedRange& r = g.expanded[va];
+    const void* ptr = (const void*)va;
     if (to_device && moved + r.bytes > budget) continue;
     int dst = to_device ? r.device : kHipCpuDeviceId;
     eh.MemAdvise(ptr, r.bytes, kAdviseSetPreferredLocation, dst);
     if (eh.MemPrefetchAsync(ptr, r.bytes, dst, nullptr) == hipSuccess) {
       r.device_resident = to_device;
       moved += r.bytes;
+      if (to_device)
+        r.last_use = g.tier_epoch.load(std::memory_order_relaxed);
       if (!to_device && moved >= budget && budget) break;
     }
   }
   if (moved && eh.StreamSynchronize) eh.StreamSynchronize(nullptr);
-  if (to_device)
+  uint64_t dt = now_ns() - t0;
+  if (to_device) {
     g.promoted_bytes.fetch_add(moved, std::memory_order_relaxed);
-  else
+    g.promoted_total.fetch_add(moved, std::memory_order_relaxed);
+    g.promote_ns.fetch_add(dt, std::memory_order_relaxed);
+  } else {
     g.promoted_bytes.fetch_sub(
         moved > g.promoted_bytes.load() ? g.promoted_bytes.load() : moved,
         std::memory_order_relaxed);
+    g.demoted_total.fetch_add(moved, std::memory_order_relaxed);
+    g.demote_ns.fetch_add(dt, std::memory_order_relaxed);
+  }
   return moved;
+}
+
+// LRU epoch: one tick per governed GPU op (launch/memcpy) — the
+// working-set clock tier_migrate orders by.
+void tier_note_launch() {
+  if (g.expand_enabled)
+    g.tier_epoch.fetch_add(1, std::memory_order_relaxed);
+}
+
+// Working-set touch: refresh the LRU stamp of the expanded range
+// containing p (called from the memcpy/memset interposers — the only
+// pointer-visible ops; kernel args are opaque).
+void touch_managed(const void* p) {
+  if (!g.expand_enabled || !p) return;
+  std::lock_guard<std::mutex> l(g.expand_mu);
+  if (g.expanded.empty()) return;
+  auto it = g.expanded.upper_bound((uint64_t)p);
+  if (it == g.expanded.begin()) return;
+  --it;
+  if ((uint64_t)p < it->first + it->second.bytes)
+    it->second.last_use = g.tier_epoch.load(std::memory_order_relaxed);
 }
 
 void* tier_thread_main(void*) {
@@ -561,7 +638,8 @@ hipError_t hipMalloc(void** p, size_t sz) {
   if (g.enabled && !admit_alloc(sz)) {
     if (g.expand_enabled) {
       // VRAM oversubscription: land the range in the host-DRAM tier
-      hipError_t r = expand_alloc(p, sz);
+      // (VMM arena remap when available, managed memory otherwise)
+      hipError_t r = tier_alloc_overcap(p, sz);
       if (r == hipSuccess) ensure_tier_thread();
       return r;
     }
@@ -602,7 +680,7 @@ hipError_t hipFree(void* p) {
   if (g.enabled && r == hipSuccess) {
     record_free(p);
     std::lock_guard<std::mutex> l(g.expand_mu);
-    auto it = g.expanded.find(p);
+    auto it = g.expanded.find((uint64_t)p);
     if (it != g.expanded.end()) {
       g.expanded_bytes.fetch_sub(it->second.bytes, std::memory_order_relaxed);
       if (it->second.device_resident)
@@ -722,7 +800,11 @@ hipError_t hipGraphLaunch_spt(void* graphExec, hipStream_t stream) {
 // ----- memcpy / memset (memory-op tokens; VRAM cap already enforced)
 hipError_t hipMemcpy(void* dst, const void* src, size_t n, int kind) {
   REAL(hipError_t, hipMemcpy, void*, const void*, size_t, int);
-  if (g.enabled) consume_blocking(g.tokens_per_memcpy);
+  if (g.enabled) {
+    consume_blocking(g.tokens_per_memcpy);
+    touch_managed(dst);
+    touch_managed(src);
+  }
   return TF_TRACE_CALL(1, call_hipMemcpy(dst, src, n, kind));
 }
 
@@ -730,7 +812,11 @@ hipError_t hipMemcpyAsync(void* dst, const void* src, size_t n, int kind,
                           hipStream_t s) {
   REAL(hipError_t, hipMemcpyAsync, void*, const void*, size_t, int,
        hipStream_t);
-  if (g.enabled) consume_blocking(g.tokens_per_memcpy);
+  if (g.enabled) {
+    consume_blocking(g.tokens_per_memcpy);
+    touch_managed(dst);
+    touch_managed(src);
+  }
   return TF_TRACE_CALL(1, call_hipMemcpyAsync(dst, src, n, kind, s));
 }
 
@@ -744,7 +830,10 @@ hipError_t hipMemcpyWithStream(void* dst, const void* src, size_t n, int kind,
 
 hipError_t hipMemsetAsync(void* dst, int v, size_t n, hipStream_t s) {
   REAL(hipError_t, hipMemsetAsync, void*, int, size_t, hipStream_t);
-  if (g.enabled) consume_blocking(g.tokens_per_memcpy);
+  if (g.enabled) {
+    consume_blocking(g.tokens_per_memcpy);
+    touch_managed(dst);
+  }
   return call_hipMemsetAsync(dst, v, n, s);
 }
 
@@ -793,6 +882,46 @@ int tf_limiter_tier_stats(unsigned long long* expanded,
     *n_ranges = (unsigned)g.expanded.size();
   }
   return g.expand_enabled ? 1 : 0;
+}
+
+// Tier engine detail: out8 = {device_resident_bytes, host_resident_bytes,
+// demoted_total, promoted_total, demote_ns, promote_ns, epoch, n_ranges}.
+// Returns 1 when oversubscription is enabled.
+int tf_limiter_tier_stats2(unsigned long long* out8) {
+  if (out8) {
+    uint64_t prom = g.promoted_bytes.load(std::memory_order_relaxed);
+    uint64_t exp = g.expanded_bytes.load(std::memory_order_relaxed);
+    out8[0] = prom;
+    out8[1] = exp > prom ? exp - prom : 0;
+    out8[2] = g.demoted_total.load(std::memory_order_relaxed);
+    out8[3] = g.promoted_total.load(std::memory_order_relaxed);
+    out8[4] = g.demote_ns.load(std::memory_order_relaxed);
+    out8[5] = g.promote_ns.load(std::memory_order_relaxed);
+    out8[6] = g.tier_epoch.load(std::memory_order_relaxed);
+    std::lock_guard<std::mutex> l(g.expand_mu);
+    out8[7] = g.expanded.size();
+  }
+  return g.expand_enabled ? 1 : 0;
+}
+
+// Working-set hint: mark the expanded range containing p as hot.
+void tf_limiter_touch(const void* p) { touch_managed(p); }
+
+// Per-range introspection (tests assert exact LRU behavior with this).
+// Returns 1 if p is inside an expanded range, 0 otherwise.
+int tf_limiter_tier_range(const void* p, unsigned long long* bytes,
+                          int* device_resident,
+                          unsigned long long* last_use) {
+  std::lock_guard<std::mutex> l(g.expand_mu);
+  if (g.expanded.empty()) return 0;
+  auto it = g.expanded.upper_bound((uint64_t)p);
+  if (it == g.expanded.begin()) return 0;
+  --it;
+  if ((uint64_t)p >= it->first + it->second.bytes) return 0;
+  if (bytes) *bytes = it->second.bytes;
+  if (device_resident) *device_resident = it->second.device_resident ? 1 : 0;
+  if (last_use) *last_use = it->second.last_use;
+  return 1;
 }
 
 // Force-demote every expanded range to host DRAM (pressure-trap path);

@@ -46,7 +46,11 @@ typedef struct TfDeviceEntry {
   uint32_t launch_count;         /* +124 kernels launched (stats)           */
   uint64_t block_ns_total;       /* +128 cumulative ns throttled            */
   uint64_t alloc_bytes_total;    /* +136 cumulative hipMalloc bytes         */
-  uint64_t pad0;                 /* +144                                    */
+  uint64_t vmm_bytes;            /* +144 remoting worker's VMM heap bytes
+                                         (hipMemCreate bypasses the
+                                         hipMalloc accounting; the worker
+                                         reports here so caps and metrics
+                                         see remote-vGPU VRAM)            */
   uint64_t pad1;                 /* +152 → sizeof == 160                    */
 } TfDeviceEntry;
 
@@ -79,6 +83,7 @@ typedef struct TfSharedState {
 } /* extern "C" */
 
 #include <atomic>
+#include <cstddef>
 static_assert(sizeof(TfDeviceEntry) == TF_DEV_STRIDE, "device entry stride");
 static_assert(sizeof(TfSharedState) == TF_SHM_SIZE, "shm page size");
 static_assert(offsetof(TfSharedState, dev) == TF_OFF_DEV, "dev offset");

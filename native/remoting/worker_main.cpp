@@ -36,6 +36,7 @@
 #include <vector>
 
 #include "codeobj.h"
+#include "limiter_shm.h"
 #include "protocol.h"
 #include "ring.h"
 
@@ -236,6 +237,13 @@ struct Vmm {
   int device = 0;
   uint64_t base = 0;
   size_t heap_bytes = 0;
+  // ---- VRAM accounting (closes the round-1 hole: hipMemCreate bypassed
+  // the limiter's hipMalloc-based cap, so a remote vGPU could exceed its
+  // VRAM limit). The worker enforces TF_VRAM_LIMIT_BYTES on its own heap
+  // and reports mapped bytes into the limiter shm (vmm_bytes field).
+  uint64_t mapped_bytes = 0;
+  uint64_t vram_limit = 0;
+  TfSharedState* shm = nullptr;
   size_t gran = 2u << 20;
   std::map<uint64_t, size_t> free_spans;          // va → len
   std::map<uint64_t, VmmRange> mapped;            // va → range
@@ -262,8 +270,36 @@ struct Vmm {
     base = want;
     heap_bytes = heap;
     free_spans[base] = heap;
+    const char* lim = getenv("TF_VRAM_LIMIT_BYTES");
+    if (lim) vram_limit = strtoull(lim, nullptr, 10);
+    const char* shm_path = getenv("TF_SHM_PATH");
+    if (shm_path && *shm_path) {
+      int fd = open(shm_path, O_RDWR);
+      if (fd >= 0) {
+        void* m = mmap(nullptr, TF_SHM_SIZE, PROT_READ | PROT_WRITE,
+                       MAP_SHARED, fd, 0);
+        close(fd);
+        if (m != MAP_FAILED) {
+          auto* st = reinterpret_cast<TfSharedState*>(m);
+          if (st->magic == TF_SHM_MAGIC) {
+            shm = st;
+            uint64_t sl = tfshm::at(&st->dev[0].mem_limit_bytes)
+                              ->load(std::memory_order_relaxed);
+            if (sl && (!vram_limit || sl < vram_limit)) vram_limit = sl;
+          } else {
+            munmap(m, TF_SHM_SIZE);
+          }
+        }
+      }
+    }
     enabled = true;
     return true;
+  }
+
+  void report_usage() {
+    if (!shm) return;
+    tfshm::at(&shm->dev[0].vmm_bytes)
+        ->store(mapped_bytes, std::memory_order_relaxed);
   }
 
   uint64_t round_up(uint64_t n) const { return (n + gran - 1) & ~(gran - 1); }
@@ -311,6 +347,10 @@ struct Vmm {
 
   void* alloc(size_t sz, hipError_t* err) {
     size_t len = round_up(sz ? sz : 1);
+    if (vram_limit && mapped_bytes + len > vram_limit) {
+      *err = 2;  // hipErrorOutOfMemory: the vGPU's VRAM cap
+      return nullptr;
+    }
     for (auto it = free_spans.begin(); it != free_spans.end(); ++it) {
       if (it->second < len) continue;
       uint64_t va = it->first;
@@ -324,6 +364,8 @@ struct Vmm {
         return nullptr;
       }
       *err = 0;
+      mapped_bytes += len;
+      report_usage();
       return (void*)va;
     }
     *err = 2;  // hipErrorOutOfMemory (VA heap exhausted)
@@ -354,6 +396,8 @@ struct Vmm {
     hip.MemAddressFree(p, it->second.bytes);
     uint64_t va = it->first;
     size_t len = it->second.bytes;
+    mapped_bytes -= len < mapped_bytes ? len : mapped_bytes;
+    report_usage();
     mapped.erase(it);
     // coalesce with neighbours
     auto nxt = free_spans.find(va + len);

@@ -132,3 +132,117 @@ def test_graph_decode_native_matches_eager():
     eager = run(False)
     graphed = run(True)
     assert torch.equal(eager, graphed), (eager.flatten(), graphed.flatten())
+
+
+# ---------------------------------------------------------------------------
+# Tier engine: LRU working-set migration + stats on the managed-memory
+# host tier (hip_limiter.cpp). A VMM host tier was probed and REJECTED —
+# ROCm 7.2 backs "host-located" handles with HBM (vmm_host_probe4.cpp),
+# so managed memory + SDMA prefetch is the genuine mechanism.
+
+LRU_CHILD = r"""
+import ctypes, json, sys
+import torch
+torch.cuda.init()
+
+lim = ctypes.CDLL(None)
+lim.tf_limiter_tier_stats2.argtypes = [ctypes.POINTER(ctypes.c_ulonglong)]
+lim.tf_limiter_touch.argtypes = [ctypes.c_void_p]
+lim.tf_limiter_promote.restype = ctypes.c_ulonglong
+lim.tf_limiter_promote.argtypes = [ctypes.c_ulonglong]
+lim.tf_limiter_demote_all.restype = ctypes.c_ulonglong
+lim.tf_limiter_tier_range.restype = ctypes.c_int
+lim.tf_limiter_tier_range.argtypes = [
+    ctypes.c_void_p, ctypes.POINTER(ctypes.c_ulonglong),
+    ctypes.POINTER(ctypes.c_int), ctypes.POINTER(ctypes.c_ulonglong)]
+
+def rng(t):
+    b = ctypes.c_ulonglong(); d = ctypes.c_int(); lu = ctypes.c_ulonglong()
+    ok = lim.tf_limiter_tier_range(ctypes.c_void_p(t.data_ptr()),
+                                   ctypes.byref(b), ctypes.byref(d),
+                                   ctypes.byref(lu))
+    return ok, d.value, lu.value
+
+def stats():
+    out = (ctypes.c_ulonglong * 8)()
+    lim.tf_limiter_tier_stats2(out)
+    return list(out)
+
+MB = 1 << 20
+# fill most of the 1 GiB cap, then three over-cap ranges -> host tier
+pad = torch.zeros(int(0.8 * (1 << 30)) // 4, device="cuda")
+a = torch.full((300 * MB // 4,), 1.0, device="cuda")
+b = torch.full((300 * MB // 4,), 2.0, device="cuda")
+c = torch.full((300 * MB // 4,), 3.0, device="cuda")
+torch.cuda.synchronize()
+for t in (a, b, c):
+    ok, dev, _ = rng(t)
+    assert ok == 1 and dev == 0, (ok, dev)
+
+# heat: touch a then c (c is hottest; b never touched)
+lim.tf_limiter_touch(ctypes.c_void_p(a.data_ptr()))
+_ = float(a.sum())  # ticks the epoch via governed ops
+lim.tf_limiter_touch(ctypes.c_void_p(c.data_ptr()))
+
+# promote with budget for ONE range -> must pick c (hottest)
+lim.tf_limiter_promote(ctypes.c_ulonglong(310 * MB))
+ra, rb, rc = rng(a), rng(b), rng(c)
+assert rc[1] == 1, (ra, rb, rc)
+assert ra[1] == 0 and rb[1] == 0, (ra, rb, rc)
+
+# promote one more -> a (b is coldest, stays)
+lim.tf_limiter_promote(ctypes.c_ulonglong(310 * MB))
+ra, rb, rc = rng(a), rng(b), rng(c)
+assert ra[1] == 1 and rc[1] == 1 and rb[1] == 0, (ra, rb, rc)
+
+# pressure demote of everything; data must be intact afterwards
+lim.tf_limiter_demote_all()
+torch.cuda.synchronize()
+assert float(a.sum()) == a.numel() * 1.0
+assert float(b.sum()) == b.numel() * 2.0
+assert float(c.sum()) == c.numel() * 3.0
+s = stats()
+bw_demote = s[2] / max(s[4], 1)   # bytes per ns == GB/s
+bw_promote = s[3] / max(s[5], 1)
+print(json.dumps({"ok": True, "demoted_total": s[2], "promoted_total": s[3],
+                  "demote_gbps": round(bw_demote, 2),
+                  "promote_gbps": round(bw_promote, 2),
+                  "n_ranges": s[7]}))
+"""
+
+
+def _limiter_env(cap_bytes: int, extra=None):
+    import os
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    env = dict(os.environ)
+    env.update({
+        "LD_PRELOAD": os.path.join(repo, "tensor_fusion_amd", "_native",
+                                   "libtfhip_limiter.so"),
+        "TF_VRAM_LIMIT_BYTES": str(cap_bytes),
+        "TF_VRAM_EXPAND": "1",
+        "TF_LIMITER_DEBUG": "1",
+    })
+    env.pop("TF_SHM_PATH", None)
+    env.update(extra or {})
+    return env
+
+
+def test_tier_lru_promotes_hottest_first():
+    """LRU working-set migration: with budget for one range, promote must
+    pick the most recently touched range; untouched ranges stay cold;
+    pressure demote preserves data. Migration bandwidth reported from the
+    engine's own byte/ns counters (tf_limiter_tier_stats2)."""
+
+    import json
+    import subprocess
+    import sys
+    out = subprocess.run(
+        [sys.executable, "-c", LRU_CHILD],
+        env=_limiter_env(1 << 30), capture_output=True, text=True,
+        timeout=600)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-4000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["ok"] and r["demoted_total"] >= 600 * (1 << 20)
+    assert r["promoted_total"] >= 600 * (1 << 20)
+    print(f"\ntier LRU: demote {r['demote_gbps']} GB/s, "
+          f"promote {r['promote_gbps']} GB/s over SDMA prefetch")
