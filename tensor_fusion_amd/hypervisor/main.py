@@ -23,7 +23,15 @@ def build_hypervisor(node: str = "node-0", mock_devices: int = 0,
     devices = DeviceController(accel)
     alloc = AllocationController(devices, shm_root=shm_root)
     erl = ErlQuotaController(devices)
-    workers = WorkerController(devices, alloc, erl=erl, shm_root=shm_root)
+    from .pressure import PressureController
+
+    def _free_fn():
+        m = devices.metrics(0)
+        total = m.vram_total or C.MI355X_VRAM_BYTES
+        return max(total - m.vram_used, 0), total
+    pressure = PressureController(_free_fn)
+    workers = WorkerController(devices, alloc, erl=erl, shm_root=shm_root,
+                               pressure=pressure)
     backend = None
     if store is not None:
         from .backend.store_backend import StoreBackend
@@ -100,6 +108,7 @@ def main():
         shm_root=args.shm_root, store=store)
     devices.start()
     erl.start()
+    workers.pressure.start()
     if backend:
         backend.start()
 

@@ -51,6 +51,7 @@ E_ACTIVE = 120
 E_LAUNCHES = 124
 E_BLOCK_NS = 128
 E_ALLOC_BYTES = 136
+E_VMM_BYTES = 144  # remoting worker's VMM heap bytes (limiter_shm.h)
 
 FLAG_FREEZE = 1 << 0
 FLAG_VRAM_PRESSURE = 1 << 1
@@ -75,6 +76,11 @@ class DeviceEntrySnapshot:
     launch_count: int
     block_ns_total: int
     alloc_bytes_total: int
+    vmm_bytes: int = 0  # remoting worker VMM heap (bypasses hipMalloc)
+
+    @property
+    def total_used(self) -> int:
+        return self.pod_memory_used + self.vmm_bytes
 
 
 class WorkerShm:
@@ -196,6 +202,7 @@ class WorkerShm:
             launch_count=self.read_u32(off + E_LAUNCHES),
             block_ns_total=self.read_u64(off + E_BLOCK_NS),
             alloc_bytes_total=self.read_u64(off + E_ALLOC_BYTES),
+            vmm_bytes=self.read_u64(off + E_VMM_BYTES),
         )
 
     def devices(self) -> List[DeviceEntrySnapshot]:

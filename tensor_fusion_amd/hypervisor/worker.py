@@ -34,10 +34,12 @@ class WorkerController:
     def __init__(self, devices: DeviceController,
                  allocator: AllocationController,
                  erl: Optional[ErlQuotaController] = None,
-                 shm_root: str = C.ShmRoot):
+                 shm_root: str = C.ShmRoot,
+                 pressure=None):
         self.devices = devices
         self.allocator = allocator
         self.erl = erl
+        self.pressure = pressure  # hypervisor.pressure.PressureController
         self.shm_root = shm_root
         self.workers: Dict[str, WorkerStatus] = {}
         self._mu = threading.RLock()
@@ -55,6 +57,10 @@ class WorkerController:
             self.workers[spec.key] = st
             if self.erl:
                 self.erl.attach(page)
+            if self.pressure:
+                self.pressure.attach(
+                    page, qos=spec.qos,
+                    provisioned_bytes=spec.vram_limit)
             return st
 
     def remove_worker(self, key: str, delete_shm: bool = True):
@@ -64,6 +70,8 @@ class WorkerController:
                 return
             if self.erl:
                 self.erl.detach(st.shm.path)
+            if self.pressure:
+                self.pressure.detach(st.shm.path)
             self.allocator.deallocate(key)
             st.shm.close()
             if delete_shm:

@@ -246,3 +246,33 @@ def test_tier_lru_promotes_hottest_first():
     assert r["promoted_total"] >= 600 * (1 << 20)
     print(f"\ntier LRU: demote {r['demote_gbps']} GB/s, "
           f"promote {r['promote_gbps']} GB/s over SDMA prefetch")
+
+
+def test_multi_tenant_oversubscription_with_pressure_controller():
+    """BASELINE config 4 (scaled shape): N tenants whose caps
+    oversubscribe the device run concurrently; each tenant's over-budget
+    slabs live in the host tier, the hypervisor PressureController
+    assigns budgets/flags, nobody OOMs, and the per-tenant spread stays
+    tight. Full 4x96 GB shape: tools/demo_oversub.py (see
+    profiles/oversub_4x96_r02.md)."""
+
+    import json
+    import os
+    import subprocess
+    import sys
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "tools", "demo_oversub.py"),
+         "--scaled", "--shm-root", "/tmp/tf-testdemo-shm"],
+        capture_output=True, text=True, timeout=900)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-3000:]
+    line = next(l for l in out.stdout.splitlines()
+                if l.startswith("SUMMARY "))
+    r = json.loads(line[len("SUMMARY "):])
+    assert not r["failures"], r["failures"]
+    assert len(r["tenant_it_s"]) == r["config"]["tenants"]
+    assert all(t > 0 for t in r["tenant_it_s"])
+    # every tenant exceeded its HBM budget and used the host tier
+    assert all(t["host_res"] > 0 for t in r["tiers"])
+    # fairness between equal-QoS tenants
+    assert r["spread_pct"] < 25.0, r
