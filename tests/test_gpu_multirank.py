@@ -1,0 +1,120 @@
+"""Multi-rank validation on ONE MI355X (VERDICT round-2 item #4).
+
+The builder pool exposes single-GPU boxes; the driver runs the real
+1/2/4/8 scaling bench. Measured constraint: RCCL 2.26.6 REFUSES two
+ranks on one device ("Duplicate GPU detected", ncclInvalidUsage), so
+2-rank RCCL-on-1-GPU is not possible on this stack. De-risking is
+therefore split:
+  - world=1 nccl(=RCCL) init + collective: the RCCL library path
+    bench.py/TPLlama use actually initializes and runs on hardware
+  - 2-rank TP decode with device tensors over the gloo backend: the
+    full cross-process TP graph (sharded weights, per-block collectives,
+    kv-cache decode) runs on the GPU; only the transport differs
+  - the same under the LD_PRELOAD limiter (gang-scheduled vGPU shape)
+What remains for the driver's 8-GPU run is RCCL fan-out over xGMI.
+"""
+import json
+import os
+import socket
+import subprocess
+import sys
+
+import pytest
+
+pytestmark = pytest.mark.gpu
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def _free_port() -> int:
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        return sk.getsockname()[1]
+
+
+def _spawn_ranks(world: int, mode: str, extra_env=None, timeout=600,
+                 model: str = "tiny"):
+    port = _free_port()
+    procs = []
+    for rank in range(world):
+        env = dict(os.environ)
+        env.update({
+            "RANK": str(rank), "LOCAL_RANK": str(rank),
+            "WORLD_SIZE": str(world),
+            "MASTER_ADDR": "127.0.0.1", "MASTER_PORT": str(port),
+            "HIP_VISIBLE_DEVICES": "0",  # both ranks share ONE GPU
+            "TF_TP_MODE": mode, "TF_TP_MODEL": model,
+            "TF_REPO": REPO,
+            "HSA_ENABLE_IPC_MODE_LEGACY": "0",
+        })
+        env.update(extra_env or {})
+        procs.append(subprocess.Popen(
+            [sys.executable, "-m",
+             "tensor_fusion_amd.parallel._tp_gpu_worker"],
+            env=env, cwd=REPO, stdout=subprocess.PIPE,
+            stderr=subprocess.PIPE, text=True))
+    outs = []
+    for p in procs:
+        try:
+            out, err = p.communicate(timeout=timeout)
+        except subprocess.TimeoutExpired:
+            for q in procs:
+                q.kill()
+            raise
+        outs.append((p.returncode, out, err))
+    return outs
+
+
+def _rank0_json(outs):
+    assert all(rc == 0 for rc, _, _ in outs), \
+        "\n".join(o[-1500:] + e[-2500:] for _, o, e in outs)
+    line = [l for l in outs[0][1].splitlines() if l.startswith("{")][-1]
+    return json.loads(line)
+
+
+def test_rccl_world1_init_and_allreduce():
+    """backend nccl (=RCCL) initializes and runs collectives on MI355X
+    (world=1 — RCCL 2.26 refuses >1 rank per device, see module doc)."""
+
+    r = _rank0_json(_spawn_ranks(1, "allreduce"))
+    assert r["ok"] is True and r["world"] == 1
+
+
+def test_tp2_allreduce_gloo_device_tensors():
+    """2 ranks sharing device 0: all-reduce of CUDA tensors across
+    processes (gloo transport; the RCCL refusal is environmental)."""
+
+    r = _rank0_json(_spawn_ranks(2, "allreduce",
+                                 extra_env={"TF_TP_BACKEND": "gloo"}))
+    assert r["ok"] is True and r["world"] == 2
+
+
+def test_tp2_decode_one_gpu():
+    """TP=2 Llama decode on one GPU (gloo transport, device tensors):
+    sharded weights reproduce the single-rank reference logits, then
+    timed decode steps run with two collectives per block (the config-5
+    workload shape; RCCL transport needs >1 physical GPU, driver run)."""
+
+    r = _rank0_json(_spawn_ranks(2, "decode", timeout=900,
+                                 extra_env={"TF_TP_BACKEND": "gloo"}))
+    assert r["ok"] is True, r
+    assert r["tok_s"] > 0
+
+
+def test_tp2_decode_under_vgpu_limiter():
+    """The gang-scheduled TP story end-to-end on one box: both TP ranks
+    run as local vGPUs under the LD_PRELOAD limiter (100% quota). The
+    limiter charges tokens at launch granularity only, so the RCCL
+    collectives inside the block are never split mid-algorithm
+     — decode must produce the same numerics."""
+
+    lim = os.path.join(REPO, "tensor_fusion_amd", "_native",
+                       "libtfhip_limiter.so")
+    r = _rank0_json(_spawn_ranks(2, "decode", extra_env={
+        "TF_TP_BACKEND": "gloo",
+        "LD_PRELOAD": lim,
+        "TF_UP_LIMIT_PERCENT": "100",
+        "TF_VRAM_LIMIT_BYTES": str(64 << 30),
+    }, timeout=900))
+    assert r["ok"] is True, r
+    assert r["tok_s"] > 0
