@@ -64,6 +64,8 @@ typedef status_t (*fn_get_activity)(processor_handle, engine_usage_t*);
 typedef status_t (*fn_get_proc_list)(processor_handle, uint32_t*, proc_info_t*);
 typedef status_t (*fn_get_bdf_id)(processor_handle, uint64_t*);
 typedef status_t (*fn_numa_affinity)(processor_handle, int32_t*);
+typedef status_t (*fn_get_str)(processor_handle, char*, uint32_t);
+typedef status_t (*fn_set_u32)(processor_handle, uint32_t);
 
 struct Api {
   void* handle = nullptr;
@@ -78,6 +80,11 @@ struct Api {
   fn_get_proc_list get_proc_list = nullptr;
   fn_get_bdf_id get_bdf_id = nullptr;
   fn_numa_affinity numa_affinity = nullptr;
+  // compute/memory partition APIs (optional — older amd-smi lacks them)
+  fn_get_str get_compute_partition = nullptr;
+  fn_set_u32 set_compute_partition = nullptr;
+  fn_get_str get_memory_partition = nullptr;
+  fn_set_u32 set_memory_partition = nullptr;
 
   bool load() {
     const char* names[] = {"libamd_smi.so", "libamd_smi.so.26"};
@@ -102,6 +109,14 @@ struct Api {
 #undef R
     numa_affinity = reinterpret_cast<fn_numa_affinity>(
         dlsym(handle, "amdsmi_get_gpu_topo_numa_affinity"));
+    get_compute_partition = reinterpret_cast<fn_get_str>(
+        dlsym(handle, "amdsmi_get_gpu_compute_partition"));
+    set_compute_partition = reinterpret_cast<fn_set_u32>(
+        dlsym(handle, "amdsmi_set_gpu_compute_partition"));
+    get_memory_partition = reinterpret_cast<fn_get_str>(
+        dlsym(handle, "amdsmi_get_gpu_memory_partition"));
+    set_memory_partition = reinterpret_cast<fn_set_u32>(
+        dlsym(handle, "amdsmi_set_gpu_memory_partition"));
     return true;
   }
 };
@@ -123,6 +138,9 @@ struct State {
   std::vector<TfAccelDevice> devices;
   tf_accel_log_fn log_cb = nullptr;
   std::mutex mu;
+  // mock backend: per-device partition modes (defaults = whole-GPU)
+  std::vector<std::string> mock_compute_mode;
+  std::vector<std::string> mock_memory_mode;
 };
 
 State& S() {
@@ -151,6 +169,23 @@ void fill_mock_devices(int n) {
     d.is_mock = 1;
     s.devices.push_back(d);
   }
+  s.mock_compute_mode.assign(n, "SPX");
+  s.mock_memory_mode.assign(n, "NPS1");
+}
+
+// amdsmi enum values (amdsmi.h): SPX=1 DPX=2 TPX=3 QPX=4 CPX=5;
+// NPS1=1 NPS2=2 NPS4=3 NPS8=4.
+int compute_mode_enum(const char* m) {
+  static const char* names[] = {"", "SPX", "DPX", "TPX", "QPX", "CPX"};
+  for (int i = 1; i <= 5; ++i)
+    if (!strcasecmp(m, names[i])) return i;
+  return 0;
+}
+int memory_mode_enum(const char* m) {
+  static const char* names[] = {"", "NPS1", "NPS2", "NPS4", "NPS8"};
+  for (int i = 1; i <= 4; ++i)
+    if (!strcasecmp(m, names[i])) return i;
+  return 0;
 }
 
 }  // namespace
@@ -353,6 +388,68 @@ int tf_accel_remove_partition(int device, const int32_t* xcds, int n_xcds) {
   if (!s.initialized || device < 0 || device >= (int)s.devices.size())
     return TF_ACCEL_ERR;
   return TF_ACCEL_OK;
+}
+
+int tf_accel_get_compute_partition(int device, char* out, int out_len) {
+  auto& s = S();
+  if (!s.initialized || device < 0 || device >= (int)s.devices.size() ||
+      !out || out_len < 4)
+    return TF_ACCEL_ERR;
+  if (s.mock) {
+    snprintf(out, out_len, "%s", s.mock_compute_mode[device].c_str());
+    return TF_ACCEL_OK;
+  }
+  if (!s.api.get_compute_partition) return TF_ACCEL_NOT_SUPPORTED;
+  return s.api.get_compute_partition(s.procs[device], out,
+                                     (uint32_t)out_len) == 0
+             ? TF_ACCEL_OK
+             : TF_ACCEL_ERR;
+}
+
+int tf_accel_set_compute_partition(int device, const char* mode) {
+  auto& s = S();
+  int e = compute_mode_enum(mode ? mode : "");
+  if (!s.initialized || device < 0 || device >= (int)s.devices.size() || !e)
+    return TF_ACCEL_ERR;
+  if (s.mock) {
+    s.mock_compute_mode[device] = mode;
+    return TF_ACCEL_OK;
+  }
+  if (!s.api.set_compute_partition) return TF_ACCEL_NOT_SUPPORTED;
+  return s.api.set_compute_partition(s.procs[device], (uint32_t)e) == 0
+             ? TF_ACCEL_OK
+             : TF_ACCEL_ERR;
+}
+
+int tf_accel_get_memory_partition(int device, char* out, int out_len) {
+  auto& s = S();
+  if (!s.initialized || device < 0 || device >= (int)s.devices.size() ||
+      !out || out_len < 4)
+    return TF_ACCEL_ERR;
+  if (s.mock) {
+    snprintf(out, out_len, "%s", s.mock_memory_mode[device].c_str());
+    return TF_ACCEL_OK;
+  }
+  if (!s.api.get_memory_partition) return TF_ACCEL_NOT_SUPPORTED;
+  return s.api.get_memory_partition(s.procs[device], out,
+                                    (uint32_t)out_len) == 0
+             ? TF_ACCEL_OK
+             : TF_ACCEL_ERR;
+}
+
+int tf_accel_set_memory_partition(int device, const char* mode) {
+  auto& s = S();
+  int e = memory_mode_enum(mode ? mode : "");
+  if (!s.initialized || device < 0 || device >= (int)s.devices.size() || !e)
+    return TF_ACCEL_ERR;
+  if (s.mock) {
+    s.mock_memory_mode[device] = mode;
+    return TF_ACCEL_OK;
+  }
+  if (!s.api.set_memory_partition) return TF_ACCEL_NOT_SUPPORTED;
+  return s.api.set_memory_partition(s.procs[device], (uint32_t)e) == 0
+             ? TF_ACCEL_OK
+             : TF_ACCEL_ERR;
 }
 
 int tf_accel_snapshot(int pid, const char* dest_dir) {

@@ -84,6 +84,17 @@ int tf_accel_compose_percent_mask_env(int device, double percent,
 
 /* partition: validate an XCD-slab partition on the device (slot accounting
  * lives in the allocator; this checks device capability). */
+/* AMD compute-partition modes (SPX/DPX/TPX/QPX/CPX) and memory modes
+ * (NPS1/NPS2/NPS4/NPS8) — the MI355X device-global partitioning the
+ * reference models per-vendor (partition_strategy.go). get returns the
+ * current mode string; set switches it (requires an idle device; the
+ * mock backend tracks modes in-process; returns TF_ACCEL_NOT_SUPPORTED
+ * when the amd-smi build lacks the APIs). */
+int tf_accel_get_compute_partition(int device, char* out, int out_len);
+int tf_accel_set_compute_partition(int device, const char* mode);
+int tf_accel_get_memory_partition(int device, char* out, int out_len);
+int tf_accel_set_memory_partition(int device, const char* mode);
+
 int tf_accel_assign_partition(int device, const int32_t* xcds, int n_xcds);
 int tf_accel_remove_partition(int device, const int32_t* xcds, int n_xcds);
 

@@ -24,6 +24,56 @@ from ..api.types import GPU, AllocRequest, GPUPartition, PartitionTemplate, Reso
 COMPUTE_WASTE_WEIGHT = 0.6
 VRAM_WASTE_WEIGHT = 0.4
 
+# AMD compute-partition modes on an 8-XCD MI355X: mode -> (slices,
+# xcds per slice, paired memory interleave). The mode is DEVICE-GLOBAL
+# (amdsmi_set_gpu_compute_partition switches the whole GPU), so slabs of
+# different modes can never coexist on one device — unlike NVIDIA MIG's
+# per-slice profiles (reference partition_strategy.go:90-236).
+PARTITION_MODES = {
+    "SPX": (1, 8, "NPS1"),
+    "DPX": (2, 4, "NPS2"),
+    "QPX": (4, 2, "NPS4"),
+    "CPX": (8, 1, "NPS4"),
+}
+
+
+def default_partition_templates() -> List[PartitionTemplate]:
+    """One template per compute-partition mode, sized by its XCD share
+    (tflops/vram proportional; NPS pairing per mode)."""
+
+    out = []
+    for mode, (slices, xcds, nps) in PARTITION_MODES.items():
+        if slices == 1:
+            continue  # SPX = whole GPU, not a partition template
+        out.append(PartitionTemplate(
+            id=f"{mode.lower()}-{xcds}xcd",
+            name=f"{xcds}xcd.{(C.MI355X_VRAM_BYTES // slices) >> 30}gb",
+            xcds=xcds,
+            compute_percent=100.0 / slices,
+            tflops=C.MI355X_BF16_TFLOPS / slices,
+            vram=C.MI355X_VRAM_BYTES // slices,
+            placements=list(range(0, C.MI355X_XCDS, xcds)),
+            mode=mode,
+            memory_mode=nps,
+        ))
+    return out
+
+
+def device_partition_mode(g: GPU) -> str:
+    """The compute-partition mode a device is committed to, derived from
+    its bound partitions ("" = unpartitioned, any mode may claim it)."""
+
+    for p in g.status.allocated_partitions:
+        tid = p.template_id
+        for mode, (_, xcds, _) in PARTITION_MODES.items():
+            if tid.startswith(mode.lower()):
+                return mode
+        if p.xcds:  # legacy slab templates: infer from slab width
+            for mode, (_, xcds, _) in PARTITION_MODES.items():
+                if xcds == len(p.xcds):
+                    return mode
+    return ""
+
 
 def match_partition_template(req: AllocRequest,
                              templates: List[PartitionTemplate]
@@ -52,8 +102,15 @@ def occupancy_bitmap(g: GPU) -> int:
 
 
 def find_slot(g: GPU, t: PartitionTemplate) -> Optional[List[int]]:
-    """First allowed placement whose XCDs are all free on this device."""
+    """First allowed placement whose XCDs are all free on this device.
+    Enforces mode exclusivity: a device committed to one compute-
+    partition mode rejects templates of any other (the mode is a
+    device-global amdsmi setting on MI355X)."""
 
+    if t.mode:
+        cur = device_partition_mode(g)
+        if cur and cur != t.mode:
+            return None
     bm = occupancy_bitmap(g)
     placements = t.placements or list(range(0, C.MI355X_XCDS, t.xcds))
     for start in placements:

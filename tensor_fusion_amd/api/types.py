@@ -484,6 +484,12 @@ class PartitionTemplate:
     tflops: float = C.MI355X_BF16_TFLOPS / 8
     vram: int = C.MI355X_VRAM_BYTES // 8
     placements: List[int] = field(default_factory=list)  # allowed start XCDs
+    # AMD compute-partition pairing: the device-global mode this template
+    # belongs to (SPX/DPX/QPX/CPX) and the memory interleave it expects
+    # (NPS1/NPS2/NPS4). Templates of different modes cannot coexist on
+    # one device (partitioning.py enforces it).
+    mode: str = ""
+    memory_mode: str = ""
 
 
 @dataclass
@@ -510,11 +516,17 @@ class ProviderConfig(TFObject):
 
 
 def default_mi355x_partition_templates() -> List[PartitionTemplate]:
-    """CPX-style partitioning of one MI355X: 8 XCDs, NPS1/NPS4 friendly sizes."""
+    """Partition templates for one MI355X, one per AMD compute-partition
+    mode (the device-global SPX/DPX/QPX/CPX split over the 8 XCDs) with
+    its paired NPS memory interleave. The mode is enforced exclusively
+    per device by allocator/partitioning.py."""
 
+    mode_by_xcds = {1: ("CPX", "NPS4"), 2: ("QPX", "NPS4"),
+                    4: ("DPX", "NPS2"), 8: ("SPX", "NPS1")}
     out = []
     for xcds in (1, 2, 4, 8):
         frac = xcds / C.MI355X_XCDS
+        mode, nps = mode_by_xcds[xcds]
         out.append(PartitionTemplate(
             id=f"xcd{xcds}",
             name=f"{xcds}xcd.{int(288 * frac)}gb",
@@ -523,6 +535,8 @@ def default_mi355x_partition_templates() -> List[PartitionTemplate]:
             tflops=C.MI355X_BF16_TFLOPS * frac,
             vram=int(C.MI355X_VRAM_BYTES * frac),
             placements=[s for s in range(0, C.MI355X_XCDS, xcds)],
+            mode=mode,
+            memory_mode=nps,
         ))
     return out
 

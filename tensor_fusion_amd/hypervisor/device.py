@@ -126,6 +126,48 @@ class Accelerator:
                 fp16_tflops=d.fp16_tflops, is_mock=bool(d.is_mock)))
         return out
 
+    # ---- AMD compute/memory partition modes (SPX..CPX x NPS1..NPS8) ----
+
+    NOT_SUPPORTED = 2  # TF_ACCEL_NOT_SUPPORTED
+
+    def compute_partition(self, device: int) -> Optional[str]:
+        """Current device-global compute-partition mode, or None when the
+        amd-smi build lacks the API."""
+
+        buf = ctypes.create_string_buffer(32)
+        rc = self._lib.tf_accel_get_compute_partition(device, buf, 32)
+        if rc == self.NOT_SUPPORTED:
+            return None
+        if rc != 0:
+            raise RuntimeError(f"tf_accel_get_compute_partition: {rc}")
+        return buf.value.decode()
+
+    def set_compute_partition(self, device: int, mode: str) -> bool:
+        rc = self._lib.tf_accel_set_compute_partition(device, mode.encode())
+        if rc == self.NOT_SUPPORTED:
+            return False
+        if rc != 0:
+            raise RuntimeError(
+                f"tf_accel_set_compute_partition({mode}): {rc}")
+        return True
+
+    def memory_partition(self, device: int) -> Optional[str]:
+        buf = ctypes.create_string_buffer(32)
+        rc = self._lib.tf_accel_get_memory_partition(device, buf, 32)
+        if rc == self.NOT_SUPPORTED:
+            return None
+        if rc != 0:
+            raise RuntimeError(f"tf_accel_get_memory_partition: {rc}")
+        return buf.value.decode()
+
+    def set_memory_partition(self, device: int, mode: str) -> bool:
+        rc = self._lib.tf_accel_set_memory_partition(device, mode.encode())
+        if rc == self.NOT_SUPPORTED:
+            return False
+        if rc != 0:
+            raise RuntimeError(f"tf_accel_set_memory_partition({mode}): {rc}")
+        return True
+
     def topology(self, n: int) -> List[List[int]]:
         arr = (ctypes.c_int32 * (n * n))()
         rc = self._lib.tf_accel_get_topology(arr, n)
