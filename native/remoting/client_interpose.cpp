@@ -28,6 +28,7 @@
 #include <arpa/inet.h>
 #include <netinet/in.h>
 #include <netinet/tcp.h>
+#include <linux/errqueue.h>
 #include <sys/socket.h>
 #include <sys/stat.h>
 #include <sys/un.h>
@@ -187,11 +188,114 @@ bool tcp_write_full(int fd, const void* p, size_t n) {
   return true;
 }
 
+// One frame = one sendmsg (header + record + up to two arena slices —
+// the payload goes straight from the hipHostRegister'd arena to the
+// socket, no staging copy). With MSG_ZEROCOPY (TF_TCP_ZEROCOPY=1 and a
+// payload past the threshold) the kernel pins those arena pages for DMA
+// instead of copying into skbs; completion notifications arrive on the
+// error queue and gate arena-chunk reuse (ZcPending below).
+struct ZcState {
+  bool enabled = false;       // SO_ZEROCOPY accepted on this socket
+  uint32_t next_seq = 0;      // kernel numbers zerocopy sends 0,1,2,...
+  uint32_t completed = 0;     // highest seq + 1 fully acked
+  struct Pending {
+    uint32_t seq;
+    uint64_t arena_end;       // arena_freed advances here on completion
+  };
+  std::vector<Pending> pending;
+};
+
+void zc_drain(int fd, ZcState& zc, Header* hdr) {
+  // non-blocking errqueue read: SO_EE_ORIGIN_ZEROCOPY ranges
+  while (true) {
+    char ctrl[128];
+    msghdr msg{};
+    msg.msg_control = ctrl;
+    msg.msg_controllen = sizeof ctrl;
+    ssize_t r = recvmsg(fd, &msg, MSG_ERRQUEUE | MSG_DONTWAIT);
+    if (r < 0) break;
+    for (cmsghdr* cm = CMSG_FIRSTHDR(&msg); cm;
+         cm = CMSG_NXTHDR(&msg, cm)) {
+      if ((cm->cmsg_level == SOL_IP && cm->cmsg_type == IP_RECVERR) ||
+          (cm->cmsg_level == SOL_IPV6 && cm->cmsg_type == 25)) {
+        auto* ee = reinterpret_cast<sock_extended_err*>(CMSG_DATA(cm));
+        if (ee->ee_origin == SO_EE_ORIGIN_ZEROCOPY) {
+          uint32_t hi = ee->ee_data;  // inclusive range [ee_info, ee_data]
+          if (hi + 1 > zc.completed) zc.completed = hi + 1;
+        }
+      }
+    }
+  }
+  // advance arena_freed for fully-acked zerocopy payloads (in order)
+  size_t done = 0;
+  for (auto& p : zc.pending) {
+    if (p.seq < zc.completed) {
+      uint64_t cur = at(&hdr->arena_freed)->load();
+      if (p.arena_end > cur) at(&hdr->arena_freed)->store(p.arena_end);
+      ++done;
+    } else {
+      break;
+    }
+  }
+  if (done) zc.pending.erase(zc.pending.begin(), zc.pending.begin() + done);
+}
+
+// Returns the number of MSG_ZEROCOPY sendmsg calls performed (each gets
+// its own kernel completion seq), or -1 on error. A partial send retries
+// with the SAME flags, so every successful call is counted.
+int tcp_send_frame(int fd, FrameHdr* fh, const uint8_t* rec, size_t len,
+                   const uint8_t* pay1, size_t n1, const uint8_t* pay2,
+                   size_t n2, bool zerocopy) {
+  iovec iov[4];
+  int cnt = 0;
+  iov[cnt++] = {fh, sizeof *fh};
+  iov[cnt++] = {const_cast<uint8_t*>(rec), len};
+  if (n1) iov[cnt++] = {const_cast<uint8_t*>(pay1), n1};
+  if (n2) iov[cnt++] = {const_cast<uint8_t*>(pay2), n2};
+  msghdr msg{};
+  msg.msg_iov = iov;
+  msg.msg_iovlen = cnt;
+  size_t total = sizeof *fh + len + n1 + n2;
+  int flags = MSG_NOSIGNAL | (zerocopy ? MSG_ZEROCOPY : 0);
+  int zc_sends = 0;
+  while (total) {
+    ssize_t r = sendmsg(fd, &msg, flags);
+    if (r < 0 && zerocopy && (errno == ENOBUFS || errno == EINVAL))
+      return -1;  // caller retries without zerocopy
+    if (r <= 0) return -1;
+    if (zerocopy) ++zc_sends;
+    total -= (size_t)r;
+    // partial send: advance the iovec
+    size_t adv = (size_t)r;
+    while (adv && msg.msg_iovlen) {
+      if (adv >= msg.msg_iov[0].iov_len) {
+        adv -= msg.msg_iov[0].iov_len;
+        ++msg.msg_iov;
+        --msg.msg_iovlen;
+      } else {
+        msg.msg_iov[0].iov_base = (uint8_t*)msg.msg_iov[0].iov_base + adv;
+        msg.msg_iov[0].iov_len -= adv;
+        adv = 0;
+      }
+    }
+  }
+  return zc_sends;
+}
+
 void* tcp_cmd_pump(void* arg) {
   Client* c = (Client*)arg;
   RingView cmd(&c->hdr->cmd, cmd_buf(c->hdr), CMD_RING_BYTES);
   uint8_t* arena = c->arena_base;
+  ZcState zc;
+  const char* zc_env = getenv("TF_TCP_ZEROCOPY");
+  if (!zc_env || atoi(zc_env) != 0) {
+    int one = 1;
+    zc.enabled = setsockopt(c->sock, SOL_SOCKET, SO_ZEROCOPY, &one,
+                            sizeof one) == 0;
+  }
+  const size_t kZcThreshold = 64u << 10;  // pinned-DMA pays past ~64 KiB
   for (;;) {
+    if (zc.enabled && !zc.pending.empty()) zc_drain(c->sock, zc, c->hdr);
     size_t len;
     uint8_t* p = cmd.try_next(&len);
     if (!p) {
@@ -214,21 +318,40 @@ void* tcp_cmd_pump(void* arg) {
       extra = (uint32_t)b->size;
     }
     FrameHdr fh{0, (uint32_t)len, arena_off, extra, 0};
-    bool ok = tcp_write_full(c->sock, &fh, sizeof fh) &&
-              tcp_write_full(c->sock, p, len);
-    if (ok && extra) {
+    const uint8_t *pay1 = nullptr, *pay2 = nullptr;
+    size_t n1 = 0, n2 = 0;
+    if (extra) {
       size_t off = arena_off % ARENA_BYTES;
       size_t first = ARENA_BYTES - off;
-      if (extra <= first) {
-        ok = tcp_write_full(c->sock, arena + off, extra);
-      } else {
-        ok = tcp_write_full(c->sock, arena + off, first) &&
-             tcp_write_full(c->sock, arena, extra - first);
+      pay1 = arena + off;
+      n1 = extra <= first ? extra : first;
+      if (extra > first) {
+        pay2 = arena;
+        n2 = extra - first;
       }
-      // shipped: the local staging chunk is reusable immediately
+    }
+    bool use_zc = zc.enabled && extra >= kZcThreshold;
+    int zc_sends = tcp_send_frame(c->sock, &fh, p, len, pay1, n1, pay2,
+                                  n2, use_zc);
+    bool ok = zc_sends >= 0;
+    if (!ok && use_zc) {  // kernel refused zerocopy: plain path once
+      zc.enabled = false;
+      use_zc = false;
+      zc_sends = tcp_send_frame(c->sock, &fh, p, len, pay1, n1, pay2, n2,
+                                false);
+      ok = zc_sends >= 0;
+    }
+    if (ok && use_zc) zc.next_seq += (uint32_t)zc_sends;
+    if (ok && extra) {
       uint64_t end = arena_off + ((extra + 63) & ~uint64_t(63));
-      uint64_t cur = at(&c->hdr->arena_freed)->load();
-      if (end > cur) at(&c->hdr->arena_freed)->store(end);
+      if (use_zc && zc_sends > 0) {
+        // arena pages stay pinned by the kernel until the errqueue ack
+        // of the LAST send of this frame: defer the freed advance
+        zc.pending.push_back({zc.next_seq - 1, end});
+      } else {
+        uint64_t cur = at(&c->hdr->arena_freed)->load();
+        if (end > cur) at(&c->hdr->arena_freed)->store(end);
+      }
     }
     cmd.pop();
     if (!ok) {

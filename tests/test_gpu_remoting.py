@@ -273,3 +273,74 @@ def test_remote_worker_under_erl_quota(tmp_path):
     # a decode step is dozens of launches; at 300 launches/s the capped
     # worker must be far below the free worker's token rate
     assert capped["tok_s"] < 0.5 * free["tok_s"], (free, capped)
+
+
+SYNC_RTT_CHILD = r"""
+import json, time
+import torch
+torch.cuda.init()
+x = torch.ones(8, device="cuda")
+torch.cuda.synchronize()
+# sync-op RTT: each synchronize is one OP_DEVICE_SYNC round trip
+N = 100
+t0 = time.perf_counter()
+for _ in range(N):
+    torch.cuda.synchronize()
+rtt_ms = (time.perf_counter() - t0) / N * 1e3
+# async burst: launches are fire-and-forget; only the final sync waits
+t0 = time.perf_counter()
+for _ in range(200):
+    x = x * 1.0001
+torch.cuda.synchronize()
+burst_ms = (time.perf_counter() - t0) * 1e3
+print(json.dumps({"sync_rtt_ms": round(rtt_ms, 3),
+                  "burst200_ms": round(burst_ms, 2)}))
+"""
+
+
+def test_tcp_latency_sensitivity(tmp_path):
+    """Cross-node latency sensitivity of the GPU-over-IP wire: the same
+    TCP command stream through a relay injecting 0 / 1 / 3 ms one-way
+    delay (no iproute2 in this image ⇒ no netns/netem; the relay is the
+    substitute — two extra processes on the same byte stream). Expect:
+    sync-op RTT tracks the injected delay (~2x one-way + base), while
+    async launch bursts stay delay-insensitive (fire-and-forget ring)."""
+
+    from tensor_fusion_amd.client.runtime import client_env, start_worker
+    base_port = 48500
+    w = start_worker("", device_index=0, tcp_port=base_port)
+    results = {}
+    try:
+        for delay_ms in (0.0, 1.0, 3.0):
+            proxy_port = base_port + 1 + int(delay_ms * 10) % 97
+            proxy = subprocess.Popen(
+                [sys.executable,
+                 os.path.join(REPO, "tools", "tcp_latency_proxy.py"),
+                 "--listen", str(proxy_port),
+                 "--connect", f"127.0.0.1:{base_port}",
+                 "--delay-ms", str(delay_ms)],
+                stdout=subprocess.PIPE, text=True)
+            try:
+                assert "PROXY_READY" in proxy.stdout.readline()
+                env = client_env("", tcp=f"127.0.0.1:{proxy_port}")
+                out = subprocess.run(
+                    [sys.executable, "-c", SYNC_RTT_CHILD], env=env,
+                    capture_output=True, text=True, timeout=600, cwd=REPO)
+                assert out.returncode == 0, \
+                    out.stdout[-800:] + out.stderr[-3000:]
+                results[delay_ms] = json.loads(
+                    out.stdout.strip().splitlines()[-1])
+            finally:
+                proxy.kill()
+    finally:
+        w.stop()
+    r0, r1, r3 = results[0.0], results[1.0], results[3.0]
+    print(f"\nsync RTT ms @0/1/3ms delay: {r0['sync_rtt_ms']} / "
+          f"{r1['sync_rtt_ms']} / {r3['sync_rtt_ms']}; "
+          f"200-launch async burst: {r0['burst200_ms']} / "
+          f"{r1['burst200_ms']} / {r3['burst200_ms']}")
+    # sync ops pay ~2x the injected one-way delay
+    assert r1["sync_rtt_ms"] >= r0["sync_rtt_ms"] + 1.0
+    assert r3["sync_rtt_ms"] >= r0["sync_rtt_ms"] + 4.0
+    # async bursts must NOT pay per-launch latency (<< 200 * delay)
+    assert r3["burst200_ms"] < 200 * 3.0
