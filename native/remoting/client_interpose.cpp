@@ -1786,6 +1786,149 @@ hipError_t hipExtStreamGetCUMask(void*, unsigned n, unsigned* mask) {
 }
 
 // not-yet-supported surfaces: fail loudly rather than run locally
+// ---- client-side VMM surface (forwarded): what PyTorch's
+// expandable_segments allocator and vLLM-class servers drive. VAs and
+// physical handles are WORKER-side values; a mapped range registers in
+// dev_ranges so memcpy/memset route it like any device pointer.
+
+extern "C" hipError_t hipMemAddressReserve(void** ptr, size_t size,
+                                           size_t align, void* hint,
+                                           unsigned long long flags) {
+  struct {
+    uint64_t size, align, hint, flags;
+  } b{size, align, (uint64_t)hint, flags};
+  uint64_t out = 0;
+  hipError_t e = send_sync(OP_VMM_RESERVE, &b, sizeof b, &out, 8);
+  if (e == hipSuccess) *ptr = (void*)out;
+  return e;
+}
+
+extern "C" hipError_t hipMemAddressFree(void* ptr, size_t size) {
+  struct {
+    uint64_t ptr, size;
+  } b{(uint64_t)ptr, size};
+  send_async(OP_VMM_ADDR_FREE, 0, &b, sizeof b);
+  return hipSuccess;
+}
+
+extern "C" hipError_t hipMemCreate(void* handle, size_t size,
+                                   const void* prop,
+                                   unsigned long long flags) {
+  struct {
+    uint64_t size, flags;
+    uint8_t prop[48];
+  } b{};
+  b.size = size;
+  b.flags = flags;
+  memcpy(b.prop, prop, 48);  // hipMemAllocationProp fits in 48 bytes
+  uint64_t out = 0;
+  hipError_t e = send_sync(OP_VMM_CREATE, &b, sizeof b, &out, 8);
+  if (e == hipSuccess) *(uint64_t*)handle = out;
+  return e;
+}
+
+extern "C" hipError_t hipMemRelease(void* handle) {
+  uint64_t h = (uint64_t)handle;
+  send_async(OP_VMM_RELEASE, 0, &h, 8);
+  return hipSuccess;
+}
+
+extern "C" hipError_t hipMemMap(void* va, size_t size, size_t offset,
+                                void* handle, unsigned long long flags) {
+  struct {
+    uint64_t va, size, off, handle, flags;
+  } b{(uint64_t)va, size, offset, (uint64_t)handle, flags};
+  hipError_t e = send_sync(OP_VMM_MAP, &b, sizeof b, nullptr, 0);
+  if (e == hipSuccess) {
+    Client& c = C();
+    std::lock_guard<std::mutex> l(c.mem_mu);
+    c.dev_ranges[(uint64_t)va] = size;
+  }
+  return e;
+}
+
+extern "C" hipError_t hipMemUnmap(void* va, size_t size) {
+  struct {
+    uint64_t va, size;
+  } b{(uint64_t)va, size};
+  hipError_t e = send_sync(OP_VMM_UNMAP, &b, sizeof b, nullptr, 0);
+  if (e == hipSuccess) {
+    Client& c = C();
+    std::lock_guard<std::mutex> l(c.mem_mu);
+    c.dev_ranges.erase((uint64_t)va);
+  }
+  return e;
+}
+
+extern "C" hipError_t hipMemSetAccess(void* va, size_t size,
+                                      const void* desc, size_t count) {
+  if (count > 8) return hipErrorNotSupported;
+  struct {
+    uint64_t va, size;
+    uint32_t count, pad;
+  } hdr{(uint64_t)va, size, (uint32_t)count, 0};
+  uint8_t body[sizeof hdr + 8 * 12];
+  memcpy(body, &hdr, sizeof hdr);
+  memcpy(body + sizeof hdr, desc, count * 12);  // {loc{type,id},flags}
+  return send_sync(OP_VMM_SET_ACCESS, body, sizeof hdr + count * 12,
+                   nullptr, 0);
+}
+
+extern "C" hipError_t hipMemGetAllocationGranularity(size_t* g,
+                                                     const void* prop,
+                                                     int opt) {
+  struct {
+    uint64_t opt;
+    uint8_t prop[48];
+  } b{};
+  b.opt = (uint64_t)opt;
+  memcpy(b.prop, prop, 48);
+  uint64_t out = 0;
+  hipError_t e = send_sync(OP_VMM_GRANULARITY, &b, sizeof b, &out, 8);
+  if (e == hipSuccess) *g = (size_t)out;
+  return e;
+}
+
+// ---- stream-ordered allocator + mempool surface (hipMallocAsync
+// backend of the PyTorch caching allocator)
+
+extern "C" hipError_t hipDeviceGetDefaultMemPool(void** pool, int dev) {
+  uint32_t d = (uint32_t)dev;
+  uint64_t out = 0;
+  hipError_t e = send_sync(OP_MEMPOOL_DEFAULT, &d, 4, &out, 8);
+  if (e == hipSuccess) *pool = (void*)out;
+  return e;
+}
+
+extern "C" hipError_t hipMemPoolSetAttribute(void* pool, int attr,
+                                             void* value) {
+  struct {
+    uint64_t pool, attr, value;
+  } b{(uint64_t)pool, (uint64_t)attr, 0};
+  // attrs are u64 or int-sized thresholds; copy 8 bytes conservatively
+  memcpy(&b.value, value, 8);
+  return send_sync(OP_MEMPOOL_SET_ATTR, &b, sizeof b, nullptr, 0);
+}
+
+extern "C" hipError_t hipMemPoolGetAttribute(void* pool, int attr,
+                                             void* value) {
+  struct {
+    uint64_t pool, attr;
+  } b{(uint64_t)pool, (uint64_t)attr};
+  uint64_t out = 0;
+  hipError_t e = send_sync(OP_MEMPOOL_GET_ATTR, &b, sizeof b, &out, 8);
+  if (e == hipSuccess) memcpy(value, &out, 8);
+  return e;
+}
+
+extern "C" hipError_t hipMemPoolTrimTo(void* pool, size_t keep) {
+  struct {
+    uint64_t pool, keep;
+  } b{(uint64_t)pool, keep};
+  send_async(OP_MEMPOOL_TRIM, 0, &b, sizeof b);
+  return hipSuccess;
+}
+
 #define TF_NOTSUP(name, ...)                         \
   hipError_t name(__VA_ARGS__) {                     \
     fprintf(stderr, "[tf-client] %s: not supported over remoting yet\n", \
@@ -1798,20 +1941,9 @@ TF_NOTSUP(hipMemcpyToSymbol, const void*, const void*, size_t, size_t, int)
 TF_NOTSUP(hipIpcGetMemHandle, void*, void*)
 TF_NOTSUP(hipIpcOpenMemHandle, void**, const void*, unsigned)
 TF_NOTSUP(hipIpcCloseMemHandle, void*)
-TF_NOTSUP(hipMemAddressReserve, void**, size_t, size_t, void*, unsigned long long)
-TF_NOTSUP(hipMemCreate, void*, size_t, const void*, unsigned long long)
-TF_NOTSUP(hipMemMap, void*, size_t, size_t, void*, unsigned long long)
-TF_NOTSUP(hipMemUnmap, void*, size_t)
-TF_NOTSUP(hipMemRelease, void*)
-TF_NOTSUP(hipMemSetAccess, void*, size_t, const void*, size_t)
-TF_NOTSUP(hipMemGetAllocationGranularity, size_t*, const void*, int)
 TF_NOTSUP(hipMemExportToShareableHandle, void*, void*, int, unsigned long long)
 TF_NOTSUP(hipMemImportFromShareableHandle, void*, void*, int)
-TF_NOTSUP(hipDeviceGetDefaultMemPool, void**, int)
-TF_NOTSUP(hipMemPoolGetAttribute, void*, int, void*)
-TF_NOTSUP(hipMemPoolSetAttribute, void*, int, void*)
 TF_NOTSUP(hipMemPoolSetAccess, void*, const void*, size_t)
-TF_NOTSUP(hipMemPoolTrimTo, void*, size_t)
 TF_NOTSUP(hipMemcpyPeerAsync, void*, int, const void*, int, size_t, void*)
 TF_NOTSUP(hipStreamWriteValue32, void*, void*, unsigned, unsigned)
 TF_NOTSUP(hipGraphNodeGetDependencies, void*, void**, size_t*)

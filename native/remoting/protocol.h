@@ -134,6 +134,25 @@ enum Op : uint32_t {
   OP_GRAPH_GET_NODES,    // {graph, cap} → reply {count, nodes[...]}
   OP_GRAPH_NODE_TYPES,   // {graph} → reply {count, type_histogram[16]}
   OP_GRAPH_KERNEL_HISTO, // {graph} → reply {text histogram of node kernels}
+
+  // client-side VMM surface (PyTorch expandable_segments, vLLM-class
+  // allocators): forwarded 1:1; handles/VAs are worker-side values
+  OP_VMM_RESERVE,        // {size, align, addr_hint, flags} → {ptr}
+  OP_VMM_ADDR_FREE,      // {ptr, size} async
+  OP_VMM_CREATE,         // {size, flags, prop[48]} → {handle}
+  OP_VMM_RELEASE,        // {handle} async
+  OP_VMM_MAP,            // {va, size, offset, handle, flags} → {err}
+  OP_VMM_UNMAP,          // {va, size} → {err}
+  OP_VMM_SET_ACCESS,     // {va, size, count, desc[count*12]} → {err}
+  OP_VMM_GRANULARITY,    // {opt, prop[48]} → {granularity}
+
+  // stream-ordered allocator + mempool surface (hipMallocAsync backend)
+  OP_MALLOC_ASYNC,       // {size, stream} → {ptr}
+  OP_FREE_ASYNC,         // {ptr, stream} async
+  OP_MEMPOOL_DEFAULT,    // {dev} → {pool}
+  OP_MEMPOOL_SET_ATTR,   // {pool, attr, value} → {err}
+  OP_MEMPOOL_GET_ATTR,   // {pool, attr} → {value}
+  OP_MEMPOOL_TRIM,       // {pool, keep} async
 };
 
 constexpr uint32_t F_WANT_REPLY = 1u << 0;

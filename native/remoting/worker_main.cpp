@@ -111,6 +111,13 @@ struct Hip {
   hipError_t (*MemSetAccess)(void*, size_t, const void*, size_t) = nullptr;
   hipError_t (*MemGetAllocationGranularity)(size_t*, const void*,
                                             int) = nullptr;
+  // stream-ordered allocator / mempool surface (optional)
+  hipError_t (*MallocAsync)(void**, size_t, void*) = nullptr;
+  hipError_t (*FreeAsync)(void*, void*) = nullptr;
+  hipError_t (*DeviceGetDefaultMemPool)(void**, int) = nullptr;
+  hipError_t (*MemPoolSetAttribute)(void*, int, void*) = nullptr;
+  hipError_t (*MemPoolGetAttribute)(void*, int, void*) = nullptr;
+  hipError_t (*MemPoolTrimTo)(void*, size_t) = nullptr;
 
   bool load() {
     const char* names[] = {"libamdhip64.so", "libamdhip64.so.7",
@@ -177,6 +184,12 @@ struct Hip {
     O(MemUnmap, "hipMemUnmap")
     O(MemSetAccess, "hipMemSetAccess")
     O(MemGetAllocationGranularity, "hipMemGetAllocationGranularity")
+    O(MallocAsync, "hipMallocAsync")
+    O(FreeAsync, "hipFreeAsync")
+    O(DeviceGetDefaultMemPool, "hipDeviceGetDefaultMemPool")
+    O(MemPoolSetAttribute, "hipMemPoolSetAttribute")
+    O(MemPoolGetAttribute, "hipMemPoolGetAttribute")
+    O(MemPoolTrimTo, "hipMemPoolTrimTo")
 #undef O
     return true;
   }
@@ -1123,6 +1136,157 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
         }
       }
       reply(c->seq, e, text.data(), (uint32_t)text.size());
+      break;
+    }
+    // ---- client-forwarded VMM surface (PyTorch expandable_segments /
+    // vLLM-class allocators). VAs and handles live in THIS process;
+    // ranges mapped here are ordinary device pointers for every other op.
+    case OP_VMM_RESERVE: {
+      struct B { uint64_t size, align, hint, flags; } b;
+      memcpy(&b, body, sizeof b);
+      void* p = nullptr;
+      hipError_t e = hip.MemAddressReserve
+          ? hip.MemAddressReserve(&p, b.size, b.align, (void*)b.hint,
+                                  b.flags)
+          : 801;
+      uint64_t r = (uint64_t)p;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_VMM_ADDR_FREE: {
+      struct B { uint64_t ptr, size; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e = hip.MemAddressFree
+          ? hip.MemAddressFree((void*)b.ptr, b.size) : 801;
+      if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
+      else set_sticky(e, "hipMemAddressFree");
+      break;
+    }
+    case OP_VMM_CREATE: {
+      struct B { uint64_t size, flags; uint8_t prop[48]; } b;
+      memcpy(&b, body, sizeof b);
+      void* h2 = nullptr;
+      hipError_t e = hip.MemCreate
+          ? hip.MemCreate(&h2, b.size, b.prop, b.flags) : 801;
+      uint64_t r = (uint64_t)h2;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_VMM_RELEASE: {
+      uint64_t h2;
+      memcpy(&h2, body, 8);
+      hipError_t e = hip.MemRelease ? hip.MemRelease((void*)h2) : 801;
+      if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
+      else set_sticky(e, "hipMemRelease");
+      break;
+    }
+    case OP_VMM_MAP: {
+      struct B { uint64_t va, size, off, handle, flags; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e = hip.MemMap
+          ? hip.MemMap((void*)b.va, b.size, b.off, (void*)b.handle,
+                       b.flags)
+          : 801;
+      reply(c->seq, e, nullptr, 0);
+      break;
+    }
+    case OP_VMM_UNMAP: {
+      struct B { uint64_t va, size; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e = hip.MemUnmap
+          ? hip.MemUnmap((void*)b.va, b.size) : 801;
+      reply(c->seq, e, nullptr, 0);
+      break;
+    }
+    case OP_VMM_SET_ACCESS: {
+      struct B { uint64_t va, size; uint32_t count, pad; } b;
+      memcpy(&b, body, sizeof b);
+      // desc array follows: count * {int type, int id, int flags}
+      hipError_t e = hip.MemSetAccess
+          ? hip.MemSetAccess((void*)b.va, b.size, body + sizeof b,
+                             b.count)
+          : 801;
+      reply(c->seq, e, nullptr, 0);
+      break;
+    }
+    case OP_VMM_GRANULARITY: {
+      struct B { uint64_t opt; uint8_t prop[48]; } b;
+      memcpy(&b, body, sizeof b);
+      size_t g2 = 0;
+      hipError_t e = hip.MemGetAllocationGranularity
+          ? hip.MemGetAllocationGranularity(&g2, b.prop, (int)b.opt)
+          : 801;
+      uint64_t r = g2;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_MALLOC_ASYNC: {
+      struct B { uint64_t size, stream; } b;
+      memcpy(&b, body, sizeof b);
+      void* p = nullptr;
+      hipError_t e;
+      if (hip.MallocAsync) {
+        e = hip.MallocAsync(&p, b.size, (hipStream_t)xl(b.stream));
+      } else if (W.vmm.enabled) {
+        p = W.vmm.alloc(b.size, &e);
+      } else {
+        e = hip.Malloc(&p, b.size);
+      }
+      uint64_t r = (uint64_t)p;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_FREE_ASYNC: {
+      struct B { uint64_t ptr, stream; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e;
+      if (hip.FreeAsync) {
+        e = hip.FreeAsync((void*)b.ptr, (hipStream_t)xl(b.stream));
+      } else if (W.vmm.enabled) {
+        e = W.vmm.free_((void*)b.ptr);
+      } else {
+        e = hip.Free((void*)b.ptr);
+      }
+      if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
+      else set_sticky(e, "hipFreeAsync");
+      break;
+    }
+    case OP_MEMPOOL_DEFAULT: {
+      uint32_t dev;
+      memcpy(&dev, body, 4);
+      void* pool = nullptr;
+      hipError_t e = hip.DeviceGetDefaultMemPool
+          ? hip.DeviceGetDefaultMemPool(&pool, (int)dev) : 801;
+      uint64_t r = (uint64_t)pool;
+      reply(c->seq, e, &r, 8);
+      break;
+    }
+    case OP_MEMPOOL_SET_ATTR: {
+      struct B { uint64_t pool, attr, value; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e = hip.MemPoolSetAttribute
+          ? hip.MemPoolSetAttribute((void*)b.pool, (int)b.attr, &b.value)
+          : 801;
+      reply(c->seq, e, nullptr, 0);
+      break;
+    }
+    case OP_MEMPOOL_GET_ATTR: {
+      struct B { uint64_t pool, attr; } b;
+      memcpy(&b, body, sizeof b);
+      uint64_t value = 0;
+      hipError_t e = hip.MemPoolGetAttribute
+          ? hip.MemPoolGetAttribute((void*)b.pool, (int)b.attr, &value)
+          : 801;
+      reply(c->seq, e, &value, 8);
+      break;
+    }
+    case OP_MEMPOOL_TRIM: {
+      struct B { uint64_t pool, keep; } b;
+      memcpy(&b, body, sizeof b);
+      hipError_t e = hip.MemPoolTrimTo
+          ? hip.MemPoolTrimTo((void*)b.pool, b.keep) : 801;
+      if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
+      else set_sticky(e, "hipMemPoolTrimTo");
       break;
     }
     case OP_SHUTDOWN:

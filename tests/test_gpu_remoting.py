@@ -344,3 +344,74 @@ def test_tcp_latency_sensitivity(tmp_path):
     assert r3["sync_rtt_ms"] >= r0["sync_rtt_ms"] + 4.0
     # async bursts must NOT pay per-launch latency (<< 200 * delay)
     assert r3["burst200_ms"] < 200 * 3.0
+
+
+EXPANDABLE_CHILD = r"""
+import json, os
+import torch
+# expandable_segments drives the VMM surface: hipMemAddressReserve /
+# hipMemCreate / hipMemMap / hipMemSetAccess / hipMemUnmap / hipMemRelease
+assert "expandable_segments" in os.environ.get("PYTORCH_ALLOC_CONF", "") \
+    or "expandable_segments" in os.environ.get("PYTORCH_HIP_ALLOC_CONF", "")
+a = torch.randn(4 << 20, device="cuda")
+b = torch.randn(4 << 20, device="cuda")
+c = (a * 2 + b).sum()
+del a
+torch.cuda.empty_cache()  # unmap path
+d = torch.randn(8 << 20, device="cuda")  # grow the segment (remap)
+torch.cuda.synchronize()
+print(json.dumps({"ok": bool(torch.isfinite(c).item()),
+                  "sum": float(c.item()), "d0": float(d[0].item())}))
+"""
+
+
+def test_expandable_segments_allocator_remoted(worker):
+    """PyTorch's expandable_segments caching-allocator backend through
+    GPU-over-IP: the client-side VMM surface (reserve/create/map/
+    set-access/unmap/release) is forwarded to the worker — the allocator
+    machinery vLLM-class servers sit on (vLLM itself is not installed in
+    this offline image; this is the same HIP surface)."""
+
+    env = client_env(worker.socket_path)
+    env["PYTORCH_ALLOC_CONF"] = "expandable_segments:True"
+    env["PYTORCH_HIP_ALLOC_CONF"] = "expandable_segments:True"
+    out = subprocess.run([sys.executable, "-c", EXPANDABLE_CHILD], env=env,
+                         capture_output=True, text=True, timeout=600,
+                         cwd=REPO)
+    assert out.returncode == 0, out.stdout[-1000:] + out.stderr[-4000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["ok"] is True
+
+
+TRANSFORMERS_CHILD = r"""
+import json
+import torch
+from transformers import LlamaConfig, LlamaForCausalLM
+cfg = LlamaConfig(vocab_size=256, hidden_size=128, num_hidden_layers=2,
+                  num_attention_heads=4, num_key_value_heads=2,
+                  intermediate_size=256, max_position_embeddings=512)
+torch.manual_seed(0)
+model = LlamaForCausalLM(cfg).to("cuda", torch.bfloat16).eval()
+ids = torch.randint(0, 256, (2, 12), device="cuda")
+with torch.no_grad():
+    out = model.generate(ids, max_new_tokens=16, do_sample=False,
+                         pad_token_id=0)
+assert out.shape == (2, 28), out.shape
+print(json.dumps({"ok": True, "tokens": out.shape[1]}))
+"""
+
+
+def test_transformers_generate_remoted(worker):
+    """HuggingFace transformers LlamaForCausalLM.generate() through the
+    remoting path (random-init tiny config — no network for weights).
+    Exercises the HF serving stack's op surface (SDPA attention, KV
+    cache, sampling machinery) as the closest available stand-in for the
+    vLLM config-3 workload."""
+
+    env = client_env(worker.socket_path)
+    out = subprocess.run([sys.executable, "-c", TRANSFORMERS_CHILD],
+                         env=env, capture_output=True, text=True,
+                         timeout=900, cwd=REPO)
+    assert out.returncode == 0, out.stdout[-1000:] + out.stderr[-5000:]
+    r = json.loads(out.stdout.strip().splitlines()[-1])
+    assert r["ok"] is True and r["tokens"] == 28
