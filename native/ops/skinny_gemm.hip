@@ -191,9 +191,119 @@ skinny_gemm_lds_kernel(const __hip_bfloat16* __restrict__ x,
   }
 }
 
+// ---- v3: pre-shuffled weight layout (the hipBLASLt-closing move) -----
+// The round-1 ablation pinned the direct kernel at 4.4 TB/s: ~8000
+// concurrent 16-row K-strided W streams are a DRAM-unfriendly pattern no
+// unroll depth fixes. The framework owns the weight layout, so shuffle W
+// ONCE at model load into the exact per-lane MFMA fragment order:
+//
+//   P[nt][kt][lane][j] = W[nt*16 + (lane&15)][kt*32 + (lane>>4)*8 + j]
+//
+// (ops/fused.py pack_skinny_weight — a single torch permute). Each wave
+// then reads ONE fully-sequential stream: per k-step the 64 lanes fetch
+// 1 KiB contiguous, step after step, tile after tile. DRAM sees pure
+// sequential bursts. No LDS, no layout math in the hot loop — the
+// pointer just increments.
+
+template <int UNROLL>
+__global__ void skinny_gemm_packed_kernel(
+    const __hip_bfloat16* __restrict__ x,
+    const __hip_bfloat16* __restrict__ wp,  // packed [N/16][K/32][64][8]
+    __hip_bfloat16* __restrict__ y, int M, int N, int K) {
+  const int lane = threadIdx.x & 63;
+  const int wave = threadIdx.x >> 6;
+  const int nt = blockIdx.x * 4 + wave;
+  const int n0 = nt * 16;
+  if (n0 >= N) return;
+
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n = n0 + col;
+  const bool arow_ok = col < M;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int ktiles = K / 32;
+  // this wave's sequential W stream: 512 elems (1 KiB) per k-tile
+  const __hip_bfloat16* wstream =
+      wp + ((size_t)nt * ktiles) * 512 + lane * 8;
+  const __hip_bfloat16* xrow = x + (size_t)(arow_ok ? col : 0) * K;
+
+  int t = 0;
+  for (; t + UNROLL <= ktiles; t += UNROLL) {
+    B16x8 a[UNROLL], b[UNROLL];
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u) {
+      int k = (t + u) * 32 + kgrp * 8;
+      a[u].raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                         : u4{0, 0, 0, 0};
+      b[u].raw = __builtin_nontemporal_load(
+          reinterpret_cast<const u4*>(wstream + (size_t)(t + u) * 512));
+    }
+#pragma unroll
+    for (int u = 0; u < UNROLL; ++u)
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u].v, b[u].v, acc, 0,
+                                                    0, 0);
+  }
+  for (; t < ktiles; ++t) {
+    int k = t * 32 + kgrp * 8;
+    B16x8 a, b;
+    a.raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                    : u4{0, 0, 0, 0};
+    b.raw = __builtin_nontemporal_load(
+        reinterpret_cast<const u4*>(wstream + (size_t)t * 512));
+    acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+  }
+
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    int row = kgrp * 4 + r;
+    if (row < M && n < N) y[(size_t)row * N + n] = __float2bfloat16(acc[r]);
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+// x[M,K] bf16; wp = packed weights (pack_skinny_weight: [N/16][K/32][64][8]);
+// y[M,N] bf16. Requires K % 32 == 0, N % 16 == 0, M <= 16.
+int tf_skinny_gemm_packed(const void* x, const void* wp, void* y, int M,
+                          int N, int K, void* stream) {
+  if (M < 1 || M > 16 || (K & 31) || (N & 15)) return 1;
+  dim3 block(256);
+  dim3 grid((N + 63) / 64);
+  static int unroll = [] {
+    const char* v = getenv("TF_SKINNY_UNROLL");
+    int u = v ? atoi(v) : 8;
+    return (u == 4 || u == 12 || u == 16) ? u : 8;
+  }();
+  switch (unroll) {
+    case 4:
+      hipLaunchKernelGGL(skinny_gemm_packed_kernel<4>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)wp, (__hip_bfloat16*)y, M,
+                         N, K);
+      break;
+    case 12:
+      hipLaunchKernelGGL(skinny_gemm_packed_kernel<12>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)wp, (__hip_bfloat16*)y, M,
+                         N, K);
+      break;
+    case 16:
+      hipLaunchKernelGGL(skinny_gemm_packed_kernel<16>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)wp, (__hip_bfloat16*)y, M,
+                         N, K);
+      break;
+    default:
+      hipLaunchKernelGGL(skinny_gemm_packed_kernel<8>, grid, block, 0,
+                         (hipStream_t)stream, (const __hip_bfloat16*)x,
+                         (const __hip_bfloat16*)wp, (__hip_bfloat16*)y, M,
+                         N, K);
+  }
+  return (int)hipGetLastError();
+}
 
 // x[M,K], w[N,K] row-major bf16; y[M,N] bf16. K % 32 == 0, M <= 16.
 int tf_skinny_gemm(const void* x, const void* w, void* y, int M, int N,

@@ -211,6 +211,21 @@ class Llama(nn.Module):
             nxt_eps = (self.blocks[i + 1].ln1.eps if i + 1 < n
                        else self.norm.eps)
             normed = fused.add_rmsnorm(m.contiguous(), res, nxt_w, nxt_eps)
+        # lm_head via the packed-layout MFMA decode GEMV when the token
+        # batch fits (M <= 16): 6.25 TB/s vs hipBLASLt's 5.7 on this
+        # shape (profiles/skinny_gemm_packed_r02.md). Weights are
+        # pre-shuffled once and cached; fall back to the plain matmul
+        # for prefill-sized batches.
+        B, T, D = normed.shape
+        if (B * T <= 16 and D % 32 == 0
+                and self.lm_head.weight.shape[0] % 16 == 0):
+            if getattr(self, "_lm_packed", None) is None:
+                self._lm_packed = fused.pack_skinny_weight(
+                    self.lm_head.weight.detach())
+            flat = normed.reshape(B * T, D).contiguous()
+            out = fused.skinny_gemm_packed(
+                flat, self._lm_packed, self.lm_head.weight.shape[0])
+            return out.view(B, T, -1)
         return self.lm_head(normed)
 
     def make_kv_cache(self, batch: int, max_seq: int, device, dtype):

@@ -86,6 +86,43 @@ def add_rmsnorm(x, residual, weight, eps: float):
     return out
 
 
+def pack_skinny_weight(weight):
+    """Pre-shuffle nn.Linear weight[N,K] into the MFMA per-lane fragment
+    order consumed by tf_skinny_gemm_packed: P[nt][kt][lane][j] =
+    W[nt*16 + (lane&15)][kt*32 + (lane>>4)*8 + j]. One-time cost at
+    model load; each compute wave then streams ONE sequential region
+    (the layout change that closes the 4.4 vs 5.7 TB/s gap to hipBLASLt
+    — profiles/skinny_gemm_packed_r02.md)."""
+
+    import torch
+    N, K = weight.shape
+    assert N % 16 == 0 and K % 32 == 0, (N, K)
+    # (nt, col, kt, kgrp, j) -> (nt, kt, kgrp, col, j); lane = kgrp*16+col
+    return (weight.view(N // 16, 16, K // 32, 4, 8)
+            .permute(0, 2, 3, 1, 4).contiguous())
+
+
+def skinny_gemm_packed(x, wp, N: int):
+    """y = x @ W.T with W pre-shuffled by pack_skinny_weight."""
+
+    import torch
+    assert x.dtype == torch.bfloat16 and x.is_contiguous()
+    M, K = x.shape[-2], x.shape[-1]
+    lb = lib()
+    if not hasattr(lb, "_skinnyp_ready"):
+        lb.tf_skinny_gemm_packed.restype = ctypes.c_int
+        lb.tf_skinny_gemm_packed.argtypes = [
+            ctypes.c_void_p, ctypes.c_void_p, ctypes.c_void_p, ctypes.c_int,
+            ctypes.c_int, ctypes.c_int, ctypes.c_void_p]
+        lb._skinnyp_ready = True
+    y = torch.empty(*x.shape[:-1], N, device=x.device, dtype=torch.bfloat16)
+    rc = lb.tf_skinny_gemm_packed(x.data_ptr(), wp.data_ptr(), y.data_ptr(),
+                                  M, N, K, _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_skinny_gemm_packed failed: {rc}")
+    return y
+
+
 def skinny_gemm(x, weight):
     """y = x @ weight.T via the gfx950 MFMA decode-GEMV kernel
     (native/ops/skinny_gemm.hip). x[M,K] bf16 with M ≤ 16; weight[N,K]
