@@ -200,7 +200,14 @@ class DefragController:
 
     def execute_due_evictions(self, now: Optional[float] = None) -> List[str]:
         """Delete pods whose eviction TTL expired (the grace window lets
-        operators cancel by removing the annotation)."""
+        operators cancel by removing the annotation).
+
+        Before deleting, the plan is RE-VALIDATED: cluster state may have
+        moved during the TTL window (new pods landed on the target nodes)
+        and an eviction whose allocation no longer fits elsewhere would
+        strand the workload — the reference re-checks placements when the
+        campaign executes (gpupool_defrag.go eviction path). A pod that
+        fails revalidation keeps running and its mark is cleared."""
 
         now = now if now is not None else time.time()
         evicted = []
@@ -208,10 +215,64 @@ class DefragController:
             mark = pod.meta.annotations.get(AnnoEvictionMark)
             if not mark:
                 continue
-            if float(mark) <= now:
-                try:
-                    self.store.delete("Pod", pod.meta.name, pod.meta.namespace)
-                    evicted.append(pod.meta.key)
-                except Exception:
-                    continue
+            if float(mark) > now:
+                continue
+            pod_key = pod.meta.key
+            alloc = self.allocator.allocation(pod_key)
+            if alloc is not None and not self._still_placeable(pod_key,
+                                                               alloc):
+                self._clear_mark(pod)
+                continue
+            try:
+                self.store.delete("Pod", pod.meta.name, pod.meta.namespace)
+                evicted.append(pod_key)
+            except Exception:
+                continue
         return evicted
+
+    def _still_placeable(self, pod_key: str, alloc) -> bool:
+        """Would this allocation still fit somewhere off its current
+        node if evicted right now?"""
+
+        node = ""
+        if alloc.gpu_names:
+            g = self.allocator.gpu(alloc.gpu_names[0])
+            if g is not None:
+                node = g.status.node
+        scratch = self.allocator.fork_for_simulation(
+            exclude_nodes=[node] if node else [])
+        req2 = copy.deepcopy(alloc.req)
+        req2.pod_name = alloc.req.pod_name + "-reval"
+        try:
+            scores, _ = scratch.check_quota_and_filter(req2)
+        except Exception:
+            return False
+        for ns_ in sorted(scores.values(), key=lambda s: s.score,
+                          reverse=True):
+            try:
+                gpus = scratch.pick_gpus(req2, ns_.node)
+                scratch.assume(req2, gpus)
+                return True
+            except Exception:
+                continue
+        return False
+
+    def _clear_mark(self, pod):
+        def _p(obj):
+            obj.meta.annotations.pop(AnnoEvictionMark, None)
+            obj.meta.annotations.pop(AnnoEvictionReason, None)
+        try:
+            self.store.patch("Pod", pod.meta.name, pod.meta.namespace, _p)
+        except Exception:
+            pass
+
+    def abort_campaign(self) -> int:
+        """Operator escape hatch: clear every pending eviction mark
+        (reference campaigns are cancellable until execution)."""
+
+        n = 0
+        for pod in self.store.list("Pod"):
+            if AnnoEvictionMark in pod.meta.annotations:
+                self._clear_mark(pod)
+                n += 1
+        return n

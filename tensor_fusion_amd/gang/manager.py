@@ -29,7 +29,9 @@ class PodGroup:
     members: Set[str] = field(default_factory=set)       # known pod keys
     scheduled: Set[str] = field(default_factory=set)     # bound pod keys
     waiting: Set[str] = field(default_factory=set)       # parked in Permit
+    waiting_since: Dict[str, float] = field(default_factory=dict)
     rejected_until: float = 0.0
+    rejections: int = 0  # consecutive all-or-nothing rejections
     created: float = field(default_factory=time.time)
 
     @property
@@ -38,7 +40,9 @@ class PodGroup:
 
 
 class GangManager:
-    BACKOFF_S = 3.0
+    BACKOFF_S = 3.0          # base backoff after a gang rejection
+    BACKOFF_MAX_S = 120.0    # exponential cap (reference HandleTimeout
+                             # re-arms with growing delay across cycles)
 
     def __init__(self, store: Optional[Store] = None):
         self._mu = threading.RLock()
@@ -79,6 +83,7 @@ class GangManager:
                 g.members.discard(pod_key)
                 g.scheduled.discard(pod_key)
                 g.waiting.discard(pod_key)
+                g.waiting_since.pop(pod_key, None)
 
     def active_groups(self) -> Set[str]:
         """Groups still pursuing quorum (feeds the allocator's TTL sweep)."""
@@ -123,6 +128,7 @@ class GangManager:
         with self._mu:
             g = self.groups[key]
             g.waiting.add(pod.meta.key)
+            g.waiting_since.setdefault(pod.meta.key, time.time())
             if g.min_members and g.quorum_now < g.min_members:
                 return g.timeout_s
             return None  # quorum met: allow, and release the others
@@ -139,20 +145,57 @@ class GangManager:
         with self._mu:
             g = self.groups[key]
             g.waiting.discard(pod.meta.key)
+            g.waiting_since.pop(pod.meta.key, None)
             g.scheduled.add(pod.meta.key)
+            g.rejections = 0  # progress resets the backoff ladder
         self._flush_status(key)
 
     def reject_group(self, key: str):
         """Strict all-or-nothing: reject every waiting member and back the
-        group off (reference :262/:1099 + HandleTimeout :977)."""
+        group off EXPONENTIALLY across cycles (reference :262/:1099 +
+        HandleTimeout :977 — a gang that keeps missing quorum must not
+        re-burn a full scheduling cycle every few seconds)."""
 
         with self._mu:
             g = self.groups.get(key)
             if g is None:
                 return
             g.waiting.clear()
-            g.rejected_until = time.time() + self.BACKOFF_S
+            g.waiting_since.clear()
+            backoff = min(self.BACKOFF_S * (2 ** g.rejections),
+                          self.BACKOFF_MAX_S)
+            g.rejections += 1
+            g.rejected_until = time.time() + backoff
         self._flush_status(key)
+
+    def backoff_remaining(self, key: str,
+                          now: Optional[float] = None) -> float:
+        now = now if now is not None else time.time()
+        with self._mu:
+            g = self.groups.get(key)
+            return max(0.0, g.rejected_until - now) if g else 0.0
+
+    def sweep_timeouts(self, now: Optional[float] = None) -> Set[str]:
+        """Reject every group whose oldest waiting member exceeded the
+        gang timeout — the watchdog the scheduler loop calls so a
+        half-assembled gang never parks reserved GPUs forever (the
+        reference's timeout path frees Assume()d devices through
+        Unreserve; here the framework's unreserve hook fires when the
+        permit wait ends rejected)."""
+
+        now = now if now is not None else time.time()
+        expired: Set[str] = set()
+        with self._mu:
+            for key, g in self.groups.items():
+                if not g.waiting:
+                    continue
+                oldest = min(g.waiting_since.get(p, now)
+                             for p in g.waiting)
+                if now - oldest > g.timeout_s:
+                    expired.add(key)
+        for key in expired:
+            self.reject_group(key)
+        return expired
 
     # -------------------------------------------------------------- status
 

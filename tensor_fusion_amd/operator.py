@@ -136,6 +136,7 @@ class Operator:
                     self.scheduler.schedule_pending()
                     self.allocator.sync_dirty()
                     self.quota.sync_dirty()
+                    self.gang.sweep_timeouts()
                     self.allocator.sweep_stale_assumed(
                         gang_active=self.gang.active_groups())
                 except Exception:

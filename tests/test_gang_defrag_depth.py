@@ -1,0 +1,164 @@
+"""Edge behavior the reference spends code on: gang timeout backoff
+across cycles (manager.go HandleTimeout :977), defrag eviction-time
+revalidation + campaign abort (gpupool_defrag.go :686-897), and a
+larger-N joint-placement property sweep."""
+import random
+
+import tensor_fusion_amd.constants as C
+from tensor_fusion_amd.api.store import Store
+from tensor_fusion_amd.api.types import (GPU, AllocRequest, Pod, Resource)
+from tensor_fusion_amd.allocator.gpuallocator import GpuAllocator
+from tensor_fusion_amd.controllers.defrag import (AnnoEvictionMark,
+                                                  DefragController)
+from tensor_fusion_amd.gang.manager import GangManager
+
+
+def gang_pod(name, key="g1", minm=3, timeout="5"):
+    p = Pod()
+    p.meta.name = name
+    p.meta.namespace = "default"
+    p.meta.annotations.update({
+        C.AnnoGangEnabled: "true",
+        C.AnnoGangMinMembers: str(minm),
+        C.AnnoGangGroupKey: key,
+        C.AnnoGangTimeout: timeout,
+    })
+    return p
+
+
+class TestGangBackoff:
+    def test_exponential_backoff_across_rejections(self):
+        m = GangManager()
+        for i in range(3):
+            m.register_pod(gang_pod(f"p{i}"))
+        waits = []
+        for _ in range(4):
+            m.permit(gang_pod("p0"))
+            m.reject_group("g1")
+            waits.append(m.backoff_remaining("g1"))
+        # 3, 6, 12, 24 (within scheduling slop)
+        assert waits[1] > waits[0] * 1.5
+        assert waits[2] > waits[1] * 1.5
+        assert waits[3] > waits[2] * 1.5
+
+    def test_backoff_capped_and_reset_on_progress(self):
+        m = GangManager()
+        for i in range(3):
+            m.register_pod(gang_pod(f"p{i}"))
+        for _ in range(10):
+            m.permit(gang_pod("p0"))
+            m.reject_group("g1")
+        assert m.backoff_remaining("g1") <= GangManager.BACKOFF_MAX_S + 1
+        # progress (a member schedules) resets the ladder
+        g = m.groups["g1"]
+        g.rejected_until = 0.0
+        m.permit(gang_pod("p0"))
+        m.mark_scheduled(gang_pod("p0"))
+        m.permit(gang_pod("p1"))
+        m.reject_group("g1")
+        assert m.backoff_remaining("g1") <= GangManager.BACKOFF_S + 1
+
+    def test_sweep_timeouts_rejects_stale_waiters(self):
+        m = GangManager()
+        for i in range(3):
+            m.register_pod(gang_pod(f"p{i}", timeout="5"))
+        m.permit(gang_pod("p0", timeout="5"))  # waits (quorum 1/3)
+        import time
+        assert m.sweep_timeouts(now=time.time() + 1) == set()
+        expired = m.sweep_timeouts(now=time.time() + 10)
+        assert expired == {"g1"}
+        assert not m.groups["g1"].waiting
+
+
+def mk_world(nodes, gpus_per_node=4):
+    store = Store()
+    alloc = GpuAllocator(store=store)
+    for n in range(nodes):
+        for i in range(gpus_per_node):
+            g = GPU()
+            g.meta.name = f"n{n}-g{i}"
+            g.status.uuid = g.meta.name
+            g.status.node = f"n{n}"
+            g.status.capacity = Resource(2500.0, 288 << 30, 100.0)
+            g.status.available = Resource(2500.0, 288 << 30, 100.0)
+            alloc.upsert_gpu_for_testing(g)
+    return store, alloc
+
+
+def place(alloc, store, name, tflops, vram):
+    req = AllocRequest(pod_name=name, workload=f"{name}-wl",
+                      request=Resource(tflops, vram, 0.0),
+                      limit=Resource(tflops, vram, 0.0))
+    scores, _ = alloc.check_quota_and_filter(req)
+    assert scores, f"{name} unschedulable"
+    node = max(scores.values(), key=lambda s: s.score).node
+    gpus = alloc.pick_gpus(req, node)
+    alloc.assume(req, gpus)
+    alloc.commit(req.pod_key)
+    alloc.notify_bound(req.pod_key)
+    p = Pod()
+    p.meta.name = name
+    p.meta.namespace = "default"
+    p.status.node = node
+    store.create(p)
+    return req.pod_key, node
+
+
+class TestDefragRevalidation:
+    def test_eviction_cancelled_when_no_longer_placeable(self):
+        store, alloc = mk_world(2, gpus_per_node=1)
+        d = DefragController(store, alloc, utilization_threshold=0.5,
+                             eviction_ttl_s=0.0, campaign_cooldown_s=0.0)
+        # n0 holds a small pod (defrag candidate); n1 is mostly free
+        key, node = place(alloc, store, "small", 200.0, 20 << 30)
+        plan = d.run_campaign(now=1e9)
+        assert plan is not None and key in plan.evict_pods
+        # the target fills up during the TTL window
+        place(alloc, store, "filler", 2400.0, 280 << 30)
+        evicted = d.execute_due_evictions(now=2e9)
+        assert evicted == []  # revalidation refused the eviction
+        pod = store.get("Pod", "small", "default")
+        assert AnnoEvictionMark not in pod.meta.annotations  # mark cleared
+
+    def test_abort_campaign_clears_marks(self):
+        store, alloc = mk_world(3, gpus_per_node=1)
+        d = DefragController(store, alloc, utilization_threshold=0.5,
+                             eviction_ttl_s=60.0, campaign_cooldown_s=0.0)
+        place(alloc, store, "a", 200.0, 20 << 30)
+        plan = d.run_campaign(now=1e9)
+        assert plan is not None
+        assert d.abort_campaign() >= 1
+        assert all(AnnoEvictionMark not in p.meta.annotations
+                   for p in store.list("Pod"))
+        assert d.execute_due_evictions(now=2e9) == []
+
+
+class TestDefragAtScale:
+    def test_joint_placement_property_sweep(self):
+        """60 nodes x 4 GPUs, randomized fragmentation: every produced
+        plan must place ALL candidate allocations in simulation, and the
+        executed evictions must never exceed the plan."""
+
+        rng = random.Random(7)
+        store, alloc = mk_world(60, gpus_per_node=4)
+        placed = []
+        for i in range(150):
+            tf = rng.choice([100.0, 300.0, 600.0, 1200.0])
+            vr = rng.choice([8, 24, 48, 96]) << 30
+            try:
+                placed.append(place(alloc, store, f"w{i}", tf, vr))
+            except AssertionError:
+                break
+        d = DefragController(store, alloc, utilization_threshold=0.45,
+                             eviction_ttl_s=0.0, campaign_cooldown_s=0.0)
+        plan = d.run_campaign(now=1e9)
+        if plan is None:
+            return  # nothing compactable with this seed — acceptable
+        # invariant: every evictee has a simulated placement off-candidates
+        assert set(plan.evict_pods) == set(plan.placements)
+        for pk, gpus in plan.placements.items():
+            for gname in gpus:
+                g = alloc.gpu(gname)
+                assert g.status.node not in plan.candidate_nodes
+        evicted = d.execute_due_evictions(now=2e9)
+        assert set(evicted) <= set(plan.evict_pods)
