@@ -244,7 +244,7 @@ def test_tui_renders_device_and_worker_frame():
 
     from rich.console import Console
 
-    from tensor_fusion_amd.tui.app import build_frame
+    from tensor_fusion_amd.tui.app import UiState, build_frame
 
     devices = [{"index": 0, "uuid": "GPU-abc123", "vram_used": 24 << 30,
                 "vram_total": 288 << 30, "busy_percent": 42.0,
@@ -256,7 +256,10 @@ def test_tui_renders_device_and_worker_frame():
                           "block_ns": 3_000_000},
                 "heartbeat_ts": 0}]
     console = Console(record=True, width=120)
-    console.print(build_frame(devices, workers))
+    console.print(build_frame(UiState(view="devices", devices=devices,
+                                      workers=workers)))
+    console.print(build_frame(UiState(view="workers", devices=devices,
+                                      workers=workers)))
     text = console.export_text()
     assert "GPU-abc123"[:10] in text
     assert "288" in text and "24.0" in text
