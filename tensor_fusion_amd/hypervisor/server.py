@@ -79,11 +79,23 @@ def build_app(devices: DeviceController, workers: WorkerController,
                 m = devices.metrics(d.index)
             except Exception:
                 pass
+            nworkers = sum(
+                1 for st in workers.list()
+                if any(dev.uuid == d.uuid for dev in st.allocation.devices))
+            part = None
+            try:
+                part = devices.accel.compute_partition(d.index)
+            except Exception:
+                pass
             out.append({
                 "uuid": d.uuid, "index": d.index, "name": d.name,
                 "numa_node": d.numa_node, "vram_total": d.vram_total,
                 "compute_units": d.compute_units, "xcd_count": d.xcd_count,
                 "fp16_tflops": d.fp16_tflops, "is_mock": d.is_mock,
+                "vram_used": m.vram_used if m else 0,
+                "busy_percent": m.gfx_activity if m else 0,
+                "compute_partition": part or "SPX",
+                "worker_count": nworkers,
                 "metrics": None if m is None else {
                     "gfx_activity": m.gfx_activity,
                     "umc_activity": m.umc_activity,
@@ -94,7 +106,93 @@ def build_app(devices: DeviceController, workers: WorkerController,
 
     @app.get("/api/v1/workers")
     def get_workers():
+        """Full per-worker records (the TUI's worker view + shm
+        inspector read this shape)."""
+
+        import time as _t
+        out = []
+        for st in workers.list():
+            a = st.allocation
+            entry = None
+            try:
+                entry = st.shm.device(0)
+            except (OSError, ValueError):
+                pass
+            ns, name = st.key.split("/", 1)
+            out.append({
+                "namespace": ns, "pod": name,
+                "workload": a.spec.workload,
+                "qos": a.spec.qos, "isolation": a.spec.isolation,
+                "shm_path": st.shm.path,
+                "device_uuid": a.devices[0].uuid if a.devices else "",
+                # shm heartbeat is CLOCK_MONOTONIC ns → wall-clock ts
+                "heartbeat_ts": (
+                    _t.time() - max(0.0, _t.monotonic()
+                                    - st.shm.heartbeat() / 1e9)
+                    if st.shm and st.shm.heartbeat() else 0),
+                "pids": st.pids or [],
+                "flags": st.shm.flags() if st.shm else 0,
+                "limits": {
+                    "vram": entry.mem_limit_bytes if entry else
+                    a.spec.vram_limit,
+                    "compute_percent": a.up_limit_percent,
+                },
+                "usage": {
+                    "vram": entry.pod_memory_used if entry else 0,
+                    "vmm_bytes": entry.vmm_bytes if entry else 0,
+                    "erl_rate": entry.erl_refill_rate if entry else 0,
+                    "erl_capacity": entry.erl_capacity if entry else 0,
+                    "erl_tokens": entry.erl_tokens if entry else 0,
+                    "launches": entry.launch_count if entry else 0,
+                    "block_ns": entry.block_ns_total if entry else 0,
+                },
+            })
+        return {"success": True, "data": out}
+
+    @app.get("/api/v1/workers/metrics")
+    def get_worker_metrics():
+        """Flat join rows (the recorder's input shape, reference
+        worker metrics join :212-290)."""
+
         return {"success": True, "data": workers.worker_metrics()}
+
+    @app.get("/api/v1/metrics")
+    def get_metrics():
+        """Node metric rows for the TUI metrics view."""
+
+        rows = []
+        devs = devices.devices()
+        used = total = 0
+        for d in devs:
+            try:
+                m = devices.metrics(d.index)
+                used += m.vram_used
+            except Exception:
+                pass
+            total += d.vram_total
+        wl = workers.list()
+        launches = blocked = 0
+        detail = []
+        for st in wl:
+            try:
+                e = st.shm.device(0)
+                launches += e.launch_count
+                blocked += e.block_ns_total
+                detail.append(f"{st.key}:{e.launch_count}")
+            except (OSError, ValueError):
+                pass
+        rows = [
+            {"name": "devices", "value": len(devs), "detail": ""},
+            {"name": "workers", "value": len(wl), "detail": ""},
+            {"name": "vram used GiB",
+             "value": f"{used / (1 << 30):.1f}/{total / (1 << 30):.0f}",
+             "detail": ""},
+            {"name": "launches", "value": launches,
+             "detail": ", ".join(detail[:8])},
+            {"name": "throttled ms", "value": f"{blocked / 1e6:.0f}",
+             "detail": ""},
+        ]
+        return {"success": True, "data": rows}
 
     @app.get("/api/v1/pod")
     def get_pod(namespace: str = Query("default"), pod: str = Query(...)):
