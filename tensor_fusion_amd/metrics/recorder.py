@@ -143,13 +143,30 @@ class MetricsRecorder:
 
     def __init__(self, out_dir: str = "", encoder: str = "influx",
                  pricing: Optional[QosPricingTable] = None, tsdb=None,
-                 max_file_bytes: int = 32 << 20, keep_files: int = 3):
+                 max_file_bytes: int = 32 << 20, keep_files: int = 3,
+                 remote_url: str = "", remote_db: str = "tensor_fusion",
+                 remote_auth: str = ""):
         self.out_dir = out_dir
         self.encoder = ENCODERS[encoder]()
         self.pricing = pricing or QosPricingTable()
         self.tsdb = tsdb
         self.max_file_bytes = max_file_bytes
         self.keep_files = keep_files
+        # external pipeline: the reference ships metric files via Vector
+        # into GreptimeDB's influx-line HTTP write endpoint
+        # (charts/templates/vector-config.yaml:36-65). Here the recorder
+        # POSTs the same line protocol directly — remote_url is the
+        # /v1/influxdb/write (GreptimeDB) or /api/v2/write (influx)
+        # endpoint; failures never block the flush (buffered + retried
+        # next cycle, bounded).
+        self.remote_url = remote_url or os.environ.get(
+            "TF_METRICS_REMOTE_URL", "")
+        self.remote_db = remote_db
+        self.remote_auth = remote_auth or os.environ.get(
+            "TF_METRICS_REMOTE_AUTH", "")
+        self._remote_backlog: List[str] = []
+        self.remote_errors = 0
+        self.remote_posts = 0
         self._lock = threading.Lock()
         self.workers: Dict[str, WorkerMetrics] = {}
         self.nodes: Dict[str, NodeMetrics] = {}
@@ -258,7 +275,33 @@ class MetricsRecorder:
             self._rotate(path)
             with open(path, "a") as f:
                 f.write("\n".join(lines) + "\n")
+        if self.remote_url:
+            self._ship_remote(lines)
         return len(lines)
+
+    def _ship_remote(self, lines: List[str]):
+        """POST influx line protocol to the external TSDB; backlog on
+        failure (bounded to ~4k lines so a dead sink can't grow RAM)."""
+
+        import requests
+        payload = self._remote_backlog + lines
+        if not payload:
+            return
+        headers = {"Content-Type": "text/plain"}
+        if self.remote_auth:
+            headers["Authorization"] = self.remote_auth
+        try:
+            r = requests.post(self.remote_url,
+                              params={"db": self.remote_db},
+                              data="\n".join(payload).encode(),
+                              headers=headers, timeout=10)
+            if r.status_code >= 300:
+                raise RuntimeError(f"sink {r.status_code}")
+            self._remote_backlog = []
+            self.remote_posts += 1
+        except Exception:
+            self.remote_errors += 1
+            self._remote_backlog = payload[-4096:]
 
     def _rotate(self, path: str):
         try:

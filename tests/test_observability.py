@@ -287,3 +287,62 @@ def test_node_scaler_info_route():
     data = r.json()
     assert data["inFlightClaims"][0]["name"] == "claim-1"
     assert data["inFlightClaims"][0]["instanceType"] == "mi355x.8x"
+
+
+def test_metrics_remote_sink_ships_and_backlogs():
+    """External pipeline (reference: rolling file → Vector → GreptimeDB
+    influx write endpoint): the recorder POSTs the same line protocol
+    directly, with a bounded backlog across sink outages."""
+
+    import threading
+
+    received = []
+    fail = {"on": True}
+    from fastapi import FastAPI, Request, Response
+    app = FastAPI()
+
+    @app.post("/v1/influxdb/write")
+    async def write(req: Request):
+        if fail["on"]:
+            return Response(status_code=503)
+        received.append((await req.body()).decode())
+        return Response(status_code=204)
+
+    import socket
+    import time
+
+    import requests
+    import uvicorn
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
+    srv = uvicorn.Server(uvicorn.Config(app, host="127.0.0.1", port=port,
+                                        log_level="error"))
+    threading.Thread(target=srv.run, daemon=True).start()
+    for _ in range(100):
+        try:
+            requests.get(f"http://127.0.0.1:{port}/x", timeout=1)
+            break
+        except Exception:
+            time.sleep(0.02)
+    try:
+        from tensor_fusion_amd.metrics.recorder import (MetricsRecorder,
+                                                        WorkerMetrics)
+        rec = MetricsRecorder(
+            remote_url=f"http://127.0.0.1:{port}/v1/influxdb/write",
+            remote_auth="token abc")
+        rec.set_worker(WorkerMetrics(namespace="ns", worker="w1",
+                                     workload="wl", qos="medium",
+                                     compute_tflops=10.0,
+                                     vram_bytes=1 << 30))
+        # sink down: flush backlogs, doesn't raise
+        rec.flush()
+        assert rec.remote_errors == 1 and not received
+        # sink recovers: backlog + new lines ship together
+        fail["on"] = False
+        rec.flush()
+        assert rec.remote_posts == 1
+        assert received and "tf_worker_metrics" in received[0]
+        assert received[0].count("tf_worker_metrics") >= 2  # backlog shipped
+    finally:
+        srv.should_exit = True
