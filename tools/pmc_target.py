@@ -28,6 +28,12 @@ def main():
         fused.skinny_gemm(x, w)
     torch.cuda.synchronize()
 
+    # packed-layout GEMV (round 2: the hipBLASLt-beating variant)
+    wp = fused.pack_skinny_weight(w)
+    for _ in range(REPS):
+        fused.skinny_gemm_packed(x, wp, w.shape[0])
+    torch.cuda.synchronize()
+
     # fused rmsnorm / add+rmsnorm at the decode row shape
     h = torch.randn(8, 4096, device=dev, dtype=torch.bfloat16)
     r = torch.randn(8, 4096, device=dev, dtype=torch.bfloat16)
