@@ -47,8 +47,8 @@ class Allocation:
     bound: bool = False
 
 
-@dataclass
-class NodeScore:
+@dataclass(slots=True)
+class NodeScore:  # slots: ~1000 of these are built per PreFilter call
     node: str
     score: float
     gpu_scores: Dict[str, float]  # gpu name -> score
@@ -121,6 +121,18 @@ class GpuAllocator:
             self._ingest(g)
 
     # ------------------------------------------------------------ queries
+
+    def gpu_uuid(self, name: str) -> str:
+        """UUID without the deepcopy gpu() pays (PreBind hot path)."""
+
+        with self._mu:
+            g = self._gpus.get(name)
+            return g.status.uuid if g else ""
+
+    def gpu_capacity_tflops(self, name: str) -> float:
+        with self._mu:
+            g = self._gpus.get(name)
+            return g.status.capacity.tflops if g else 0.0
 
     def gpu(self, name: str) -> Optional[GPU]:
         with self._mu:

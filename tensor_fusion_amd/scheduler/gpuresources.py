@@ -193,10 +193,8 @@ class GPUResourcesFit(Plugin):
             alloc = self.allocator.commit(req.pod_key)
         except AllocationError as e:
             return Status.error(str(e))
-        gpu_uuids = []
-        for n in alloc.gpu_names:
-            g = self.allocator.gpu(n)
-            gpu_uuids.append(g.status.uuid if g else n)
+        gpu_uuids = [self.allocator.gpu_uuid(n) or n
+                     for n in alloc.gpu_names]
         annos = {
             C.AnnoGpuIds: ",".join(gpu_uuids),
             C.AnnoContainerGpus: ",".join(alloc.gpu_names),
@@ -205,8 +203,8 @@ class GPUResourcesFit(Plugin):
             idx = self.index_allocator.occupy(node, pod.meta.key)
             annos[C.AnnoPodIndex] = str(idx)
         if req.isolation_mode == C.IsolationHard:
-            gpu0 = self.allocator.gpu(alloc.gpu_names[0])
-            cap_tf = gpu0.status.capacity.tflops if gpu0 else C.MI355X_BF16_TFLOPS
+            cap_tf = self.allocator.gpu_capacity_tflops(
+                alloc.gpu_names[0]) or C.MI355X_BF16_TFLOPS
             pct = req.limit.compute_percent or \
                 (100.0 * req.limit.tflops / cap_tf if cap_tf else 100.0)
             mask, cus = cu_mask_for_percent(max(pct, 0.5))
