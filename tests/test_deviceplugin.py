@@ -229,3 +229,56 @@ class TestCheckpointDetector:
         assert "n0" in scores
         picked = alloc.pick_gpus(req, "n0")
         assert picked == ["n0-g1"]  # external GPU never allocated
+
+
+class FakePodResourcesLister(grpc.GenericRpcHandler):
+    """kubelet's PodResources service (the half kubelet implements)."""
+
+    def __init__(self, pods):
+        from tensor_fusion_amd.k8s.podresources import MSG as PR
+        self.PR = PR
+        self.pods = pods
+
+    def service(self, details):
+        from tensor_fusion_amd.k8s.podresources import M_LIST
+        if details.method == M_LIST:
+            return grpc.unary_unary_rpc_method_handler(
+                self._list,
+                request_deserializer=self.PR[
+                    "ListPodResourcesRequest"].FromString,
+                response_serializer=lambda m: m.SerializeToString())
+        return None
+
+    def _list(self, req, ctx):
+        resp = self.PR["ListPodResourcesResponse"]()
+        for ns, name, res, ids in self.pods:
+            pr = resp.pod_resources.add(name=name, namespace=ns)
+            c = pr.containers.add(name="main")
+            d = c.devices.add(resource_name=res)
+            d.device_ids.extend(ids)
+        return resp
+
+
+class TestPodResources:
+    def test_list_and_maps(self, tmp_path):
+        from tensor_fusion_amd.k8s.podresources import PodResourcesClient
+        sock = str(tmp_path / "podres.sock")
+        server = grpc.server(ThreadPoolExecutor(max_workers=2))
+        server.add_generic_rpc_handlers((FakePodResourcesLister([
+            ("default", "w0", "tensor-fusion.ai/index-3", ["3-1"]),
+            ("default", "legacy", "amd.com/gpu", ["uuid-aaa"]),
+            ("kube-system", "dns", "cpu", []),
+        ]),))
+        server.add_insecure_port(f"unix://{sock}")
+        server.start()
+        try:
+            cli = PodResourcesClient(sock)
+            pods = cli.list()
+            assert len(pods) == 3
+            # our index devices
+            assert cli.device_map() == {"default/w0": ["3-1"]}
+            # foreign GPU holders (live complement to the checkpoint file)
+            assert cli.foreign_gpu_devices() == {
+                "default/legacy": ["uuid-aaa"]}
+        finally:
+            server.stop(0.2)
