@@ -96,6 +96,10 @@ struct Client {
   std::mutex reg_mu;
   std::unordered_map<void*, FatBin*> fatbins;        // handle → fatbin
   std::unordered_map<const void*, std::pair<FatBin*, std::string>> stubs;
+  // device globals: host shadow ptr → (fatbin, device name, size);
+  // resolved worker addresses cached in dvar_addr
+  std::unordered_map<const void*, std::pair<FatBin*, std::string>> dvars;
+  std::unordered_map<const void*, std::pair<uint64_t, uint64_t>> dvar_addr;
   std::unordered_map<const void*, KernLayout*> launch_cache;
   std::unordered_map<uint64_t, KernLayout*> module_fn_layouts;  // worker fn
   uint64_t next_image_id = 1;
@@ -1343,7 +1347,14 @@ void __hipRegisterFunction(void** handle, const void* hostFun, char*,
   c.stubs[hostFun] = {it->second, deviceName};
 }
 
-void __hipRegisterVar(void**, void*, char*, char*, int, size_t, int, int) {}
+void __hipRegisterVar(void** handle, void* var, char*, char* deviceVar,
+                      int, size_t, int, int) {
+  Client& c = C();
+  std::lock_guard<std::mutex> l(c.reg_mu);
+  auto it = c.fatbins.find(handle);
+  if (it == c.fatbins.end()) return;
+  c.dvars[var] = {it->second, deviceVar};
+}
 void __hipRegisterManagedVar(void**, void**, void*, const char*, size_t,
                              unsigned) {}
 void __hipRegisterSurface(void**, void*, char*, char*, int, int) {}
@@ -1929,6 +1940,90 @@ extern "C" hipError_t hipMemPoolTrimTo(void* pool, size_t keep) {
   return hipSuccess;
 }
 
+// ---- device globals (__device__ vars): __hipRegisterVar recorded the
+// host shadow → (fatbin, name); resolution ships the module (if not
+// already) and asks the worker for hipModuleGetGlobal. The range is
+// registered in dev_ranges so plain memcpys route to the device.
+
+bool resolve_symbol(const void* symbol, uint64_t* dptr, uint64_t* bytes) {
+  Client& c = C();
+  {
+    std::lock_guard<std::mutex> l(c.reg_mu);
+    auto hit = c.dvar_addr.find(symbol);
+    if (hit != c.dvar_addr.end()) {
+      *dptr = hit->second.first;
+      *bytes = hit->second.second;
+      return true;
+    }
+  }
+  FatBin* fb = nullptr;
+  std::string name;
+  {
+    std::lock_guard<std::mutex> l(c.reg_mu);
+    auto it = c.dvars.find(symbol);
+    if (it == c.dvars.end()) return false;
+    fb = it->second.first;
+    name = it->second.second;
+  }
+  uint64_t mod = ship_image(fb);
+  if (!mod) return false;
+  std::vector<uint8_t> body(8 + name.size() + 1);
+  memcpy(body.data(), &mod, 8);
+  memcpy(body.data() + 8, name.c_str(), name.size() + 1);
+  struct {
+    uint64_t dptr, bytes;
+  } r{0, 0};
+  hipError_t e = send_sync(OP_GET_GLOBAL, body.data(), body.size(), &r,
+                           sizeof r);
+  if (e != hipSuccess || !r.dptr) return false;
+  {
+    std::lock_guard<std::mutex> l(c.reg_mu);
+    c.dvar_addr[symbol] = {r.dptr, r.bytes};
+  }
+  {
+    std::lock_guard<std::mutex> l(c.mem_mu);
+    c.dev_ranges[r.dptr] = r.bytes;
+  }
+  *dptr = r.dptr;
+  *bytes = r.bytes;
+  return true;
+}
+
+extern "C" hipError_t hipGetSymbolAddress(void** devPtr,
+                                          const void* symbol) {
+  uint64_t d = 0, b = 0;
+  if (!resolve_symbol(symbol, &d, &b)) return 1 /*hipErrorInvalidValue*/;
+  *devPtr = (void*)d;
+  return hipSuccess;
+}
+
+extern "C" hipError_t hipGetSymbolSize(size_t* size, const void* symbol) {
+  uint64_t d = 0, b = 0;
+  if (!resolve_symbol(symbol, &d, &b)) return 1;
+  *size = (size_t)b;
+  return hipSuccess;
+}
+
+extern "C" hipError_t hipMemcpyToSymbol(const void* symbol, const void* src,
+                                        size_t count, size_t offset,
+                                        int kind) {
+  uint64_t d = 0, b = 0;
+  if (!resolve_symbol(symbol, &d, &b)) return 1;
+  if (offset + count > b && b) return 1;
+  return hipMemcpy((void*)(d + offset), src, count,
+                   kind == 0 ? 1 /*default for ToSymbol is H2D*/ : kind);
+}
+
+extern "C" hipError_t hipMemcpyFromSymbol(void* dst, const void* symbol,
+                                          size_t count, size_t offset,
+                                          int kind) {
+  uint64_t d = 0, b = 0;
+  if (!resolve_symbol(symbol, &d, &b)) return 1;
+  if (offset + count > b && b) return 1;
+  return hipMemcpy(dst, (const void*)(d + offset), count,
+                   kind == 0 ? 2 /*default FromSymbol is D2H*/ : kind);
+}
+
 #define TF_NOTSUP(name, ...)                         \
   hipError_t name(__VA_ARGS__) {                     \
     fprintf(stderr, "[tf-client] %s: not supported over remoting yet\n", \
@@ -1936,8 +2031,6 @@ extern "C" hipError_t hipMemPoolTrimTo(void* pool, size_t keep) {
     return hipErrorNotSupported;                     \
   }
 
-TF_NOTSUP(hipGetSymbolAddress, void**, const void*)
-TF_NOTSUP(hipMemcpyToSymbol, const void*, const void*, size_t, size_t, int)
 TF_NOTSUP(hipIpcGetMemHandle, void*, void*)
 TF_NOTSUP(hipIpcOpenMemHandle, void**, const void*, unsigned)
 TF_NOTSUP(hipIpcCloseMemHandle, void*)

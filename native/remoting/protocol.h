@@ -153,6 +153,7 @@ enum Op : uint32_t {
   OP_MEMPOOL_SET_ATTR,   // {pool, attr, value} → {err}
   OP_MEMPOOL_GET_ATTR,   // {pool, attr} → {value}
   OP_MEMPOOL_TRIM,       // {pool, keep} async
+  OP_GET_GLOBAL,         // {module, name[]} → reply {dptr, size}
 };
 
 constexpr uint32_t F_WANT_REPLY = 1u << 0;

@@ -20,6 +20,13 @@
     }                                                               \
   } while (0)
 
+__device__ float g_scale[4] = {1.f, 2.f, 3.f, 4.f};
+
+__global__ void scale_by_global(float* y, int n) {
+  int i = blockIdx.x * blockDim.x + threadIdx.x;
+  if (i < n) y[i] *= g_scale[i & 3];
+}
+
 __global__ void saxpy(float a, const float* x, float* y, int n) {
   int i = blockIdx.x * blockDim.x + threadIdx.x;
   if (i < n) y[i] = a * x[i] + y[i];
@@ -133,11 +140,33 @@ int main() {
     printf("small_h2d_us=%.3f\n", h2d_us);
   }
 
+  // device-global symbols: hipMemcpyToSymbol / GetSymbolAddress /
+  // FromSymbol (exercises __hipRegisterVar + OP_GET_GLOBAL remotely)
+  bool sym_ok = false;
+  {
+    float vals[4] = {10.f, 20.f, 30.f, 40.f};
+    CK(hipMemcpyToSymbol(HIP_SYMBOL(g_scale), vals, sizeof vals, 0,
+                         hipMemcpyHostToDevice));
+    float four[4] = {7.f, 7.f, 7.f, 7.f};
+    CK(hipMemcpy(dy, four, sizeof four, hipMemcpyHostToDevice));
+    hipLaunchKernelGGL(scale_by_global, dim3(1), dim3(4), 0, 0, dy, 4);
+    float got[4] = {0};
+    CK(hipMemcpy(got, dy, sizeof got, hipMemcpyDeviceToHost));
+    float back[4] = {0};
+    CK(hipMemcpyFromSymbol(back, HIP_SYMBOL(g_scale), sizeof back, 0,
+                           hipMemcpyDeviceToHost));
+    void* addr = nullptr;
+    CK(hipGetSymbolAddress(&addr, HIP_SYMBOL(g_scale)));
+    sym_ok = got[0] == 70.f && got[3] == 280.f && back[2] == 30.f &&
+             addr != nullptr;
+    printf("symbol_ok=%d\n", (int)sym_ok);
+  }
+
   CK(hipFree(dx));
   CK(hipFree(dy));
   CK(hipFree(dsum));
   CK(hipDeviceSynchronize());
-  bool ok = rel > -1e-3 && rel < 1e-3 && probe == 0.0f;
+  bool ok = rel > -1e-3 && rel < 1e-3 && probe == 0.0f && sym_ok;
   printf(ok ? "TESTAPP_OK\n" : "TESTAPP_FAIL\n");
   return ok ? 0 : 1;
 }

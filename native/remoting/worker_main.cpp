@@ -118,6 +118,8 @@ struct Hip {
   hipError_t (*MemPoolSetAttribute)(void*, int, void*) = nullptr;
   hipError_t (*MemPoolGetAttribute)(void*, int, void*) = nullptr;
   hipError_t (*MemPoolTrimTo)(void*, size_t) = nullptr;
+  hipError_t (*ModuleGetGlobal)(void**, size_t*, void*, const char*) =
+      nullptr;
 
   bool load() {
     const char* names[] = {"libamdhip64.so", "libamdhip64.so.7",
@@ -190,6 +192,7 @@ struct Hip {
     O(MemPoolSetAttribute, "hipMemPoolSetAttribute")
     O(MemPoolGetAttribute, "hipMemPoolGetAttribute")
     O(MemPoolTrimTo, "hipMemPoolTrimTo")
+    O(ModuleGetGlobal, "hipModuleGetGlobal")
 #undef O
     return true;
   }
@@ -1287,6 +1290,21 @@ void handle(tfrpc::CmdHdr* c, uint8_t* body) {
           ? hip.MemPoolTrimTo((void*)b.pool, b.keep) : 801;
       if (c->flags & F_WANT_REPLY) reply(c->seq, e, nullptr, 0);
       else set_sticky(e, "hipMemPoolTrimTo");
+      break;
+    }
+    case OP_GET_GLOBAL: {
+      uint64_t module;
+      memcpy(&module, body, 8);
+      const char* name = (const char*)body + 8;
+      void* dptr = nullptr;
+      size_t bytes = 0;
+      hipError_t e = hip.ModuleGetGlobal
+          ? hip.ModuleGetGlobal(&dptr, &bytes, (void*)xl(module), name)
+          : 801;
+      struct {
+        uint64_t dptr, bytes;
+      } r{(uint64_t)dptr, (uint64_t)bytes};
+      reply(c->seq, e, &r, sizeof r);
       break;
     }
     case OP_SHUTDOWN:
