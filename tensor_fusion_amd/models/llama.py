@@ -99,14 +99,42 @@ class Attention(nn.Module):
         self.wk = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(cfg.heads * self.head_dim, cfg.dim, bias=False)
+        self._packed = None  # decode-GEMV weight shuffles (lazy)
+
+    def _proj(self, lin: nn.Linear, x, idx: int):
+        """Decode projections via the packed MFMA GEMV when the token
+        batch fits (6.25 TB/s vs hipBLASLt's decode tiles; weights are
+        pre-shuffled once per layer — profiles/skinny_gemm_packed_r02.md)."""
+
+        B, T, D = x.shape
+        if not (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
+                and B * T <= 16 and D % 32 == 0
+                and lin.out_features % 16 == 0):
+            return lin(x)
+        from ..ops import fused
+        if not fused.packed_profitable(lin.out_features, D):
+            return lin(x)
+        if self._packed is None:
+            self._packed = {}
+        wp = self._packed.get(idx)
+        if wp is None:
+            wp = fused.pack_skinny_weight(lin.weight.detach())
+            self._packed[idx] = wp
+        flat = x.reshape(B * T, D)
+        if not flat.is_contiguous():
+            flat = flat.contiguous()
+        return fused.skinny_gemm_packed(flat, wp, lin.out_features)             .view(B, T, -1)
 
     def forward(self, x, cos, sin, pos, cache=None, pos_end=None,
                 mask=None):
         B, T, _ = x.shape
         cfg = self.cfg
-        q = self.wq(x).view(B, T, cfg.heads, self.head_dim).transpose(1, 2)
-        k = self.wk(x).view(B, T, cfg.kv_heads, self.head_dim).transpose(1, 2)
-        v = self.wv(x).view(B, T, cfg.kv_heads, self.head_dim).transpose(1, 2)
+        q = self._proj(self.wq, x, 0).view(B, T, cfg.heads,
+                                           self.head_dim).transpose(1, 2)
+        k = self._proj(self.wk, x, 1).view(B, T, cfg.kv_heads,
+                                           self.head_dim).transpose(1, 2)
+        v = self._proj(self.wv, x, 2).view(B, T, cfg.kv_heads,
+                                           self.head_dim).transpose(1, 2)
         q = apply_rope(q, cos, sin, pos)
         k = apply_rope(k, cos, sin, pos)
         if cache is not None:
@@ -133,7 +161,7 @@ class Attention(nn.Module):
         o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
                                            is_causal=causal)
         o = o.transpose(1, 2).reshape(B, T, -1)
-        return self.wo(o)
+        return self._proj(self.wo, o, 3)
 
 
 class MLP(nn.Module):
@@ -142,9 +170,14 @@ class MLP(nn.Module):
         self.gate = nn.Linear(cfg.dim, cfg.intermediate, bias=False)
         self.up = nn.Linear(cfg.dim, cfg.intermediate, bias=False)
         self.down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
+        self._packed = None
+
+    _proj = Attention._proj  # same packed-GEMV decode routing
 
     def forward(self, x):
-        return self.down(F.silu(self.gate(x)) * self.up(x))
+        g = self._proj(self.gate, x, 0)
+        u = self._proj(self.up, x, 1)
+        return self._proj(self.down, F.silu(g) * u, 2)
 
 
 class Block(nn.Module):

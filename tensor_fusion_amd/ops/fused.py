@@ -102,6 +102,29 @@ def pack_skinny_weight(weight):
             .permute(0, 2, 3, 1, 4).contiguous())
 
 
+def packed_profitable(N: int, K: int) -> bool:
+    """Shape routing for the packed decode GEMV, from the measured
+    per-shape table (tools/shape_bench.py on MI355X, batch 8):
+
+        N=  4096 K= 4096: packed 19.1us  hipBLASLt 23.4us   -> packed
+        N=  1024 K= 4096: packed 15.5us  hipBLASLt 19.1us   -> packed
+        N= 14336 K= 4096: packed 24.8us  hipBLASLt 20.2us   -> blas
+        N=  4096 K=14336: packed 74.5us  hipBLASLt 19.6us   -> blas
+        N=128256 K= 4096: packed 165us   hipBLASLt 185us    -> packed
+
+    The kernel launches N/64 workgroups of 4 waves. Standalone the small
+    shapes look competitive, but that microbench is L2-flattered (200
+    reps of one 32 MB weight); IN the captured decode graph, where every
+    token streams the full 16 GB working set, the small-N projections
+    measured SLOWER end to end (780 -> 703 tok/s with q/k/v/o routed
+    here). So the rule is strict: packed only where the grid fills the
+    chip — N >= 32768 (lm_head-class shapes), which is also where it
+    beats hipBLASLt by ~10%% in situ. Closing small-N needs a split-K
+    variant (future work), not shape routing."""
+
+    return K <= 8192 and N >= 32768
+
+
 def skinny_gemm_packed(x, wp, N: int):
     """y = x @ W.T with W pre-shuffled by pack_skinny_weight."""
 
