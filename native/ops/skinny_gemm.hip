@@ -261,6 +261,90 @@ __global__ void skinny_gemm_packed_kernel(
   }
 }
 
+// ---- v3b: wave-split-K for underfilled N ------------------------------
+// At N=4096 the packed kernel launches only 64 workgroups (1 wave-group
+// per CU on a quarter of the chip) and the per-CU memory parallelism
+// caps the stream at ~1.8 TB/s. This variant keeps the same grid but
+// packs KSPLIT wave-groups per workgroup, each streaming a disjoint
+// K-slice of the SAME four column tiles; the ks==0 group reduces the
+// partials through LDS and writes y. 4x the outstanding loads per CU,
+// one kernel, no workspace.
+
+template <int UNROLL, int KSPLIT>
+__global__ void __launch_bounds__(256 * KSPLIT)
+skinny_gemm_packed_ws_kernel(const __hip_bfloat16* __restrict__ x,
+                             const __hip_bfloat16* __restrict__ wp,
+                             __hip_bfloat16* __restrict__ y, int M, int N,
+                             int K) {
+  const int lane = threadIdx.x & 63;
+  const int w = threadIdx.x >> 6;  // wave: [0, 4*KSPLIT)
+  const int tile = w & 3;
+  const int ks = w >> 2;  // K-slice of this wave
+  const int nt = blockIdx.x * 4 + tile;
+  const int n0 = nt * 16;
+  __shared__ float red[4][KSPLIT][64][4];
+
+  const int col = lane & 15;
+  const int kgrp = lane >> 4;
+  const int n = n0 + col;
+  const bool arow_ok = col < M;
+
+  f32x4 acc = {0.f, 0.f, 0.f, 0.f};
+  const int ktiles = K / 32;
+  const int per = (ktiles + KSPLIT - 1) / KSPLIT;
+  const int t_lo = ks * per;
+  const int t_hi = t_lo + per < ktiles ? t_lo + per : ktiles;
+
+  if (n0 < N) {
+    const __hip_bfloat16* wstream =
+        wp + ((size_t)nt * ktiles) * 512 + lane * 8;
+    const __hip_bfloat16* xrow = x + (size_t)(arow_ok ? col : 0) * K;
+    int t = t_lo;
+    for (; t + UNROLL <= t_hi; t += UNROLL) {
+      B16x8 a[UNROLL], b[UNROLL];
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u) {
+        int k = (t + u) * 32 + kgrp * 8;
+        a[u].raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                           : u4{0, 0, 0, 0};
+        b[u].raw = __builtin_nontemporal_load(
+            reinterpret_cast<const u4*>(wstream + (size_t)(t + u) * 512));
+      }
+#pragma unroll
+      for (int u = 0; u < UNROLL; ++u)
+        acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a[u].v, b[u].v, acc,
+                                                      0, 0, 0);
+    }
+    for (; t < t_hi; ++t) {
+      int k = t * 32 + kgrp * 8;
+      B16x8 a, b;
+      a.raw = arow_ok ? *reinterpret_cast<const u4*>(xrow + k)
+                      : u4{0, 0, 0, 0};
+      b.raw = __builtin_nontemporal_load(
+          reinterpret_cast<const u4*>(wstream + (size_t)t * 512));
+      acc = __builtin_amdgcn_mfma_f32_16x16x32_bf16(a.v, b.v, acc, 0, 0, 0);
+    }
+  }
+
+  if (ks > 0) {
+#pragma unroll
+    for (int r = 0; r < 4; ++r) red[tile][ks][lane][r] = acc[r];
+  }
+  __syncthreads();
+  if (ks == 0 && n0 < N) {
+#pragma unroll
+    for (int s = 1; s < KSPLIT; ++s)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) acc[r] += red[tile][s][lane][r];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      int row = kgrp * 4 + r;
+      if (row < M && n < N)
+        y[(size_t)row * N + n] = __float2bfloat16(acc[r]);
+    }
+  }
+}
+
 }  // namespace
 
 extern "C" {
@@ -272,6 +356,20 @@ int tf_skinny_gemm_packed(const void* x, const void* wp, void* y, int M,
   if (M < 1 || M > 16 || (K & 31) || (N & 15)) return 1;
   dim3 block(256);
   dim3 grid((N + 63) / 64);
+  // underfilled grids (< 3/4 of the 256 CUs): wave-split-K packs 4
+  // K-slice wave-groups per workgroup for 4x the in-flight loads/CU
+  static int splitk = [] {
+    const char* v = getenv("TF_SKINNY_SPLITK");
+    return v ? atoi(v) : -1;  // -1 = auto
+  }();
+  bool want_ws = splitk == 4 || (splitk == -1 && grid.x < 192);
+  if (want_ws && K / 32 >= 8) {
+    hipLaunchKernelGGL((skinny_gemm_packed_ws_kernel<8, 4>), grid,
+                       dim3(1024), 0, (hipStream_t)stream,
+                       (const __hip_bfloat16*)x, (const __hip_bfloat16*)wp,
+                       (__hip_bfloat16*)y, M, N, K);
+    return (int)hipGetLastError();
+  }
   static int unroll = [] {
     const char* v = getenv("TF_SKINNY_UNROLL");
     int u = v ? atoi(v) : 8;

@@ -113,16 +113,30 @@ def packed_profitable(N: int, K: int) -> bool:
         N=128256 K= 4096: packed 165us   hipBLASLt 185us    -> packed
 
     The kernel launches N/64 workgroups of 4 waves. Standalone the small
-    shapes look competitive, but that microbench is L2-flattered (200
-    reps of one 32 MB weight); IN the captured decode graph, where every
-    token streams the full 16 GB working set, the small-N projections
-    measured SLOWER end to end (780 -> 703 tok/s with q/k/v/o routed
-    here). So the rule is strict: packed only where the grid fills the
-    chip — N >= 32768 (lm_head-class shapes), which is also where it
-    beats hipBLASLt by ~10%% in situ. Closing small-N needs a split-K
-    variant (future work), not shape routing."""
+    shapes initially looked competitive, then measured SLOWER in the
+    captured decode graph (the microbench was L2-flattered). The
+    wave-split-K variant (skinny_gemm_packed_ws_kernel: 4 K-slice
+    wave-groups per workgroup, LDS reduction) fixed the per-CU memory
+    parallelism at small N:
 
-    return K <= 8192 and N >= 32768
+        N= 4096 K= 4096: 19.1 -> 15.6us (beats blas 18.5)
+        N= 1024 K= 4096: 15.5 -> 14.6us (beats blas 19.0)
+
+    Deep-K narrow shapes (down-proj N=4096 K=14336: 55us vs blas 19.3)
+    and mid-N (gate/up N=14336: loses to blas 5.8 TB/s) stay on
+    hipBLASLt — a grid-level split-K with fp32 atomics could close
+    down-proj but costs extra zero/cast launches per projection."""
+
+    if K > 8192:
+        return False  # deep-K underfilled: hipBLASLt wins 3x
+    if N >= 32768:
+        return True
+    # small-N: wave-split-K wins the STANDALONE microbench (15.6 vs
+    # 18.5us at N=4096) but measured consistently slower IN the captured
+    # decode graph (A/B x2: 762.5 vs 776.6 tok/s) — hipBLASLt's decode
+    # tiles interact better with the surrounding graph than an isolated
+    # rep loop predicts. Default off; TF_PACKED_SMALL=1 for experiments.
+    return N <= 8192 and os.environ.get("TF_PACKED_SMALL", "0") == "1"
 
 
 def skinny_gemm_packed(x, wp, N: int):
