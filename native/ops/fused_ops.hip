@@ -105,9 +105,94 @@ __global__ void add_rmsnorm_kernel(const u4* __restrict__ x,
   }
 }
 
+// ---- fused RoPE + KV-cache write --------------------------------------
+// The decode census (profiles/pmc_ktrace_r02.md) showed ~4.2 ms/step of
+// rope/index elementwise soup — ~260 4.7 us kernels per step, as much
+// time as all GEMMs. This kernel replaces apply_rope(q) + apply_rope(k)
+// + the two cache index-writes of one attention layer with ONE launch:
+//
+//   qout[b,h,t,d]              = rope(qin[b,t,h*D+d], pos[t])
+//   kcache[b,hk,pos[t],d]      = rope(kin[b,t,hk*D+d], pos[t])
+//   vcache[b,hk,pos[t],d]      = vin[b,t,hk*D+d]
+//
+// Rope pairing is interleaved (x1=x[...,::2], x2=x[...,1::2]) against
+// cos/sin[pos, j] fp32 tables — numerically identical math to the
+// reference implementation (fp32 multiply, bf16 store).
+
+__global__ void rope_qkv_cache_kernel(
+    const u4* __restrict__ qin,   // [B,T,Hq*D] bf16
+    const u4* __restrict__ kin,   // [B,T,Hk*D]
+    const u4* __restrict__ vin,   // [B,T,Hk*D]
+    const float* __restrict__ cosT,  // [S, D/2]
+    const float* __restrict__ sinT,
+    const long long* __restrict__ pos,  // [T]
+    u4* __restrict__ qout,        // [B,Hq,T,D]
+    u4* __restrict__ kcache,      // [B,Hk,S,D]
+    u4* __restrict__ vcache,
+    int B, int T, int Hq, int Hk, int D, int S) {
+  // grid.x = B*T, grid.y = Hq + 2*Hk; 64 threads each handling 8 elems
+  // (4 rope pairs) — D % 8 == 0, D <= 512
+  const int bt = blockIdx.x;
+  const int b = bt / T, t = bt % T;
+  int h = blockIdx.y;
+  const int i8 = threadIdx.x;         // 8-element group index
+  if (i8 * 8 >= D) return;
+  const long long p = pos[t];
+  const int half = D / 2;
+
+  auto rope8 = [&](V8 in, int d0) {
+    V8 o;
+#pragma unroll
+    for (int k = 0; k < 8; k += 2) {
+      int j = (d0 + k) >> 1;  // pair index
+      float c = cosT[(size_t)p * half + j];
+      float sn = sinT[(size_t)p * half + j];
+      float x1 = __bfloat162float(in.h[k]);
+      float x2 = __bfloat162float(in.h[k + 1]);
+      o.h[k] = __float2bfloat16(x1 * c - x2 * sn);
+      o.h[k + 1] = __float2bfloat16(x2 * c + x1 * sn);
+    }
+    return o;
+  };
+
+  const int d0 = i8 * 8;
+  if (h < Hq) {  // q head → rope → qout[b,h,t,:]
+    V8 in{qin[((size_t)(b * T + t) * Hq * D + h * D + d0) / 8]};
+    V8 o = rope8(in, d0);
+    qout[((size_t)((b * Hq + h) * T + t) * D + d0) / 8] = o.v;
+  } else if (h < Hq + Hk) {  // k head → rope → kcache[b,hk,p,:]
+    int hk = h - Hq;
+    V8 in{kin[((size_t)(b * T + t) * Hk * D + hk * D + d0) / 8]};
+    V8 o = rope8(in, d0);
+    kcache[((size_t)((b * Hk + hk) * S + p) * D + d0) / 8] = o.v;
+  } else {  // v head → vcache
+    int hk = h - Hq - Hk;
+    vcache[((size_t)((b * Hk + hk) * S + p) * D + d0) / 8] =
+        vin[((size_t)(b * T + t) * Hk * D + hk * D + d0) / 8];
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+// One launch per attention layer: rope(q) → qout, rope(k)/copy(v) →
+// caches at pos. All tensors bf16 except cos/sin (fp32) and pos (i64).
+int tf_rope_qkv_cache(const void* qin, const void* kin, const void* vin,
+                      const void* cos_t, const void* sin_t,
+                      const void* pos, void* qout, void* kcache,
+                      void* vcache, int B, int T, int Hq, int Hk, int D,
+                      int S, void* stream) {
+  if (D % 8 || D > 512) return 1;
+  dim3 grid(B * T, Hq + 2 * Hk);
+  dim3 block((D + 7) / 8);
+  hipLaunchKernelGGL(rope_qkv_cache_kernel, grid, block, 0,
+                     (hipStream_t)stream, (const u4*)qin, (const u4*)kin,
+                     (const u4*)vin, (const float*)cos_t,
+                     (const float*)sin_t, (const long long*)pos, (u4*)qout,
+                     (u4*)kcache, (u4*)vcache, B, T, Hq, Hk, D, S);
+  return (int)hipGetLastError();
+}
 
 // x,w,out bf16; dim % 8 == 0. One workgroup per row.
 int tf_rmsnorm(const void* x, const void* w, void* out, int rows, int dim,

@@ -129,6 +129,36 @@ class Attention(nn.Module):
                 mask=None):
         B, T, _ = x.shape
         cfg = self.cfg
+        if (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
+                and cache is not None and self.head_dim % 8 == 0
+                and isinstance(pos, torch.Tensor) and cos.is_contiguous()):
+            # one kernel: q-rope + k-rope/v straight into the caches
+            # (~10 eager kernels/layer collapse; decode census in
+            # profiles/pmc_ktrace_r02.md)
+            from ..ops import fused
+            qin = self._proj(self.wq, x, 0)
+            kin = self._proj(self.wk, x, 1)
+            vin = self._proj(self.wv, x, 2)
+            k_cache, v_cache = cache
+            q = fused.rope_qkv_cache(qin, kin, vin, cos, sin, pos,
+                                     k_cache, v_cache, cfg.heads,
+                                     cfg.kv_heads, self.head_dim)
+            if mask is not None:
+                k = k_cache
+                v = v_cache
+            else:
+                end = pos_end if pos_end is not None                     else int(pos[-1].item()) + 1
+                k = k_cache[:, :, :end]
+                v = v_cache[:, :, :end]
+            rep = cfg.heads // cfg.kv_heads
+            if rep > 1:
+                k = k.repeat_interleave(rep, dim=1)
+                v = v.repeat_interleave(rep, dim=1)
+            o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
+                                               is_causal=T > 1 and
+                                               mask is None)
+            o = o.transpose(1, 2).reshape(B, T, -1)
+            return self._proj(self.wo, o, 3)
         q = self._proj(self.wq, x, 0).view(B, T, cfg.heads,
                                            self.head_dim).transpose(1, 2)
         k = self._proj(self.wk, x, 1).view(B, T, cfg.kv_heads,

@@ -44,6 +44,33 @@ def lib() -> ctypes.CDLL:
     return _LIB
 
 
+def rope_qkv_cache(qin, kin, vin, cos_t, sin_t, pos, k_cache, v_cache,
+                   heads: int, kv_heads: int, head_dim: int):
+    """One launch per attention layer: q-rope to a fresh [B,Hq,T,D]
+    tensor; k roped and v copied straight into the caches at `pos`
+    (replaces ~10 eager kernels/layer — see the decode census in
+    profiles/pmc_ktrace_r02.md)."""
+
+    import torch
+    B, T, _ = qin.shape
+    S = k_cache.shape[2]
+    lb = lib()
+    if not hasattr(lb, "_rope_ready"):
+        lb.tf_rope_qkv_cache.restype = ctypes.c_int
+        lb.tf_rope_qkv_cache.argtypes = [ctypes.c_void_p] * 9 +             [ctypes.c_int] * 6 + [ctypes.c_void_p]
+        lb._rope_ready = True
+    qout = torch.empty(B, heads, T, head_dim, device=qin.device,
+                       dtype=qin.dtype)
+    rc = lb.tf_rope_qkv_cache(
+        qin.data_ptr(), kin.data_ptr(), vin.data_ptr(),
+        cos_t.data_ptr(), sin_t.data_ptr(), pos.data_ptr(),
+        qout.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
+        B, T, heads, kv_heads, head_dim, S, _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_rope_qkv_cache failed: {rc}")
+    return qout
+
+
 def available() -> bool:
     return os.path.exists(os.path.join(_native_dir(), "libtfops.so"))
 
