@@ -172,9 +172,38 @@ __global__ void rope_qkv_cache_kernel(
   }
 }
 
+__global__ void silu_mul_kernel(const u4* __restrict__ g,
+                                const u4* __restrict__ u,
+                                u4* __restrict__ out, size_t n8) {
+  size_t i = (size_t)blockIdx.x * blockDim.x + threadIdx.x;
+  if (i >= n8) return;
+  V8 a{g[i]}, b{u[i]}, o;
+#pragma unroll
+  for (int k = 0; k < 8; ++k) {
+    float x = __bfloat162float(a.h[k]);
+    float s = x / (1.0f + __expf(-x));
+    o.h[k] = __float2bfloat16(s * __bfloat162float(b.h[k]));
+  }
+  out[i] = o.v;
+}
+
 }  // namespace
 
 extern "C" {
+
+// out = silu(g) * u, bf16, numel % 8 == 0 (one kernel instead of two
+// eager ones on the MLP hot path).
+int tf_silu_mul(const void* g, const void* u, void* out,
+                long long numel, void* stream) {
+  if (numel % 8) return 1;
+  size_t n8 = (size_t)numel / 8;
+  int block = 256;
+  size_t grid = (n8 + block - 1) / block;
+  hipLaunchKernelGGL(silu_mul_kernel, dim3((unsigned)grid), dim3(block), 0,
+                     (hipStream_t)stream, (const u4*)g, (const u4*)u,
+                     (u4*)out, n8);
+  return (int)hipGetLastError();
+}
 
 // One launch per attention layer: rope(q) → qout, rope(k)/copy(v) →
 // caches at pos. All tensors bf16 except cos/sin (fp32) and pos (i64).

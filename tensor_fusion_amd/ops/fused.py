@@ -71,6 +71,25 @@ def rope_qkv_cache(qin, kin, vin, cos_t, sin_t, pos, k_cache, v_cache,
     return qout
 
 
+def silu_mul(g, u):
+    """out = silu(g) * u in one gfx950 kernel (bf16)."""
+
+    import torch
+    lb = lib()
+    if not hasattr(lb, "_silu_ready"):
+        lb.tf_silu_mul.restype = ctypes.c_int
+        lb.tf_silu_mul.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                   ctypes.c_void_p, ctypes.c_longlong,
+                                   ctypes.c_void_p]
+        lb._silu_ready = True
+    out = torch.empty_like(g)
+    rc = lb.tf_silu_mul(g.data_ptr(), u.data_ptr(), out.data_ptr(),
+                        g.numel(), _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_silu_mul failed: {rc}")
+    return out
+
+
 def available() -> bool:
     return os.path.exists(os.path.join(_native_dir(), "libtfops.so"))
 

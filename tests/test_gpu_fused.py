@@ -324,3 +324,17 @@ print(json.dumps(outs))
     flat0 = [t for step in outs["0"] for row in step for t in row]
     agree = sum(a == b for a, b in zip(flat1, flat0)) / len(flat1)
     assert agree >= 0.9, (agree, outs)
+
+
+def test_silu_mul_matches_reference():
+    from tensor_fusion_amd.ops import fused
+    torch.manual_seed(2)
+    g = torch.randn(8, 14336, device="cuda", dtype=torch.bfloat16)
+    u = torch.randn_like(g)
+    got = fused.silu_mul(g, u)
+    torch.cuda.synchronize()
+    want = torch.nn.functional.silu(g.float()) * u.float()
+    # bf16 store quantization: tolerance relative to magnitude
+    err = (got.float() - want).abs()
+    rel = (err / want.abs().clamp(min=1.0)).max().item()
+    assert rel < 1e-2, rel
