@@ -120,16 +120,16 @@ __global__ void add_rmsnorm_kernel(const u4* __restrict__ x,
 // reference implementation (fp32 multiply, bf16 store).
 
 __global__ void rope_qkv_cache_kernel(
-    const u4* __restrict__ qin,   // [B,T,Hq*D] bf16
-    const u4* __restrict__ kin,   // [B,T,Hk*D]
-    const u4* __restrict__ vin,   // [B,T,Hk*D]
+    const u4* __restrict__ qin,   // [B,T,*] bf16, row stride qs elems
+    const u4* __restrict__ kin,   // (may be slices of ONE merged qkv
+    const u4* __restrict__ vin,   //  GEMM output — strides in elems)
     const float* __restrict__ cosT,  // [S, D/2]
     const float* __restrict__ sinT,
     const long long* __restrict__ pos,  // [T]
     u4* __restrict__ qout,        // [B,Hq,T,D]
     u4* __restrict__ kcache,      // [B,Hk,S,D]
     u4* __restrict__ vcache,
-    int B, int T, int Hq, int Hk, int D, int S) {
+    int B, int T, int Hq, int Hk, int D, int S, int qs, int ks, int vs) {
   // grid.x = B*T, grid.y = Hq + 2*Hk; 64 threads each handling 8 elems
   // (4 rope pairs) — D % 8 == 0, D <= 512
   const int bt = blockIdx.x;
@@ -157,18 +157,36 @@ __global__ void rope_qkv_cache_kernel(
 
   const int d0 = i8 * 8;
   if (h < Hq) {  // q head → rope → qout[b,h,t,:]
-    V8 in{qin[((size_t)(b * T + t) * Hq * D + h * D + d0) / 8]};
+    V8 in{qin[((size_t)(b * T + t) * qs + h * D + d0) / 8]};
     V8 o = rope8(in, d0);
     qout[((size_t)((b * Hq + h) * T + t) * D + d0) / 8] = o.v;
   } else if (h < Hq + Hk) {  // k head → rope → kcache[b,hk,p,:]
     int hk = h - Hq;
-    V8 in{kin[((size_t)(b * T + t) * Hk * D + hk * D + d0) / 8]};
+    V8 in{kin[((size_t)(b * T + t) * ks + hk * D + d0) / 8]};
     V8 o = rope8(in, d0);
     kcache[((size_t)((b * Hk + hk) * S + p) * D + d0) / 8] = o.v;
   } else {  // v head → vcache
     int hk = h - Hq - Hk;
     vcache[((size_t)((b * Hk + hk) * S + p) * D + d0) / 8] =
-        vin[((size_t)(b * T + t) * Hk * D + hk * D + d0) / 8];
+        vin[((size_t)(b * T + t) * vs + hk * D + d0) / 8];
+  }
+}
+
+__global__ void silu_mul_gu_kernel(const u4* __restrict__ gu,
+                                   u4* __restrict__ out, int n8,
+                                   int row8) {
+  // grid.x = rows; each row: first n8 groups = gate, next n8 = up
+  const u4* row = gu + (size_t)blockIdx.x * row8;
+  u4* orow = out + (size_t)blockIdx.x * n8;
+  for (int i = threadIdx.x; i < n8; i += blockDim.x) {
+    V8 a{row[i]}, b{row[n8 + i]}, o;
+#pragma unroll
+    for (int k = 0; k < 8; ++k) {
+      float x = __bfloat162float(a.h[k]);
+      float sg = x / (1.0f + __expf(-x));
+      o.h[k] = __float2bfloat16(sg * __bfloat162float(b.h[k]));
+    }
+    orow[i] = o.v;
   }
 }
 
@@ -191,8 +209,22 @@ __global__ void silu_mul_kernel(const u4* __restrict__ g,
 
 extern "C" {
 
+// out[r, i] = silu(gu[r, i]) * gu[r, I+i] — the gate+up halves of ONE
+// merged GEMM output (bf16; I % 8 == 0).
+int tf_silu_mul_gu(const void* gu, void* out, int rows, int inter,
+                   void* stream);
+
 // out = silu(g) * u, bf16, numel % 8 == 0 (one kernel instead of two
 // eager ones on the MLP hot path).
+int tf_silu_mul_gu(const void* gu, void* out, int rows, int inter,
+                   void* stream) {
+  if (inter % 8) return 1;
+  hipLaunchKernelGGL(silu_mul_gu_kernel, dim3(rows), dim3(256), 0,
+                     (hipStream_t)stream, (const u4*)gu, (u4*)out,
+                     inter / 8, inter / 4);
+  return (int)hipGetLastError();
+}
+
 int tf_silu_mul(const void* g, const void* u, void* out,
                 long long numel, void* stream) {
   if (numel % 8) return 1;
@@ -211,15 +243,19 @@ int tf_rope_qkv_cache(const void* qin, const void* kin, const void* vin,
                       const void* cos_t, const void* sin_t,
                       const void* pos, void* qout, void* kcache,
                       void* vcache, int B, int T, int Hq, int Hk, int D,
-                      int S, void* stream) {
-  if (D % 8 || D > 512) return 1;
+                      int S, int q_stride, int k_stride, int v_stride,
+                      void* stream) {
+  if (D % 8 || D > 512 || (q_stride % 8) || (k_stride % 8) ||
+      (v_stride % 8))
+    return 1;
   dim3 grid(B * T, Hq + 2 * Hk);
   dim3 block((D + 7) / 8);
   hipLaunchKernelGGL(rope_qkv_cache_kernel, grid, block, 0,
                      (hipStream_t)stream, (const u4*)qin, (const u4*)kin,
                      (const u4*)vin, (const float*)cos_t,
                      (const float*)sin_t, (const long long*)pos, (u4*)qout,
-                     (u4*)kcache, (u4*)vcache, B, T, Hq, Hk, D, S);
+                     (u4*)kcache, (u4*)vcache, B, T, Hq, Hk, D, S,
+                     q_stride, k_stride, v_stride);
   return (int)hipGetLastError();
 }
 

@@ -100,6 +100,7 @@ class Attention(nn.Module):
         self.wv = nn.Linear(cfg.dim, cfg.kv_heads * self.head_dim, bias=False)
         self.wo = nn.Linear(cfg.heads * self.head_dim, cfg.dim, bias=False)
         self._packed = None  # decode-GEMV weight shuffles (lazy)
+        self._wqkv = None    # merged qkv weight (fused path, lazy)
 
     def _proj(self, lin: nn.Linear, x, idx: int):
         """Decode projections via the packed MFMA GEMV when the token
@@ -134,11 +135,20 @@ class Attention(nn.Module):
                 and isinstance(pos, torch.Tensor) and cos.is_contiguous()):
             # one kernel: q-rope + k-rope/v straight into the caches
             # (~10 eager kernels/layer collapse; decode census in
-            # profiles/pmc_ktrace_r02.md)
+            # profiles/pmc_ktrace_r02.md). qkv = ONE merged GEMM (3
+            # launches -> 1; the rope kernel reads the column slices
+            # with their row stride)
             from ..ops import fused
-            qin = self._proj(self.wq, x, 0)
-            kin = self._proj(self.wk, x, 1)
-            vin = self._proj(self.wv, x, 2)
+            if self._wqkv is None:
+                self._wqkv = torch.cat(
+                    [self.wq.weight, self.wk.weight, self.wv.weight],
+                    dim=0).detach().contiguous()
+            qkv = F.linear(x, self._wqkv)
+            dq = cfg.heads * self.head_dim
+            dk = cfg.kv_heads * self.head_dim
+            qin = qkv[..., :dq]
+            kin = qkv[..., dq:dq + dk]
+            vin = qkv[..., dq + dk:]
             k_cache, v_cache = cache
             q = fused.rope_qkv_cache(qin, kin, vin, cos, sin, pos,
                                      k_cache, v_cache, cfg.heads,
@@ -201,16 +211,25 @@ class MLP(nn.Module):
         self.up = nn.Linear(cfg.dim, cfg.intermediate, bias=False)
         self.down = nn.Linear(cfg.intermediate, cfg.dim, bias=False)
         self._packed = None
+        self._wgu = None  # merged gate+up weight (fused path, lazy)
 
     _proj = Attention._proj  # same packed-GEMV decode routing
 
     def forward(self, x):
+        if (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
+                and self.gate.out_features % 8 == 0):
+            # gate+up = ONE merged GEMM, then one silu*mul kernel over
+            # the two column halves
+            from ..ops import fused
+            if self._wgu is None:
+                self._wgu = torch.cat(
+                    [self.gate.weight, self.up.weight],
+                    dim=0).detach().contiguous()
+            gu = F.linear(x, self._wgu)
+            h = fused.silu_mul_gu(gu, self.gate.out_features)
+            return self._proj(self.down, h, 2)
         g = self._proj(self.gate, x, 0)
         u = self._proj(self.up, x, 1)
-        if (_FUSED_OPS and x.is_cuda and x.dtype == torch.bfloat16
-                and g.numel() % 8 == 0):
-            from ..ops import fused
-            return self._proj(self.down, fused.silu_mul(g, u), 2)
         return self._proj(self.down, F.silu(g) * u, 2)
 
 

@@ -57,18 +57,45 @@ def rope_qkv_cache(qin, kin, vin, cos_t, sin_t, pos, k_cache, v_cache,
     lb = lib()
     if not hasattr(lb, "_rope_ready"):
         lb.tf_rope_qkv_cache.restype = ctypes.c_int
-        lb.tf_rope_qkv_cache.argtypes = [ctypes.c_void_p] * 9 +             [ctypes.c_int] * 6 + [ctypes.c_void_p]
+        lb.tf_rope_qkv_cache.argtypes = ([ctypes.c_void_p] * 9 +
+                                         [ctypes.c_int] * 9 +
+                                         [ctypes.c_void_p])
         lb._rope_ready = True
     qout = torch.empty(B, heads, T, head_dim, device=qin.device,
                        dtype=qin.dtype)
+    # q/k/v may be column slices of ONE merged qkv GEMM output: pass
+    # each slice's base pointer + row stride (in elements)
     rc = lb.tf_rope_qkv_cache(
         qin.data_ptr(), kin.data_ptr(), vin.data_ptr(),
         cos_t.data_ptr(), sin_t.data_ptr(), pos.data_ptr(),
         qout.data_ptr(), k_cache.data_ptr(), v_cache.data_ptr(),
-        B, T, heads, kv_heads, head_dim, S, _stream())
+        B, T, heads, kv_heads, head_dim, S,
+        qin.stride(-2), kin.stride(-2), vin.stride(-2), _stream())
     if rc != 0:
         raise RuntimeError(f"tf_rope_qkv_cache failed: {rc}")
     return qout
+
+
+def silu_mul_gu(gu, inter: int):
+    """out[r,:I] = silu(gu[r,:I]) * gu[r,I:2I] — the two halves of one
+    merged gate+up GEMM output, fused in one kernel."""
+
+    import torch
+    lb = lib()
+    if not hasattr(lb, "_silugu_ready"):
+        lb.tf_silu_mul_gu.restype = ctypes.c_int
+        lb.tf_silu_mul_gu.argtypes = [ctypes.c_void_p, ctypes.c_void_p,
+                                      ctypes.c_int, ctypes.c_int,
+                                      ctypes.c_void_p]
+        lb._silugu_ready = True
+    rows = gu.numel() // (2 * inter)
+    out = torch.empty(*gu.shape[:-1], inter, device=gu.device,
+                      dtype=gu.dtype)
+    rc = lb.tf_silu_mul_gu(gu.data_ptr(), out.data_ptr(), rows, inter,
+                           _stream())
+    if rc != 0:
+        raise RuntimeError(f"tf_silu_mul_gu failed: {rc}")
+    return out
 
 
 def silu_mul(g, u):
