@@ -204,6 +204,8 @@ def packed_profitable(N: int, K: int) -> bool:
         return False  # deep-K underfilled: hipBLASLt wins 3x
     if N >= 32768:
         return True
+    if N >= 16384 and os.environ.get("TF_PACKED_MID", "0") == "1":
+        return True  # chip-filling-ish (gate+up N=28672): A/B gated
     # small-N: wave-split-K wins the STANDALONE microbench (15.6 vs
     # 18.5us at N=4096) but measured consistently slower IN the captured
     # decode graph (A/B x2: 762.5 vs 776.6 tok/s) — hipBLASLt's decode

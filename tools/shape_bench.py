@@ -3,7 +3,7 @@ sys.path.insert(0, ".")
 import torch
 from tensor_fusion_amd.ops import fused
 torch.manual_seed(0)
-for (N, K) in [(4096, 4096), (1024, 4096), (14336, 4096), (4096, 14336), (128256, 4096)]:
+for (N, K) in [(6144, 4096), (28672, 4096), (14336, 4096), (128256, 4096)]:
     x = torch.randn(8, K, device="cuda", dtype=torch.bfloat16)
     w = torch.randn(N, K, device="cuda", dtype=torch.bfloat16)
     wp = fused.pack_skinny_weight(w)
