@@ -271,3 +271,68 @@ def test_operator_cli_k8s_mode(tmp_path):
         except subprocess.TimeoutExpired:
             proc.kill()
         us.should_exit = True
+
+
+def test_node_plane_kubernetes_backend_over_the_wire(tmp_path):
+    """Node onboarding over the wire: a hypervisor built
+    with --backend kubernetes semantics (K8sStore) discovers its (mock)
+    devices and publishes GPU + GPUNode CRs THROUGH the apiserver; a
+    worker pod bound to the node appears via the watch and gets its shm
+    page + allocation; deleting the pod tears the worker down."""
+
+    import tensor_fusion_amd.constants as C
+    from tensor_fusion_amd.hypervisor.main import build_hypervisor
+
+    srv, base, us = serve_in_thread()
+    kubectl = K8sClient(base)
+    for crd in all_crds().values():
+        kubectl.create(crd)
+    store = K8sStore(K8sClient(base), namespace="default").start()
+    try:
+        devices, workers, erl, backend = build_hypervisor(
+            node="knode-0", mock_devices=2,
+            shm_root=str(tmp_path / "shm"), store=store)
+        devices.start()
+        backend.start()
+
+        # GPU CRs (with capacity + topology) and the GPUNode exist on
+        # the WIRE, not just locally
+        gpus = _wait(lambda: kubectl.list_items("GPU") or None)
+        assert gpus and len(gpus) == 2
+        st = gpus[0]["status"]
+        assert st["capacity"]["vram"] == C.MI355X_VRAM_BYTES
+        assert st["node"] == "knode-0"
+        gn = kubectl.get("GPUNode", "knode-0")
+        assert gn["status"]["gpuCount"] == 2
+
+        # a worker pod bound to this node (as the scheduler would leave
+        # it) flows in via the informer → worker + shm appear
+        uuid0 = gpus[0]["status"]["uuid"]
+        pod = {
+            "apiVersion": "v1", "kind": "Pod",
+            "metadata": {
+                "name": "w-kube", "namespace": "default",
+                "labels": {C.LabelComponent: C.ComponentWorker},
+                "annotations": {
+                    C.AnnoGpuIds: uuid0,
+                    C.AnnoContainerGpus: gpus[0]["metadata"]["name"],
+                    C.AnnoVramRequest: str(8 << 30),
+                    C.AnnoVramLimit: str(8 << 30),
+                    C.AnnoTflopsLimit: "600",
+                },
+            },
+            "spec": {"containers": [{"name": "main"}],
+                     "nodeName": "knode-0"},
+        }
+        kubectl.create(pod)
+        w = _wait(lambda: workers.get("default/w-kube"))
+        assert w is not None
+        assert w.allocation.spec.gpu_uuids == [uuid0]
+        assert (tmp_path / "shm" / "default" / "w-kube" / "shm").exists()
+
+        kubectl.delete("Pod", "w-kube", "default")
+        gone = _wait(lambda: workers.get("default/w-kube") is None)
+        assert gone
+    finally:
+        store.stop()
+        us.should_exit = True
