@@ -14,7 +14,7 @@ import json
 import urllib.parse
 import uuid
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, Optional
 
 import requests
 

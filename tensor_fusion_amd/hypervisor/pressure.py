@@ -21,7 +21,6 @@ Runs alongside the ERL loop at the same 500 ms-class cadence.
 from __future__ import annotations
 
 import threading
-import time
 from dataclasses import dataclass, field
 from typing import Callable, Dict, List, Optional, Tuple
 

@@ -27,7 +27,7 @@ import grpc
 from .. import constants as C
 from .dp_proto import (API_VERSION, KUBELET_SOCKET, M_ALLOCATE,
                        M_LISTWATCH, M_OPTIONS, M_PREFERRED, M_PRESTART,
-                       MSG, PKG, REGISTRATION_SERVICE)
+                       MSG, REGISTRATION_SERVICE)
 
 # Allocate() answer for one pod index: env + device nodes + annotations.
 AllocationResolver = Callable[[int], Optional[dict]]

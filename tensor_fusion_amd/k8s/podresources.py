@@ -12,7 +12,7 @@ its own worker set and the checkpoint detector's foreign-device view.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 import grpc
 from google.protobuf import descriptor_pb2, descriptor_pool, message_factory
