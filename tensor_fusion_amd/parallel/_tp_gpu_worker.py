@@ -42,6 +42,32 @@ if mode == "allreduce":
     if rank == 0:
         print(json.dumps({"ok": ok and ok2, "world": world,
                           "mode": "allreduce"}), flush=True)
+elif mode == "ep":
+    from tensor_fusion_amd.parallel.ep import ExpertParallelMLP
+    DIM, INTER, NEXP, N = 64, 128, 4, 53
+    torch.manual_seed(1)
+    ref = ExpertParallelMLP(DIM, INTER, NEXP, ep_size=1).float()
+    torch.manual_seed(2)
+    x = torch.randn(N, DIM)
+    with torch.no_grad():
+        want = ref(x)
+    ep = ExpertParallelMLP(DIM, INTER, NEXP, ep_size=world).float()
+    per = NEXP // world
+    with torch.no_grad():
+        ep.router.weight.copy_(ref.router.weight)
+        for i in range(per):
+            src = ref.experts[rank * per + i]
+            ep.experts[i].gate.weight.copy_(src.gate.weight)
+            ep.experts[i].up.weight.copy_(src.up.weight)
+            ep.experts[i].down.weight.copy_(src.down.weight)
+    ep = ep.to("cuda")
+    xg = x.to("cuda")
+    with torch.no_grad():
+        got = ep(xg).cpu()
+    err = (got - want).abs().max().item()
+    if rank == 0:
+        print(json.dumps({"ok": err < 1e-4, "rel_err": err,
+                          "world": world, "mode": "ep"}), flush=True)
 elif mode == "decode":
     from tensor_fusion_amd.models.llama import CONFIGS, Llama, decode_bench
     from tensor_fusion_amd.parallel.tp import TPLlama

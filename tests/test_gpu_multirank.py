@@ -118,3 +118,14 @@ def test_tp2_decode_under_vgpu_limiter():
     }, timeout=900))
     assert r["ok"] is True, r
     assert r["tok_s"] > 0
+
+
+def test_ep2_all_to_all_device_tensors():
+    """Expert-parallel MoE with device tensors across 2 ranks on one
+    GPU (gloo transport): token routing + all_to_all exchanges + return
+    scatter reproduce the dense reference. The RCCL all-to-all over
+    real xGMI fan-out is the driver's multi-GPU run."""
+
+    r = _rank0_json(_spawn_ranks(2, "ep",
+                                 extra_env={"TF_TP_BACKEND": "gloo"}))
+    assert r["ok"] is True, r
