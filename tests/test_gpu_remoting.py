@@ -492,3 +492,47 @@ def test_auto_freeze_idle_worker_and_resume_on_dial(tmp_path):
         if cli.poll() is None:
             cli.kill()
         mgr.stop("default/afw")
+
+
+def test_tcp_wire_robustness_garbage_inputs(tmp_path):
+    """A worker's TCP port must survive hostile/buggy peers: a wrong
+    handshake magic is rejected, a valid handshake followed by garbage
+    frames ends that session cleanly, and a LEGITIMATE client afterwards
+    still gets a full working session (the accept loop recovers)."""
+
+    import socket as _s
+    import struct
+
+    from tensor_fusion_amd.client.runtime import client_env, start_worker
+    port = 48777
+    w = start_worker("", device_index=0, tcp_port=port)
+    try:
+        # 1) wrong magic → closed without echo
+        c = _s.create_connection(("127.0.0.1", port), timeout=10)
+        c.sendall(b"NOPE")
+        c.settimeout(3)
+        got = b""
+        try:
+            got = c.recv(4)
+        except _s.timeout:
+            pass
+        assert got == b""  # no magic echo, connection dropped
+        c.close()
+
+        # 2) good magic, then a hostile frame header (absurd rec_len)
+        c = _s.create_connection(("127.0.0.1", port), timeout=10)
+        c.sendall(struct.pack("<I", 0x54465443))
+        assert c.recv(4) == struct.pack("<I", 0x54465443)
+        c.sendall(struct.pack("<IIQII", 0, 0xFFFFFFFF, 0, 0, 0))
+        c.sendall(os.urandom(4096))
+        c.close()
+
+        # 3) a real client afterwards works end to end
+        time.sleep(0.5)
+        env = client_env("", tcp=f"127.0.0.1:{port}")
+        out = subprocess.run([os.path.join(NATIVE, "tf_remote_testapp")],
+                             capture_output=True, text=True, timeout=180,
+                             env=env)
+        assert "TESTAPP_OK" in out.stdout, out.stdout + out.stderr
+    finally:
+        w.stop()

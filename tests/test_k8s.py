@@ -231,3 +231,40 @@ class TestBridge:
             assert wire["spec"]["nodeName"] == "some-node"
         finally:
             store.stop()
+
+
+class TestBridgeConflicts:
+    def test_patch_retries_through_external_conflict(self, api):
+        """An external writer bumps the resourceVersion between the
+        bridge's read and write → Conflict → the RMW refetches and
+        retries (the controller-runtime idiom)."""
+
+        cli = K8sClient(api[1])
+        store = K8sStore(cli, kinds=["GPU"], namespace="conf").start()
+        try:
+            store.create(mk_gpu("cf-g0", "cf-n0"))
+            calls = {"n": 0}
+            real_update = store.update
+
+            def racing_update(obj, check_rv=True):
+                calls["n"] += 1
+                if calls["n"] == 1:
+                    # external actor wins the race before our write
+                    cli.patch("GPU", "cf-g0",
+                              {"metadata": {"labels": {"ext": "1"}}})
+                    # local cache still holds the old rv → force the
+                    # conflict the live system would hit
+                    from tensor_fusion_amd.api.store import Conflict
+                    raise Conflict("simulated external writer")
+                return real_update(obj, check_rv)
+
+            store.update = racing_update
+            store.patch("GPU", "cf-g0", "",
+                        lambda o: setattr(o.status, "phase", "Migrating"))
+            store.update = real_update
+            wire = cli.get("GPU", "cf-g0")
+            assert wire["status"]["phase"] == "Migrating"
+            assert wire["metadata"]["labels"].get("ext") == "1"
+            assert calls["n"] == 2  # one conflict, one success
+        finally:
+            store.stop()
