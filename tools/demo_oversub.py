@@ -144,6 +144,9 @@ def main():
     ap.add_argument("--duration", type=float, default=40.0)
     ap.add_argument("--reserve-gb", type=float, default=8.0)
     ap.add_argument("--shm-root", default="/tmp/tf-demo-shm")
+    ap.add_argument("--qos-mix", default="",
+                    help="comma list per tenant, e.g. critical,high,"
+                         "medium,low (default: all medium)")
     ap.add_argument("--scaled", action="store_true",
                     help="CI shape: 3 tenants x 3 GB cap, 4 GB alloc")
     args = ap.parse_args()
@@ -170,13 +173,16 @@ def main():
                       "solo_rc": solo["rc"]}), flush=True)
 
     # ---- N concurrent tenants under the pressure controller
+    qos_mix = ([q.strip() for q in args.qos_mix.split(",")]
+               if args.qos_mix else [C.QosMedium] * args.tenants)
     pages = []
     for i in range(args.tenants):
         p = S.WorkerShm.create(os.path.join(args.shm_root, f"t{i}", "shm"))
         p.set_device(0, f"tenant-{i}", up_limit_percent=100,
                      mem_limit_bytes=cap, total_cus=256,
                      refill_rate=0.0, capacity=0.0)
-        pc.attach(p, qos=C.QosMedium, provisioned_bytes=cap)
+        pc.attach(p, qos=qos_mix[i % len(qos_mix)],
+                  provisioned_bytes=cap)
         pages.append(p)
     pc.start()
 
@@ -209,6 +215,7 @@ def main():
                    "alloc_gb": args.alloc_gb, "hot_gb": args.hot_gb,
                    "provisioned_total_gb": args.tenants * args.cap_gb},
         "solo_it_s": solo_its,
+        "qos_mix": qos_mix,
         "tenant_it_s": tenant_its,
         "mean_slowdown_x": round(
             solo_its / (sum(tenant_its) / len(tenant_its)), 2)
