@@ -536,3 +536,77 @@ def test_tcp_wire_robustness_garbage_inputs(tmp_path):
         assert "TESTAPP_OK" in out.stdout, out.stdout + out.stderr
     finally:
         w.stop()
+
+
+HOT_MIGRATE_CLIENT = r"""
+import ctypes, os, sys, time
+hip = ctypes.CDLL(None)
+
+def ck(rc, what):
+    if rc != 0:
+        print(f"FAIL {what} rc={rc}", flush=True)
+        sys.exit(1)
+
+N = 1 << 18
+a = ctypes.c_void_p(); b = ctypes.c_void_p()
+ck(hip.hipMalloc(ctypes.byref(a), N * 4), "malloc-a")
+ck(hip.hipMalloc(ctypes.byref(b), N * 4), "malloc-b")
+host = (ctypes.c_float * N)(*([1.0] * N))
+ck(hip.hipMemcpy(a, host, N * 4, 1), "h2d")
+print("LOOPING", flush=True)
+stop_marker = sys.argv[1]
+iters = 0
+# continuous device work: copy a->b and read one element back, no pauses
+while not os.path.exists(stop_marker):
+    ck(hip.hipMemcpy(b, a, N * 4, 3), f"d2d-{iters}")
+    probe = (ctypes.c_float * 1)()
+    ck(hip.hipMemcpy(probe, b, 4, 2), f"d2h-{iters}")
+    if abs(probe[0] - 1.0) > 1e-9:
+        print(f"FAIL corrupt probe {probe[0]} at {iters}", flush=True)
+        sys.exit(1)
+    iters += 1
+back = (ctypes.c_float * N)()
+ck(hip.hipMemcpy(back, a, N * 4, 2), "final-d2h")
+ok = all(abs(back[i] - 1.0) < 1e-9 for i in range(0, N, 16384))
+print(f"DONE iters={iters} ok={ok}", flush=True)
+sys.exit(0 if (ok and iters > 0) else 1)
+"""
+
+
+def test_live_migration_under_continuous_load(tmp_path):
+    """Migrate the worker while the client is MID-LOOP issuing device
+    work with no cooperation — ops stall during the snapshot/restore
+    window, then complete; no op ever fails and the data survives
+    (config-5's live-migrate-one-vGPU story under load)."""
+
+    from tensor_fusion_amd.client.runtime import (client_env,
+                                                  migrate_worker,
+                                                  start_worker)
+    sock = str(tmp_path / "vgpu.sock")
+    snap = str(tmp_path / "snap.bin")
+    stop = str(tmp_path / "stop")
+    w = start_worker(sock, device_index=0, snapshot_path=snap)
+    env = client_env(sock)
+    cli = subprocess.Popen([sys.executable, "-c", HOT_MIGRATE_CLIENT,
+                            stop],
+                           env=env, stdout=subprocess.PIPE,
+                           stderr=subprocess.PIPE, text=True, cwd=REPO)
+    w2 = None
+    try:
+        assert "LOOPING" in cli.stdout.readline()
+        time.sleep(1.0)  # client is deep in its loop
+        w2 = migrate_worker(w, snap, new_device_index=0)
+        time.sleep(1.0)  # client keeps hammering the restored worker
+        with open(stop, "w") as f:
+            f.write("x")
+        out, err = cli.communicate(timeout=180)
+        assert cli.returncode == 0, out + err
+        assert "DONE" in out and "ok=True" in out, out + err
+        iters = int(out.split("iters=")[1].split()[0])
+        assert iters > 10, f"client barely ran: {iters}"
+    finally:
+        if cli.poll() is None:
+            cli.kill()
+        if w2 is not None:
+            w2.stop()
+        w.stop()
