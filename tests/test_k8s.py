@@ -268,3 +268,72 @@ class TestBridgeConflicts:
             assert calls["n"] == 2  # one conflict, one success
         finally:
             store.stop()
+
+
+class TestSerdeFuzz:
+    def test_random_objects_round_trip(self):
+        """Randomized dataclass instances survive to_k8s → from_k8s for
+        every CRD kind (field-name/camelCase/nesting bugs surface
+        here)."""
+
+        import random
+
+        import dataclasses
+        import typing
+
+        rng = random.Random(42)
+
+        def fill(obj, depth=0):
+            if depth > 4:
+                return
+            hints = typing.get_type_hints(type(obj))
+            for f in dataclasses.fields(obj):
+                if f.name in ("kind", "meta"):
+                    continue
+                t = hints.get(f.name)
+                cur = getattr(obj, f.name)
+                if t is int:
+                    setattr(obj, f.name, rng.randint(0, 1 << 40))
+                elif t is float:
+                    setattr(obj, f.name, round(rng.uniform(0, 5000), 3))
+                elif t is str:
+                    setattr(obj, f.name, f"s{rng.randint(0, 999)}")
+                elif t is bool:
+                    setattr(obj, f.name, rng.random() < 0.5)
+                elif dataclasses.is_dataclass(cur):
+                    fill(cur, depth + 1)
+                elif isinstance(cur, dict) and rng.random() < 0.7:
+                    vt = typing.get_args(t)[1] if typing.get_args(t) \
+                        else str
+                    if vt is int:
+                        cur[f"k{rng.randint(0,9)}"] = rng.randint(0, 99)
+                    elif vt is str:
+                        cur[f"k{rng.randint(0,9)}"] = "v"
+                elif isinstance(cur, list) and typing.get_args(t):
+                    (et,) = typing.get_args(t)
+                    if et is str and rng.random() < 0.7:
+                        cur.append(f"item{rng.randint(0,99)}")
+                    elif dataclasses.is_dataclass(et) and \
+                            rng.random() < 0.5 and depth < 3:
+                        inst = et()
+                        fill(inst, depth + 1)
+                        cur.append(inst)
+
+        import dataclasses as dc
+        for kind in serde.CRD_KINDS:
+            cls = getattr(T, kind)
+            for trial in range(8):
+                obj = cls()
+                obj.meta.name = f"{kind.lower()}-{trial}"
+                if kind not in serde.CLUSTER_SCOPED:
+                    obj.meta.namespace = "ns"
+                fill(obj)
+                wire = serde.to_k8s(obj)
+                back = serde.from_k8s(wire)
+                a = dc.asdict(obj)
+                b = dc.asdict(back)
+                # server-owned meta noise tolerated
+                for d in (a, b):
+                    for k in ("uid", "creation_ts", "resource_version"):
+                        d["meta"].pop(k, None)
+                assert a == b, (kind, trial)

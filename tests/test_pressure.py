@@ -109,3 +109,46 @@ class TestPressure:
         assert pc.tick().budgets
         pc.detach(p.path)
         assert not pc.tick().budgets
+
+
+class TestBudgetInvariants:
+    def test_waterfill_invariants_randomized(self, tmp_path):
+        """Random tenant mixes: budgets never exceed caps, never exceed
+        usable HBM in total, respect QoS ordering at equal caps, and
+        spare capacity from capped tenants is redistributed."""
+
+        import random
+        rng = random.Random(9)
+        for trial in range(25):
+            n = rng.randint(1, 6)
+            free = rng.randint(1, 280) * GB
+            pc = PressureController(lambda: (free, TOTAL),
+                                    reserve_bytes=4 * GB)
+            caps, qoses, pages = [], [], []
+            for i in range(n):
+                cap = rng.choice([24, 48, 96, 192]) * GB
+                qos = rng.choice([C.QosLow, C.QosMedium, C.QosHigh,
+                                  C.QosCritical])
+                p = mk_page(tmp_path, f"t{trial}-{i}")
+                pc.attach(p, qos=qos, provisioned_bytes=cap)
+                caps.append(cap)
+                qoses.append(qos)
+                pages.append(p)
+            d = pc.tick()
+            budgets = [d.budgets[p.path] for p in pages]
+            usable = TOTAL - 4 * GB
+            assert sum(budgets) <= usable + n  # rounding slack
+            for b, cap in zip(budgets, caps):
+                assert 0 <= b <= cap
+            # QoS ordering among equal-cap, uncapped tenants
+            from tensor_fusion_amd.hypervisor.pressure import QOS_WEIGHT
+            for i in range(n):
+                for j in range(n):
+                    if caps[i] == caps[j] and budgets[i] < caps[i] \
+                            and budgets[j] < caps[j]:
+                        if QOS_WEIGHT[qoses[i]] > QOS_WEIGHT[qoses[j]]:
+                            assert budgets[i] >= budgets[j] - 1
+            # fully-provisionable fleets get their full caps
+            if sum(caps) <= usable:
+                for b, cap in zip(budgets, caps):
+                    assert b == cap
