@@ -190,3 +190,43 @@ def test_partitioned_pod_gets_partition_id():
     assert res[0].status == Code.Success
     pod = store.get("Pod", "part", "default")
     assert pod.meta.annotations[C.AnnoPartitionId].endswith("xcd1-0")
+
+
+def test_feasible_scan_rotates_across_cycles():
+    """The numFeasibleNodes truncation must not always consider the
+    same fleet prefix (advisor finding: fixed dict-order scans hotspot
+    the first ~100 nodes; kube-scheduler rotates nextStartNodeIndex)."""
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.api.types import Node, Pod
+    from tensor_fusion_amd.scheduler.framework import (Code, Plugin,
+                                                       Scheduler, Status)
+
+    store = Store()
+    for i in range(400):
+        n = Node()
+        n.meta.name = f"n{i:03d}"
+        store.create(n)
+
+    seen_first: list = []
+
+    class Recorder(Plugin):
+        name = "rec"
+
+        def filter(self, state, pod, node):
+            if not seen_first or seen_first[-1][0] != pod.meta.name:
+                seen_first.append((pod.meta.name, node))
+            return Status.ok_()
+
+    sched = Scheduler(store, [Recorder()],
+                      bind_fn=lambda pod, node: None)
+    for i in range(6):
+        p = Pod()
+        p.meta.name = f"p{i}"
+        p.meta.namespace = "default"
+        p.scheduler_name = sched.scheduler_name
+        store.create(p)
+        sched.schedule_pod(p)
+    firsts = {node for (_, node) in seen_first}
+    # with rotation the first-considered node differs across cycles
+    assert len(firsts) >= 4, seen_first
