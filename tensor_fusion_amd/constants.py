@@ -9,6 +9,7 @@ brand-new MI355X-native implementation.
 
 # ---------------------------------------------------------------- domain
 Domain = "tensor-fusion.ai"
+Version = "0.2.0"  # round-2 build
 
 # ---------------------------------------------------------------- labels
 LabelComponent = f"{Domain}/component"  # client | worker | hypervisor | operator

@@ -76,7 +76,45 @@ def make_dp_resolver(store, node: str, workers):
     return resolver
 
 
+def mount_shm(path: str, size_mb: int = 64, dry_run: bool = False) -> str:
+    """`mount-shm` init-container subcommand: mount a tmpfs at the
+    limiter-shm root so worker pods and the hypervisor share pages via
+    a hostPath (reference shm_init/mount_shm.go:17-92). Idempotent: if
+    `path` is already a tmpfs mountpoint it is left alone. Returns the
+    action taken ("mounted" | "already-mounted" | "dir-only")."""
+
+    import os
+    import subprocess
+    os.makedirs(path, exist_ok=True)
+    try:
+        with open("/proc/mounts") as f:
+            for line in f:
+                parts = line.split()
+                if len(parts) >= 3 and parts[1] == os.path.realpath(path) \
+                        and parts[2] == "tmpfs":
+                    return "already-mounted"
+    except OSError:
+        pass
+    cmd = ["mount", "-t", "tmpfs", "-o",
+           f"size={size_mb}m,mode=0755", "tensor-fusion-shm", path]
+    if dry_run or os.geteuid() != 0:
+        return "dir-only"  # tests / rootless: plain dir still works
+    try:
+        subprocess.run(cmd, check=True, capture_output=True)
+        return "mounted"
+    except (subprocess.CalledProcessError, FileNotFoundError):
+        return "dir-only"
+
+
 def main():
+    import sys
+    if len(sys.argv) > 1 and sys.argv[1] == "mount-shm":
+        mp = argparse.ArgumentParser(prog="hypervisor mount-shm")
+        mp.add_argument("--path", default=C.ShmRoot)
+        mp.add_argument("--size-mb", type=int, default=64)
+        a = mp.parse_args(sys.argv[2:])
+        print(mount_shm(a.path, a.size_mb))
+        return
     ap = argparse.ArgumentParser()
     ap.add_argument("--node", default="node-0")
     ap.add_argument("--backend", default="single",

@@ -60,6 +60,29 @@ class HypervisorMetrics:
                 throttled_ratio=min(1.0, block_ns / max(1, now_ns))))
         return self.recorder.flush()
 
+    def telemetry_ping(self) -> Optional[dict]:
+        """Anonymous usage ping (reference metrics.go:40-46 PostHog).
+        OPT-IN here: only fires when TF_TELEMETRY_URL is set and
+        TF_TELEMETRY_DISABLED is not — no endpoint is baked in. Payload
+        is aggregate-only (node/device counts, no names)."""
+
+        import os
+        url = os.environ.get("TF_TELEMETRY_URL", "")
+        if not url or os.environ.get("TF_TELEMETRY_DISABLED"):
+            return None
+        payload = {
+            "event": "hypervisor_heartbeat",
+            "gpu_count": len(self.devices.devices()),
+            "worker_count": len(self.workers.worker_metrics()),
+            "version": C.Version,
+        }
+        try:
+            import requests
+            requests.post(url, json=payload, timeout=3)
+        except Exception:
+            pass  # telemetry must never disturb the node plane
+        return payload
+
     def start(self, interval_s: float = 60.0):
         self._stop.clear()
 
@@ -67,6 +90,7 @@ class HypervisorMetrics:
             while not self._stop.wait(interval_s):
                 try:
                     self.collect_once()
+                    self.telemetry_ping()
                 except Exception:
                     pass
         self._thread = threading.Thread(target=loop, daemon=True,

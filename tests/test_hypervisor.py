@@ -266,3 +266,37 @@ def test_hypervisor_metrics_loop(tmp_path):
     pts = tsdb.query("tf_node_metrics", "gpu_count", tags={"node": "node-x"})
     assert pts and pts[-1][1] >= 1  # mock device count is process-sticky
     assert (tmp_path / "m" / "metrics.log").exists()
+
+
+def test_mount_shm_subcommand(tmp_path):
+    """`hypervisor mount-shm` (reference shm_init/mount_shm.go:17-92):
+    idempotent tmpfs mount of the limiter-shm root; rootless/test mode
+    degrades to a plain directory."""
+
+    from tensor_fusion_amd.hypervisor.main import mount_shm
+    p = str(tmp_path / "run" / "tensor-fusion")
+    action = mount_shm(p, dry_run=True)
+    assert action in ("dir-only", "already-mounted")
+    import os
+    assert os.path.isdir(p)
+    # idempotent
+    assert mount_shm(p, dry_run=True) in ("dir-only", "already-mounted")
+
+
+def test_telemetry_ping_opt_in(monkeypatch, tmp_path):
+    """Telemetry fires only when TF_TELEMETRY_URL is set and never
+    raises on an unreachable sink (reference metrics.go:40-46)."""
+
+    from tensor_fusion_amd.hypervisor.main import build_hypervisor
+    from tensor_fusion_amd.hypervisor.metrics import HypervisorMetrics
+    devices, workers, erl, backend = build_hypervisor(
+        node="n0", mock_devices=1, shm_root=str(tmp_path))
+    hm = HypervisorMetrics("n0", devices, workers,
+                           out_dir=str(tmp_path / "m"))
+    monkeypatch.delenv("TF_TELEMETRY_URL", raising=False)
+    assert hm.telemetry_ping() is None  # off by default
+    monkeypatch.setenv("TF_TELEMETRY_URL", "http://127.0.0.1:1/none")
+    out = hm.telemetry_ping()  # sink unreachable: payload still built
+    assert out and out["gpu_count"] == 1 and out["version"]
+    monkeypatch.setenv("TF_TELEMETRY_DISABLED", "1")
+    assert hm.telemetry_ping() is None
