@@ -170,6 +170,9 @@ class DevicePluginManager:
         self.max_indexes = max_indexes
         self._servers: List[grpc.Server] = []
         self._plugins: Dict[int, _PluginServicer] = {}
+        self._stop = threading.Event()
+        self._watcher: Optional[threading.Thread] = None
+        self.registrations = 0  # total register RPCs (visible to tests)
 
     # ------------------------------------------------------------ serve
 
@@ -193,7 +196,43 @@ class DevicePluginManager:
             self._plugins[idx] = plugin
             if register:
                 self._register(idx, endpoint)
+        if register:
+            self._watcher = threading.Thread(
+                target=self._watch_kubelet, daemon=True,
+                name="dp-kubelet-watch")
+            self._watcher.start()
         return self
+
+    def _kubelet_ino(self):
+        """Identity of the current kubelet socket: (inode, ctime_ns).
+        Inode alone is unreliable — tmpfs reuses inode numbers
+        immediately, so a fast restart could look unchanged."""
+
+        try:
+            st = os.stat(self.kubelet_socket)
+            return (st.st_ino, st.st_ctime_ns)
+        except OSError:
+            return None
+
+    def _watch_kubelet(self, poll_s: float = 2.0):
+        """kubelet restarts wipe its plugin registry and recreate its
+        registration socket — detect the new socket inode and
+        re-register every index plugin (reference deviceplugin.go
+        restart loop)."""
+
+        last = self._kubelet_ino()
+        while not self._stop.wait(getattr(self, "_watch_kubelet_poll",
+                                          poll_s)):
+            ident = self._kubelet_ino()
+            if ident is not None and ident != last:
+                for idx in self._plugins:
+                    try:
+                        self._register(idx, f"tf-index-{idx}.sock")
+                    except grpc.RpcError:
+                        ident = None  # kubelet not ready: retry next poll
+                        break
+            if ident is not None:
+                last = ident
 
     def _register(self, idx: int, endpoint: str, timeout: float = 5.0):
         ch = grpc.insecure_channel(f"unix://{self.kubelet_socket}")
@@ -206,6 +245,7 @@ class DevicePluginManager:
                 version=API_VERSION, endpoint=endpoint,
                 resource_name=f"{C.IndexResourcePrefix}{idx}")
             stub(req, timeout=timeout)
+            self.registrations += 1
         finally:
             ch.close()
 
@@ -214,6 +254,7 @@ class DevicePluginManager:
             p.refresh()
 
     def stop(self, grace: float = 0.5):
+        self._stop.set()
         for p in self._plugins.values():
             p.stop()
         for s in self._servers:

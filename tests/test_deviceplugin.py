@@ -282,3 +282,48 @@ class TestPodResources:
                 "default/legacy": ["uuid-aaa"]}
         finally:
             server.stop(0.2)
+
+
+class TestKubeletRestart:
+    def test_reregister_after_kubelet_socket_recreated(self, tmp_path):
+        """kubelet restart wipes its plugin registry and recreates its
+        registration socket — the manager must notice the new socket
+        inode and re-register every plugin (reference deviceplugin.go
+        restart loop)."""
+
+        sock = str(tmp_path / "kubelet.sock")
+
+        def serve():
+            servicer = FakeKubeletRegistration()
+            server = grpc.server(ThreadPoolExecutor(max_workers=2))
+            server.add_generic_rpc_handlers((servicer,))
+            server.add_insecure_port(f"unix://{sock}")
+            server.start()
+            return servicer, server
+
+        s1, srv1 = serve()
+        mgr = DevicePluginManager(_resolver, socket_dir=str(tmp_path),
+                                  kubelet_socket=sock, max_indexes=2)
+        # fast poll for the test
+        mgr._watch_kubelet_poll = 0.1
+        mgr.start(register=True)
+        try:
+            assert len(s1.registered) == 2
+            first = mgr.registrations
+            # hand-drive the watcher logic across a "restart":
+            old_ino = mgr._kubelet_ino()
+            srv1.stop(0)
+            import contextlib
+            import os
+            with contextlib.suppress(OSError):
+                os.unlink(sock)  # grpc may remove its socket on stop
+            s2, srv2 = serve()
+            assert mgr._kubelet_ino() != old_ino
+            deadline = time.time() + 10
+            while time.time() < deadline and len(s2.registered) < 2:
+                time.sleep(0.1)
+            assert len(s2.registered) == 2, s2.registered
+            assert mgr.registrations == first + 2
+            srv2.stop(0.2)
+        finally:
+            mgr.stop()
