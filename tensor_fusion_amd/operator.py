@@ -206,6 +206,45 @@ def build_operator(persist_dir: Optional[str] = None,
         defrag=defrag, config=cfg)
 
 
+def run_leader_election(cli, lease_name: str, namespace: str,
+                        duration_s: int = 15, identity: str = "",
+                        on_lost=None, _max_wait: float = 0.0) -> str:
+    """Block until this replica holds the operator Lease, then keep
+    renewing it from a daemon thread (reference cmd/main.go manager
+    leader election). Losing the lease calls `on_lost` (default:
+    hard-exit so kubernetes restarts us into a follower)."""
+
+    import os
+    import socket
+    import time as _t
+    identity = identity or f"{socket.gethostname()}-{os.getpid()}"
+    waited = 0.0
+    while not cli.acquire_lease(lease_name, namespace, identity,
+                                duration_s=duration_s):
+        _t.sleep(duration_s / 5)
+        waited += duration_s / 5
+        if _max_wait and waited >= _max_wait:
+            return ""  # tests: give up instead of blocking forever
+
+    def renew():
+        while True:
+            _t.sleep(duration_s / 3)
+            try:
+                ok = cli.acquire_lease(lease_name, namespace, identity,
+                                       duration_s=duration_s)
+            except Exception:
+                ok = True  # transient apiserver error: keep trying
+            if not ok:
+                if on_lost:
+                    on_lost()
+                    return
+                os._exit(1)
+
+    threading.Thread(target=renew, daemon=True,
+                     name="leader-lease-renew").start()
+    return identity
+
+
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--persist-dir", default="")
@@ -226,6 +265,12 @@ def main():
     ap.add_argument("--webhook-certs", default="",
                     help="dir with tls.crt/tls.key for the webhook server "
                          "(generated with tensor_fusion_amd.k8s.certs)")
+    ap.add_argument("--leader-elect", action="store_true",
+                    help="Lease-based leader election: block until this "
+                         "replica holds the lease, exit if it loses it "
+                         "(k8s restarts us; followers keep serving HTTP "
+                         "reads and proxy writes to the leader)")
+    ap.add_argument("--lease-name", default="tensor-fusion-operator")
     args = ap.parse_args()
 
     store = None
@@ -244,6 +289,8 @@ def main():
                     if not e.conflict:
                         raise
         store = K8sStore(cli, namespace=args.namespace).start()
+        if args.leader_elect:
+            run_leader_election(cli, args.lease_name, args.namespace)
 
     op = build_operator(persist_dir=args.persist_dir or None,
                         metrics_dir=args.metrics_dir,

@@ -336,3 +336,41 @@ def test_node_plane_kubernetes_backend_over_the_wire(tmp_path):
     finally:
         store.stop()
         us.should_exit = True
+
+
+def test_leader_election_over_wire():
+    """Two operator identities contend for the Lease over the real
+    wire; the loser backs off, the winner renews, and once the holder
+    stops renewing past the staleness window the loser takes over."""
+
+    import time
+
+    from tensor_fusion_amd.k8s.fake_apiserver import serve_in_thread
+    from tensor_fusion_amd.operator import run_leader_election
+
+    srv, url, us = serve_in_thread()
+    c1 = K8sClient(url)
+    c2 = K8sClient(url)
+
+    lost = []
+    id1 = run_leader_election(c1, "op-lease", "default", duration_s=1,
+                              identity="op-1", on_lost=lambda: lost.append(1))
+    assert id1 == "op-1"
+    # second replica cannot take a held, renewed lease
+    id2 = run_leader_election(c2, "op-lease", "default", duration_s=1,
+                              identity="op-2", _max_wait=1.5)
+    assert id2 == ""
+    # holder stops renewing (simulate crash): acquire_lease from op-2
+    # succeeds once the lease is 2x stale
+    # break op-1's renewals (simulated crash) and wait out the
+    # 2x-duration staleness window
+    c1.base_url = "http://127.0.0.1:1"
+    deadline = time.time() + 15
+    got = False
+    while time.time() < deadline:
+        if c2.acquire_lease("op-lease", "default", "op-2", duration_s=1):
+            got = True
+            break
+        time.sleep(0.3)
+    assert got
+    us.should_exit = True
