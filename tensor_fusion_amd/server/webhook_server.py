@@ -159,3 +159,51 @@ def create_webhook_app(mutator: PodMutator):
                     "body": json.dumps(out).encode()})
 
     return app
+
+
+def serve_multiprocess(app, port: int, workers: int = 2,
+                       host: str = "0.0.0.0",
+                       ssl_certfile: str = "", ssl_keyfile: str = "",
+                       app_factory=None):
+    """Serve the webhook from `workers` forked processes accepting on
+    ONE shared listening socket — admission is stateless per request
+    (the mutator reads a read-mostly store snapshot), so the kernel's
+    accept queue load-balances across processes and throughput scales
+    with cores instead of being GIL-bound. Returns (socket, [pids]);
+    close the socket and signal the pids to stop.
+
+    Reference parity: the Go webhook serves from one multiplexed
+    process; this is the CPython equivalent of its goroutine
+    concurrency."""
+
+    import os
+    import socket as _socket
+
+    import uvicorn
+
+    # proto must be IPPROTO_TCP (not 0): accepted sockets inherit it,
+    # and asyncio's _set_nodelay only disables Nagle when proto says
+    # TCP — with proto 0 every response stalls ~40 ms on delayed ACK.
+    sock = _socket.socket(_socket.AF_INET, _socket.SOCK_STREAM,
+                          _socket.IPPROTO_TCP)
+    sock.setsockopt(_socket.SOL_SOCKET, _socket.SO_REUSEADDR, 1)
+    sock.bind((host, port))
+    sock.listen(4096)
+    pids = []
+    for _ in range(max(1, workers)):
+        pid = os.fork()
+        if pid == 0:
+            kw = {}
+            if ssl_certfile:
+                kw = {"ssl_certfile": ssl_certfile,
+                      "ssl_keyfile": ssl_keyfile}
+            # app_factory (zero-arg, called post-fork) exists because
+            # threads — informers, watch streams — do not survive
+            # fork: state that must stay live is rebuilt in the child
+            child_app = app_factory() if app_factory else app
+            cfg = uvicorn.Config(child_app, log_level="warning", **kw)
+            srv = uvicorn.Server(cfg)
+            srv.run(sockets=[sock])
+            os._exit(0)
+        pids.append(pid)
+    return sock, pids

@@ -76,3 +76,71 @@ def test_webhook_qps_passthrough_path():
     qps = _qps(client, _plain_pod)
     print(f"webhook passthrough path: {qps:.0f} admissions/s")
     assert qps > 150, qps
+
+
+def test_multiprocess_webhook_serving():
+    """serve_multiprocess: N forked acceptors on one shared TCP socket,
+    app_factory called post-fork, Nagle disabled (a response must not
+    take the 40 ms delayed-ACK stall)."""
+
+    import json
+    import os
+    import signal
+    import socket
+    import time
+
+    import requests
+
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.server.webhook_server import (
+        create_webhook_app, serve_multiprocess)
+    from tensor_fusion_amd.webhook import PodMutator
+
+    with socket.socket() as sk:
+        sk.bind(("127.0.0.1", 0))
+        port = sk.getsockname()[1]
+
+    def factory():
+        return create_webhook_app(PodMutator(Store()))
+
+    lsock, pids = serve_multiprocess(None, port, workers=2,
+                                     host="127.0.0.1",
+                                     app_factory=factory)
+    try:
+        assert len(pids) == 2
+        body = json.dumps({
+            "apiVersion": "admission.k8s.io/v1", "kind": "AdmissionReview",
+            "request": {"uid": "u1", "namespace": "default",
+                        "object": {"metadata": {"name": "p",
+                                                "namespace": "default"},
+                                   "spec": {"containers": [
+                                       {"name": "m", "image": "i"}]}}}})
+        s = requests.Session()
+        deadline = time.time() + 15
+        r = None
+        while time.time() < deadline:
+            try:
+                r = s.post(f"http://127.0.0.1:{port}/mutate-v1-pod",
+                           data=body,
+                           headers={"content-type": "application/json"},
+                           timeout=2)
+                break
+            except requests.ConnectionError:
+                time.sleep(0.2)
+        assert r is not None and r.status_code == 200
+        assert r.json()["response"]["allowed"] is True
+        # Nagle check: a warm request must be far below the 40 ms
+        # delayed-ACK floor
+        lats = []
+        for _ in range(10):
+            t0 = time.perf_counter()
+            s.post(f"http://127.0.0.1:{port}/mutate-v1-pod", data=body,
+                   headers={"content-type": "application/json"},
+                   timeout=2)
+            lats.append(time.perf_counter() - t0)
+        assert min(lats) < 0.035, f"nagle stall? {min(lats)*1e3:.1f}ms"
+    finally:
+        for pid in pids:
+            os.kill(pid, signal.SIGTERM)
+            os.waitpid(pid, 0)
+        lsock.close()

@@ -70,14 +70,22 @@ def _load_proc(port: int, n: int, conc: int, seed: int, q):
     q.put((wall, lat))
 
 
-def run(n_total: int, procs: int, conc: int) -> dict:
+def run(n_total: int, procs: int, conc: int,
+        server_workers: int = 1) -> dict:
     app = create_webhook_app(PodMutator(Store()))
     with socket.socket() as sk:
         sk.bind(("127.0.0.1", 0))
         port = sk.getsockname()[1]
-    server = uvicorn.Server(uvicorn.Config(
-        app, host="127.0.0.1", port=port, log_level="error"))
-    threading.Thread(target=server.run, daemon=True).start()
+    server = None
+    if server_workers > 1:
+        from tensor_fusion_amd.server.webhook_server import \
+            serve_multiprocess
+        lsock, worker_pids = serve_multiprocess(
+            app, port, workers=server_workers, host="127.0.0.1")
+    else:
+        server = uvicorn.Server(uvicorn.Config(
+            app, host="127.0.0.1", port=port, log_level="error"))
+        threading.Thread(target=server.run, daemon=True).start()
     import requests
     for _ in range(200):
         try:
@@ -99,13 +107,20 @@ def run(n_total: int, procs: int, conc: int) -> dict:
     for p in ps:
         p.join()
     wall = time.perf_counter() - t0
-    server.should_exit = True
+    if server is not None:
+        server.should_exit = True
+    else:
+        import signal
+        for pid in worker_pids:
+            os.kill(pid, signal.SIGTERM)
+        lsock.close()
     lat = sorted(x for (_, ls) in results for x in ls)
     return {
         "qps": per * procs / wall,
         "p50_ms": statistics.median(lat) * 1e3,
         "p99_ms": lat[int(len(lat) * 0.99) - 1] * 1e3,
         "n": per * procs, "procs": procs, "conc_per_proc": conc,
+        "server_workers": server_workers,
     }
 
 
@@ -114,7 +129,8 @@ if __name__ == "__main__":
     ap.add_argument("--n", type=int, default=4000)
     ap.add_argument("--procs", type=int, default=4)
     ap.add_argument("--conc", type=int, default=10)
+    ap.add_argument("--server-workers", type=int, default=1)
     args = ap.parse_args()
-    r = run(args.n, args.procs, args.conc)
+    r = run(args.n, args.procs, args.conc, args.server_workers)
     print(json.dumps({k: round(v, 2) if isinstance(v, float) else v
                       for k, v in r.items()}))
