@@ -303,16 +303,10 @@ class GpuAllocator:
         nid = nid[good_mask]
         # scores
         sname = type(self.strategy).__name__
-        tw, vw = self.strategy.tflops_weight, self.strategy.vram_weight
-        af = (tw * (1.0 - (soa["av_t"][idx] - r.tflops) / soa["cap_t"][idx])
-              + vw * (1.0 - (soa["av_v"][idx] - r.vram) / soa["cap_v"][idx]))
-        af = np.clip(af, 0.0, 1.0)
-        if sname == "CompactFirst":
-            gpu_score = 100.0 * af
-        else:
-            gpu_score = 100.0 * (1.0 - af)
         nn = len(soa["node_names"])
         if sname == "NodeCompactGPULowLoad":
+            # node score is pure utilization — skip the per-GPU affinity
+            # score entirely (it is only consumed by the other branches)
             usage = (0.5 * (1.0 - soa["av_t"][idx] / soa["cap_t"][idx])
                      + 0.5 * (1.0 - soa["av_v"][idx] / soa["cap_v"][idx]))
             sums = np.bincount(nid, weights=usage, minlength=nn)
@@ -321,6 +315,15 @@ class GpuAllocator:
                 node_score = np.where(cnts > 0, 100.0 * sums /
                                       np.maximum(cnts, 1), 0.0)
         else:
+            tw = self.strategy.tflops_weight
+            vw = self.strategy.vram_weight
+            af = (tw * (1.0 - (soa["av_t"][idx] - r.tflops)
+                        / soa["cap_t"][idx])
+                  + vw * (1.0 - (soa["av_v"][idx] - r.vram)
+                          / soa["cap_v"][idx]))
+            af = np.clip(af, 0.0, 1.0)
+            gpu_score = 100.0 * af if sname == "CompactFirst" \
+                else 100.0 * (1.0 - af)
             # mean of top gpu_count scores per node; k==1 → per-node max
             if req.gpu_count == 1:
                 node_score = np.full(nn, -1.0)

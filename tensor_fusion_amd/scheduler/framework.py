@@ -174,6 +174,7 @@ class Scheduler:
         self._ready_nodes: Dict[str, bool] = {
             n.meta.name: n.status_phase == "Ready"
             for n in store.list("Node")}
+        self._ready_list: Optional[List[str]] = None  # cache of _nodes()
         store.on_change("Node", self._on_node_event)
 
     def _on_node_event(self, event: str, obj):
@@ -183,6 +184,7 @@ class Scheduler:
             else:
                 self._ready_nodes[obj.meta.name] = (
                     obj.status_phase == "Ready")
+            self._ready_list = None
 
     # ------------------------------------------------------------- binding
 
@@ -204,7 +206,10 @@ class Scheduler:
 
     def _nodes(self) -> List[str]:
         with self._node_mu:
-            return [n for n, ready in self._ready_nodes.items() if ready]
+            if self._ready_list is None:
+                self._ready_list = [n for n, ready in
+                                    self._ready_nodes.items() if ready]
+            return self._ready_list
 
     # -------------------------------------------------------------- cycle
 
