@@ -172,12 +172,25 @@ class Store:
         bounded-retry loop under thread contention)."""
 
         with self._Mutate(self):
-            obj = self.get(kind, name, namespace)
-            before = copy.deepcopy(obj)
+            key = f"{namespace}/{name}" if namespace else name
+            bucket = self._objs.get(kind, {})
+            cur = bucket.get(key)
+            if cur is None:
+                raise NotFound(f"{kind} {key}")
+            obj = copy.deepcopy(cur)
             fn(obj)
-            if obj == before:  # dataclass field-wise eq, cheaper than
-                return obj     # asdict x2; no-op: no rv bump, no events
-            return self.update(obj)
+            # no-op detection against the STORED original (dataclass
+            # field-wise eq) — saves the old second "before" deepcopy
+            if obj == cur:
+                return obj
+            self._rv += 1
+            obj.meta.resource_version = self._rv
+            bucket[key] = obj  # private copy: fn ran on it under the lock
+            self._persist(kind)
+            self._notify("MODIFIED", obj)
+            # callers get their own copy so later mutation of the return
+            # value cannot alias the stored object
+            return copy.deepcopy(obj)
 
     def delete(self, kind: str, name: str, namespace: str = "") -> None:
         key = f"{namespace}/{name}" if namespace else name
