@@ -310,3 +310,25 @@ def test_adjust_allocation_updates_soa_filter_view():
                         limit=Resource(cap.tflops, cap.vram, 100.0))
     scores2, _ = a.check_quota_and_filter(req2)
     assert scores2 == {}, scores2
+
+
+def test_lazy_node_scores_mapping_semantics():
+    """LazyNodeScores must behave exactly like {node: NodeScore} for
+    every access pattern the cycle uses (getitem/get/contains/keys/
+    values/items/iteration) while storing raw floats until read."""
+
+    from tensor_fusion_amd.allocator.gpuallocator import (LazyNodeScores,
+                                                          NodeScore)
+    m = LazyNodeScores(zip(["n1", "n2", "n3"], [1.5, 2.5, 0.5]))
+    assert set(m) == {"n1", "n2", "n3"} and len(m) == 3
+    assert "n2" in m and "nx" not in m
+    ns = m["n2"]
+    assert isinstance(ns, NodeScore)
+    assert ns.node == "n2" and ns.score == 2.5 and ns.gpu_scores == {}
+    assert m.get("n2") is ns  # materialized once, cached
+    assert m.get("nx") is None
+    # values()/items() materialize everything (defrag sorts by .score)
+    vals = sorted(m.values(), key=lambda s: s.score, reverse=True)
+    assert [v.node for v in vals] == ["n2", "n1", "n3"]
+    assert all(isinstance(v, NodeScore) for _, v in m.items())
+    assert max(m, key=lambda n: m[n].score) == "n2"
