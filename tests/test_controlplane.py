@@ -594,3 +594,33 @@ def test_owner_reference_gc_cascades():
     assert op.store.try_get("GPUNode", "gc-node") is None
     assert op.store.try_get("Pod", "hypervisor-gc-node",
                             "tensor-fusion-sys") is None
+
+
+def test_fast_deepcopy_semantics():
+    """TFObject.__deepcopy__ (structure-aware) must match copy.deepcopy
+    exactly: full independence, asdict equality, nested containers."""
+
+    import copy
+    from dataclasses import asdict
+
+    from tensor_fusion_amd.api.types import (GPU, Pod, Resource,
+                                             fast_deepcopy)
+    g = GPU()
+    g.meta.name = "g0"
+    g.meta.labels = {"a": "1"}
+    g.status.capacity = Resource(2500.0, 288 << 30, 100.0)
+    g.status.running_apps = [{"name": "w1", "namespace": "ns"}]
+    c = copy.deepcopy(g)  # routes through __deepcopy__
+    assert c == g and asdict(c) == asdict(g)
+    assert c.status is not g.status
+    assert c.status.running_apps is not g.status.running_apps
+    assert c.status.running_apps[0] is not g.status.running_apps[0]
+    c.status.running_apps[0]["name"] = "w2"
+    assert g.status.running_apps[0]["name"] == "w1"
+    c.meta.labels["a"] = "2"
+    assert g.meta.labels["a"] == "1"
+    # helper form used by the store
+    p = Pod()
+    p.meta.annotations = {"k": "v"}
+    q = fast_deepcopy(p)
+    assert q == p and q.meta.annotations is not p.meta.annotations
