@@ -374,3 +374,21 @@ def test_leader_election_over_wire():
         time.sleep(0.3)
     assert got
     us.should_exit = True
+
+
+def test_demo_cluster_script_runs(capsys):
+    """tools/demo_cluster.py (the human-facing walkthrough) must keep
+    working — it exercises the same wire flow this module asserts."""
+
+    import importlib.util
+    import pathlib
+    spec = importlib.util.spec_from_file_location(
+        "demo_cluster", pathlib.Path(__file__).resolve().parent.parent
+        / "tools" / "demo_cluster.py")
+    mod = importlib.util.module_from_spec(spec)
+    spec.loader.exec_module(mod)
+    mod.main()
+    out = capsys.readouterr().out
+    assert "12 CRDs" in out
+    assert "native+" in out, out  # connection URL published
+    assert "demo complete." in out
