@@ -91,6 +91,17 @@ def main():
         print(json.dumps({"mode": mode, "steps": args.steps}))
         return
 
+    # sliced-KV eager step — what bench.py's eager row actually runs
+    # (host-known position, no additive mask)
+    pos_host = [args.ctx + 1]
+
+    def step_sliced():
+        pos_buf.copy_(torch.tensor([pos_host[0]]), non_blocking=True)
+        logits = model(cur, pos=pos_buf, caches=caches,
+                       pos_end=pos_host[0] + 1)
+        cur.copy_(logits.argmax(-1))
+
+    sliced_gpu, sliced_wall = time_gpu(step_sliced)
     eager_gpu, eager_wall = time_gpu(step)
 
     side = torch.cuda.Stream()
@@ -107,6 +118,9 @@ def main():
 
     out = {
         "batch": args.batch, "ctx": args.ctx,
+        "sliced_eager_gpu_ms": round(sliced_gpu, 4),
+        "sliced_eager_wall_ms": round(sliced_wall, 4),
+        "tok_s_sliced_eager": round(args.batch / sliced_wall * 1e3, 1),
         "eager_gpu_ms": round(eager_gpu, 4),
         "eager_wall_ms": round(eager_wall, 4),
         "graph_gpu_ms": round(graph_gpu, 4),
