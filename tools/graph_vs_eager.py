@@ -53,6 +53,15 @@ def main():
         cur.copy_(logits.argmax(-1))
         pos_buf.add_(one)
 
+    if os.environ.get("TF_DIAG_WARM", ""):
+        # ramp SCLK/power state the way bench's sustained child runs do
+        a = torch.randn(8192, 8192, device="cuda", dtype=torch.bfloat16)
+        t0 = time.perf_counter()
+        while time.perf_counter() - t0 < float(
+                os.environ.get("TF_DIAG_WARM", "20")):
+            a @ a
+        torch.cuda.synchronize()
+
     def time_gpu(fn, n=24):
         for _ in range(6):
             fn()
