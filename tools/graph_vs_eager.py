@@ -19,6 +19,9 @@ import torch
 
 sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
+# BEFORE the model import: models/llama.py snapshots TF_FUSED_OPS at
+# import time (bench children get it from their process env)
+os.environ.setdefault("TF_FUSED_OPS", "1")
 from tensor_fusion_amd.models.llama import build_model  # noqa: E402
 
 
@@ -32,7 +35,6 @@ def main():
     args = ap.parse_args()
     mode = os.environ.get("TF_DIAG_MODE", "")
 
-    os.environ.setdefault("TF_FUSED_OPS", "1")
     model = build_model(args.model, device="cuda", dtype=torch.bfloat16)
     cfg = model.cfg
     total = args.ctx + args.steps + 8
