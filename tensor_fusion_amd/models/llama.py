@@ -47,6 +47,9 @@ CONFIGS = {
 
 
 _FUSED_OPS = os.environ.get("TF_FUSED_OPS") == "1"
+# grouped-query SDPA without materializing repeat_interleave'd caches
+# (4x cache traffic in decode); TF_SDPA_GQA=0 restores the repeat path
+_SDPA_GQA = os.environ.get("TF_SDPA_GQA", "1") != "0"
 
 
 class RMSNorm(nn.Module):
@@ -161,12 +164,14 @@ class Attention(nn.Module):
                 k = k_cache[:, :, :end]
                 v = v_cache[:, :, :end]
             rep = cfg.heads // cfg.kv_heads
-            if rep > 1:
+            if rep > 1 and not _SDPA_GQA:
                 k = k.repeat_interleave(rep, dim=1)
                 v = v.repeat_interleave(rep, dim=1)
             o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
                                                is_causal=T > 1 and
-                                               mask is None)
+                                               mask is None,
+                                               enable_gqa=rep > 1 and
+                                               _SDPA_GQA)
             o = o.transpose(1, 2).reshape(B, T, -1)
             return self._proj(self.wo, o, 3)
         q = self._proj(self.wq, x, 0).view(B, T, cfg.heads,
@@ -194,12 +199,14 @@ class Attention(nn.Module):
                 k = k_cache[:, :, :end]
                 v = v_cache[:, :, :end]
         rep = cfg.heads // cfg.kv_heads
-        if rep > 1:
+        if rep > 1 and not _SDPA_GQA:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
         causal = T > 1 and mask is None
         o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
-                                           is_causal=causal)
+                                           is_causal=causal,
+                                           enable_gqa=rep > 1 and
+                                           _SDPA_GQA)
         o = o.transpose(1, 2).reshape(B, T, -1)
         return self._proj(self.wo, o, 3)
 
