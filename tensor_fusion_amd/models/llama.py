@@ -164,14 +164,16 @@ class Attention(nn.Module):
                 k = k_cache[:, :, :end]
                 v = v_cache[:, :, :end]
             rep = cfg.heads // cfg.kv_heads
-            if rep > 1 and not _SDPA_GQA:
+            # grouped SDPA only without a mask: the masked+GQA combo
+            # drops to a slow backend (measured 7.4 -> 10.3 ms/step)
+            gqa = rep > 1 and _SDPA_GQA and mask is None
+            if rep > 1 and not gqa:
                 k = k.repeat_interleave(rep, dim=1)
                 v = v.repeat_interleave(rep, dim=1)
             o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
                                                is_causal=T > 1 and
                                                mask is None,
-                                               enable_gqa=rep > 1 and
-                                               _SDPA_GQA)
+                                               enable_gqa=gqa)
             o = o.transpose(1, 2).reshape(B, T, -1)
             return self._proj(self.wo, o, 3)
         q = self._proj(self.wq, x, 0).view(B, T, cfg.heads,
@@ -199,14 +201,14 @@ class Attention(nn.Module):
                 k = k_cache[:, :, :end]
                 v = v_cache[:, :, :end]
         rep = cfg.heads // cfg.kv_heads
-        if rep > 1 and not _SDPA_GQA:
+        gqa = rep > 1 and _SDPA_GQA and mask is None
+        if rep > 1 and not gqa:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
         causal = T > 1 and mask is None
         o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
                                            is_causal=causal,
-                                           enable_gqa=rep > 1 and
-                                           _SDPA_GQA)
+                                           enable_gqa=gqa)
         o = o.transpose(1, 2).reshape(B, T, -1)
         return self._proj(self.wo, o, 3)
 
