@@ -412,3 +412,65 @@ class TestInformerConvergenceFuzz:
             inf.stop()
         finally:
             us.should_exit = True
+
+
+class TestInformer410Recovery:
+    def test_informer_relists_after_410_gone(self):
+        """A watch that dies with 410 Gone (RV compacted away) must
+        trigger a fresh relist, not a stall (informer.py gone->relist
+        path). Driven by a stub client so the 410 is deterministic."""
+
+        import threading
+        import time as _t
+
+        from tensor_fusion_amd.k8s.client import ApiError
+        from tensor_fusion_amd.k8s.informer import Informer
+
+        relists = []
+        phase = {"n": 0}
+
+        class StubClient:
+            def list(self, kind, namespace="", label_selector=""):
+                relists.append(kind)
+                items = [{"metadata": {"name": "a",
+                                       "resourceVersion": "1"}}]
+                if len(relists) > 1:  # post-410 state has b too
+                    items.append({"metadata": {"name": "b",
+                                               "resourceVersion": "7"}})
+                return {"items": items,
+                        "metadata": {"resourceVersion": "7"}}
+
+            def watch(self, kind, namespace="", resource_version="",
+                      label_selector="", timeout_s=300):
+                phase["n"] += 1
+                if phase["n"] == 1:
+                    # one live event, then the server compacts: 410
+                    yield ("MODIFIED", {"metadata": {
+                        "name": "a", "resourceVersion": "2"}})
+                    raise ApiError(410, "Expired")
+                while True:  # healthy stream after the relist
+                    _t.sleep(0.05)
+                    yield ("MODIFIED", {"metadata": {
+                        "name": "b", "resourceVersion": "8"}})
+
+        seen = {}
+        mu = threading.Lock()
+
+        def on_event(typ, obj):
+            with mu:
+                seen[obj["metadata"]["name"]] = \
+                    obj["metadata"]["resourceVersion"]
+
+        inf = Informer(StubClient(), "GPU", on_event=on_event)
+        inf.start()
+        assert inf.wait_synced(5)
+        deadline = _t.time() + 10
+        while _t.time() < deadline:
+            with mu:
+                if seen.get("b") == "8":
+                    break
+            _t.sleep(0.05)
+        inf.stop()
+        assert len(relists) >= 2, "410 must force a relist"
+        with mu:
+            assert seen.get("b") == "8", seen
