@@ -102,11 +102,15 @@ class TPAttention(nn.Module):
                 k = k_cache[:, :, :end]
                 v = v_cache[:, :, :end]
         rep = self.heads // self.kv_heads
-        if rep > 1:
+        # grouped SDPA on the mask-free path only (the masked+GQA combo
+        # picks a slow backend — profiles/graph_vs_eager_r02.md)
+        gqa = rep > 1 and mask is None
+        if rep > 1 and not gqa:
             k = k.repeat_interleave(rep, dim=1)
             v = v.repeat_interleave(rep, dim=1)
         o = F.scaled_dot_product_attention(q, k, v, attn_mask=mask,
-                                           is_causal=T > 1 and mask is None)
+                                           is_causal=T > 1 and mask is None,
+                                           enable_gqa=gqa)
         o = o.transpose(1, 2).reshape(B, T, -1)
         return self.wo(o)
 
