@@ -615,6 +615,7 @@ class Node(TFObject):
     capacity: Dict[str, float] = field(default_factory=dict)
     status_phase: str = "Ready"
     address: str = "127.0.0.1"
+    taints: List[Dict[str, str]] = field(default_factory=list)
 
 
 ALL_KINDS = [

@@ -9,6 +9,8 @@ brand-new MI355X-native implementation.
 
 # ---------------------------------------------------------------- domain
 Domain = "tensor-fusion.ai"
+NodeUsedByTaintKey = f"{Domain}/used-by"  # legacy NoSchedule taint
+TensorFusionSystemName = "tensor-fusion"
 Version = "0.2.0"  # round-2 build
 
 # ---------------------------------------------------------------- labels

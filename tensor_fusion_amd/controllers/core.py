@@ -130,6 +130,7 @@ class NodeReconciler(Reconciler):
 
     def reconcile(self, req: Request):
         node = self.store.get(self.kind, req.name)
+        self._remove_legacy_taint(node)
         pool = self._match_pool(node)
         if pool is None:
             return 0.0
@@ -145,6 +146,26 @@ class NodeReconciler(Reconciler):
             except AlreadyExists:
                 pass
         return 0.0
+
+    def _remove_legacy_taint(self, node) -> bool:
+        """Drop the deprecated `tensor-fusion.ai/used-by=tensor-fusion:
+        NoSchedule` taint from nodes (reference node_controller.go:300
+        removeTensorFusionTaint; skipped while the node is deleting)."""
+
+        if node.meta.deletion_ts is not None:
+            return False
+        taints = getattr(node, "taints", None) or []
+        keep = [t for t in taints
+                if not (t.get("key") == C.NodeUsedByTaintKey
+                        and t.get("value") == C.TensorFusionSystemName
+                        and t.get("effect") == "NoSchedule")]
+        if len(keep) == len(taints):
+            return False
+
+        def _p(obj):
+            obj.taints = keep
+        self.store.patch(self.kind, node.meta.name, "", _p)
+        return True
 
     def _match_pool(self, node) -> Optional[GPUPool]:
         for pool in self.store.list("GPUPool"):

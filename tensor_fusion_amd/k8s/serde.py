@@ -339,9 +339,12 @@ def node_to_k8s(n: T.Node) -> Dict[str, Any]:
     if labels:
         meta.setdefault("labels", {}).update(labels)
     ready = n.status_phase == "Ready"
+    spec: Dict[str, Any] = {}
+    if n.taints:
+        spec["taints"] = [dict(t) for t in n.taints]
     return {
         "apiVersion": "v1", "kind": "Node", "metadata": meta,
-        "spec": {},
+        "spec": spec,
         "status": {
             "capacity": {k: str(v) for k, v in n.capacity.items()},
             "addresses": [{"type": "InternalIP", "address": n.address}],
@@ -367,4 +370,6 @@ def node_from_k8s(d: Dict[str, Any]) -> T.Node:
     for a in st.get("addresses") or []:
         if a.get("type") == "InternalIP":
             n.address = a.get("address", n.address)
+    n.taints = [dict(t) for t in
+                (d.get("spec") or {}).get("taints") or []]
     return n

@@ -624,3 +624,48 @@ def test_fast_deepcopy_semantics():
     p.meta.annotations = {"k": "v"}
     q = fast_deepcopy(p)
     assert q == p and q.meta.annotations is not p.meta.annotations
+
+
+def test_node_legacy_taint_removed():
+    """NodeReconciler strips the deprecated
+    tensor-fusion.ai/used-by=tensor-fusion:NoSchedule taint from nodes
+    (reference node_controller.go:300); unrelated taints survive and
+    deleting nodes are left alone."""
+
+    import tensor_fusion_amd.constants as C
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.api.types import Node
+    from tensor_fusion_amd.controllers.core import NodeReconciler, Request
+
+    store = Store()
+    rec = NodeReconciler(store)
+    n = Node()
+    n.meta.name = "n0"
+    n.taints = [
+        {"key": C.NodeUsedByTaintKey, "value": C.TensorFusionSystemName,
+         "effect": "NoSchedule"},
+        {"key": "other.io/maintenance", "value": "x",
+         "effect": "NoExecute"},
+    ]
+    store.create(n)
+    rec.reconcile(Request("Node", "n0", ""))
+    got = store.get("Node", "n0")
+    assert [t["key"] for t in got.taints] == ["other.io/maintenance"]
+
+    # deleting node: untouched
+    n2 = Node()
+    n2.meta.name = "n1"
+    n2.meta.deletion_ts = 1.0
+    n2.taints = [{"key": C.NodeUsedByTaintKey,
+                  "value": C.TensorFusionSystemName,
+                  "effect": "NoSchedule"}]
+    store.create(n2)
+    rec.reconcile(Request("Node", "n1", ""))
+    assert len(store.get("Node", "n1").taints) == 1
+
+    # serde: taints survive the corev1 round trip
+    from tensor_fusion_amd.k8s import serde
+    wire = serde.node_to_k8s(got)
+    assert wire["spec"]["taints"] == got.taints
+    back = serde.node_from_k8s(wire)
+    assert back.taints == got.taints
