@@ -187,6 +187,39 @@ class GPUNodeReconciler(Reconciler):
             return [Request("GPUNode", obj.status.node)]
         return []
 
+    PROBE_NS = "tensor-fusion-sys"
+
+    def _ensure_driver_probe(self, node: str):
+        """One-shot ROCm driver probe pod, created BEFORE the hypervisor
+        (reference gpunode_controller.go:790-863 runs a probe Job and
+        gates hypervisor deployment on its success). Returns the probe's
+        rocm-version annotation when it has Succeeded, else None."""
+
+        name = f"driver-probe-{node}"
+        probe = self.store.try_get("Pod", name, self.PROBE_NS)
+        if probe is None:
+            pod = Pod()
+            pod.meta.name = name
+            pod.meta.namespace = self.PROBE_NS
+            pod.meta.labels[C.LabelComponent] = "driver-probe"
+            pod.meta.labels[C.LabelNode] = node
+            pod.meta.owner = f"GPUNode//{node}"
+            pod.containers = [Container(
+                name="probe",
+                command=["python", "-c",
+                         "import torch; assert torch.cuda.is_available();"
+                         "print(torch.version.hip)"])]
+            pod.status.node = node
+            try:
+                self.store.create(pod)
+            except AlreadyExists:
+                pass
+            return None
+        if probe.status.phase == "Succeeded":
+            return probe.meta.annotations.get(
+                f"{C.Domain}/rocm-version", "unknown")
+        return None
+
     def reconcile(self, req: Request):
         gn = self.store.get(self.kind, req.name)
         gpus = [g for g in self.store.list("GPU")
@@ -196,6 +229,17 @@ class GPUNodeReconciler(Reconciler):
         for g in gpus:
             total = total.add(g.status.capacity)
             avail = avail.add(g.status.available)
+
+        # driver probe gates hypervisor rollout — unless devices are
+        # already discovered (working driver is then self-evident, e.g.
+        # operator restart over a live fleet)
+        rocm = self._ensure_driver_probe(req.name)
+        if rocm is None and not gpus:
+            def _pending(obj):
+                obj.status.phase = "Pending"
+                obj.status.hypervisor_ready = False
+            self.store.patch(self.kind, req.name, "", _pending)
+            return 2.0  # re-check until the probe succeeds
 
         hyp_name = f"hypervisor-{req.name}"
         hyp = self.store.try_get("Pod", hyp_name, "tensor-fusion-sys")
@@ -224,6 +268,8 @@ class GPUNodeReconciler(Reconciler):
             s.gpus = sorted(g.meta.name for g in gpus)
             s.total, s.available = total, avail
             s.hypervisor_ready = True
+            if rocm and rocm != "unknown":
+                s.rocm_version = rocm
             s.phase = "Running" if gpus else "Pending"
         self.store.patch(self.kind, req.name, "", _p)
         return 0.0

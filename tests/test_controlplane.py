@@ -587,8 +587,22 @@ def test_owner_reference_gc_cascades():
     gn = op.store.try_get("GPUNode", "gc-node")
     assert gn is not None and gn.meta.owner == "Node//gc-node"
     op.tick()
+    # driver probe gates the hypervisor rollout (gpunode_controller.go
+    # :790-863): no hypervisor pod until the probe pod Succeeds
+    probe = op.store.try_get("Pod", "driver-probe-gc-node",
+                             "tensor-fusion-sys")
+    assert probe is not None and probe.meta.owner == "GPUNode//gc-node"
+    assert op.store.try_get("Pod", "hypervisor-gc-node",
+                            "tensor-fusion-sys") is None
+
+    def _ok(obj):
+        obj.status.phase = "Succeeded"
+        obj.meta.annotations["tensor-fusion.ai/rocm-version"] = "7.2.0"
+    op.store.patch("Pod", "driver-probe-gc-node", "tensor-fusion-sys", _ok)
+    op.tick()
     hyp = op.store.try_get("Pod", "hypervisor-gc-node", "tensor-fusion-sys")
     assert hyp is not None and hyp.meta.owner == "GPUNode//gc-node"
+    assert op.store.get("GPUNode", "gc-node").status.rocm_version == "7.2.0"
 
     op.store.delete("Node", "gc-node")
     assert op.store.try_get("GPUNode", "gc-node") is None
