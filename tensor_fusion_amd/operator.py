@@ -55,6 +55,7 @@ class Operator:
     config: Optional[ConfigWatcher] = None
     _threads: List[threading.Thread] = field(default_factory=list)
     _stop: threading.Event = field(default_factory=threading.Event)
+    _last_rebalance: float = 0.0
 
     # ------------------------------------------------------ admission
 
@@ -108,9 +109,36 @@ class Operator:
         self.allocator.sync_dirty()
         self.quota.sync_dirty()
         self.autoscaler.tick()
+        self._maybe_rebalance()
         self._record_pool_metrics()
         self.metrics.flush()
         self.alerts.evaluate()
+
+    def _maybe_rebalance(self, now: Optional[float] = None):
+        """Periodic re-balancer (reference SchedulingConfigTemplate
+        :241): when a pool enables defrag, or a scheduling template
+        sets rebalanceIntervalS, run a defrag campaign on that cadence
+        (DefragController applies its own campaign cooldown on top)."""
+
+        import time as _t
+        now = now if now is not None else _t.time()
+        interval = 0
+        for t in self.store.list("SchedulingConfigTemplate"):
+            if t.rebalance_interval_s > 0:
+                interval = (t.rebalance_interval_s if not interval
+                            else min(interval, t.rebalance_interval_s))
+        enabled = interval > 0 or any(
+            p.node_manager.defrag_enabled
+            for p in self.store.list("GPUPool"))
+        if not enabled:
+            return
+        if interval and now - self._last_rebalance < interval:
+            return
+        self._last_rebalance = now
+        try:
+            self.defrag.run_campaign(now)
+        except Exception:
+            pass
 
     def _record_pool_metrics(self):
         for pool in self.store.list("GPUPool"):
@@ -133,6 +161,7 @@ class Operator:
         def sched_loop():
             while not self._stop.wait(scheduler_interval_s):
                 try:
+                    self._maybe_rebalance()
                     self.scheduler.schedule_pending()
                     self.allocator.sync_dirty()
                     self.quota.sync_dirty()

@@ -162,3 +162,43 @@ class TestDefragAtScale:
                 assert g.status.node not in plan.candidate_nodes
         evicted = d.execute_due_evictions(now=2e9)
         assert set(evicted) <= set(plan.evict_pods)
+
+
+def test_rebalancer_triggers_defrag_campaigns():
+    """The re-balancer (SchedulingConfigTemplate :241) runs defrag
+    campaigns on the configured interval; without any enablement it
+    never fires."""
+
+    from tensor_fusion_amd.api.types import (GPUPool,
+                                             SchedulingConfigTemplate)
+    from tensor_fusion_amd.operator import build_operator
+
+    op = build_operator()
+    calls = []
+    op.defrag.run_campaign = lambda now=None: calls.append(now)
+
+    op._maybe_rebalance(now=100.0)
+    assert not calls  # nothing enabled
+
+    t = SchedulingConfigTemplate()
+    t.meta.name = "tmpl"
+    t.rebalance_interval_s = 60
+    op.store.create(t)
+    op._maybe_rebalance(now=100.0)
+    assert len(calls) == 1
+    op._maybe_rebalance(now=130.0)  # inside the interval
+    assert len(calls) == 1
+    op._maybe_rebalance(now=161.0)
+    assert len(calls) == 2
+
+    # pool-level defrag_enabled alone also fires (no interval gate;
+    # the DefragController's own cooldown applies)
+    op2 = build_operator()
+    calls2 = []
+    op2.defrag.run_campaign = lambda now=None: calls2.append(now)
+    pool = GPUPool()
+    pool.meta.name = "p"
+    pool.node_manager.defrag_enabled = True
+    op2.store.create(pool)
+    op2._maybe_rebalance(now=5.0)
+    assert calls2
