@@ -215,8 +215,36 @@ class Accelerator:
         arr = (ctypes.c_int32 * len(xcds))(*xcds)
         return self._lib.tf_accel_assign_partition(device, arr, len(xcds)) == 0
 
+    def remove_partition(self, device: int, xcds: List[int]) -> bool:
+        arr = (ctypes.c_int32 * len(xcds))(*xcds)
+        return self._lib.tf_accel_remove_partition(
+            device, arr, len(xcds)) == 0
+
     def snapshot(self, pid: int, dest: str) -> int:
         return self._lib.tf_accel_snapshot(pid, dest.encode())
+
+    def resume(self, pid: int, src: str) -> int:
+        return self._lib.tf_accel_resume(pid, src.encode())
+
+    def device_count(self) -> int:
+        n = ctypes.c_int(0)
+        if self._lib.tf_accel_device_count(ctypes.byref(n)) != 0:
+            raise RuntimeError("tf_accel_device_count failed")
+        return n.value
+
+    def register_log_callback(self, fn) -> None:
+        """fn(level:int, msg:str); keeps the ctypes thunk alive on self
+        (the C side stores the raw pointer)."""
+
+        cb_t = ctypes.CFUNCTYPE(None, ctypes.c_int, ctypes.c_char_p)
+
+        def thunk(level, msg):
+            try:
+                fn(int(level), (msg or b"").decode(errors="replace"))
+            except Exception:
+                pass
+        self._log_cb = cb_t(thunk)
+        self._lib.tf_accel_register_log_callback(self._log_cb)
 
 
 class DeviceController:
