@@ -426,7 +426,15 @@ def serve_in_thread(server: Optional[FakeApiServer] = None,
         with socket.socket() as sk:
             sk.bind(("127.0.0.1", 0))
             port = sk.getsockname()[1]
-    cfg = uvicorn.Config(srv.app, host="127.0.0.1", port=port,
+    async def _raise_threadpool(scope, receive, send):
+        # every active watch stream occupies a threadpool slot (sync
+        # generators run via iterate_in_threadpool); the anyio default
+        # of 40 deadlocks multi-process informer tests
+        import anyio.to_thread
+        anyio.to_thread.current_default_thread_limiter().total_tokens = 256
+        await srv.app(scope, receive, send)
+
+    cfg = uvicorn.Config(_raise_threadpool, host="127.0.0.1", port=port,
                         log_level="error")
     us = uvicorn.Server(cfg)
     th = threading.Thread(target=us.run, daemon=True)

@@ -222,7 +222,8 @@ def test_operator_cli_k8s_mode(tmp_path):
     proc = subprocess.Popen(
         [sys.executable, "-m", "tensor_fusion_amd.operator", "--k8s",
          "--install-crds", "--http-port", str(http_port),
-         "--webhook-port", str(wh_port)],
+         "--webhook-port", str(wh_port),
+         "--leader-elect", "--webhook-workers", "2"],
         env=env, cwd=repo, stdout=subprocess.PIPE,
         stderr=subprocess.PIPE, text=True)
     try:
