@@ -106,7 +106,35 @@ class PodMutator:
             profile.is_local_gpu = True
         if not pod.meta.annotations.get(C.AnnoQos):
             profile.qos = self.calculate_qos(profile)
+        self._apply_recommendation(pod, profile)
         return profile
+
+    def _apply_recommendation(self, pod: Pod, profile: WorkloadProfile):
+        """Autoscaler feedback at admission (reference webhook :349):
+        when the pod's workload carries a status.recommendation and the
+        user did NOT pin resources explicitly via annotations, admit
+        with the recommended requests/limits so new replicas start at
+        the learned size."""
+
+        wl_name = pod.meta.labels.get(C.LabelWorkload)
+        if not wl_name and pod.meta.annotations.get(C.AnnoAutoscale):
+            wl_name = f"{pod.meta.name}-wl"
+        if not wl_name:
+            return
+        wl = self.store.try_get("TensorFusionWorkload", wl_name,
+                                pod.meta.namespace)
+        rec = getattr(getattr(wl, "status", None), "recommendation", None)
+        if not rec or not rec.resources.requests.vram and \
+                not rec.resources.requests.tflops:
+            return
+        a = pod.meta.annotations
+        req, lim = rec.resources.requests, rec.resources.limits
+        if C.AnnoTflopsRequest not in a and req.tflops:
+            profile.resources.requests.tflops = req.tflops
+            profile.resources.limits.tflops = lim.tflops or req.tflops
+        if C.AnnoVramRequest not in a and req.vram:
+            profile.resources.requests.vram = req.vram
+            profile.resources.limits.vram = lim.vram or req.vram
 
     @staticmethod
     def calculate_qos(profile: WorkloadProfile) -> str:

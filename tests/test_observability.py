@@ -346,3 +346,47 @@ def test_metrics_remote_sink_ships_and_backlogs():
         assert received[0].count("tf_worker_metrics") >= 2  # backlog shipped
     finally:
         srv.should_exit = True
+
+
+def test_recommendation_applied_at_admission():
+    """Autoscaler feedback loop closes through the webhook (reference
+    pod_webhook :349): a workload's status.recommendation resizes NEW
+    replicas admitted without explicit resource annotations; explicit
+    annotations always win."""
+
+    import tensor_fusion_amd.constants as C
+    from tensor_fusion_amd.api.store import Store
+    from tensor_fusion_amd.api.types import (Pod, Recommendation,
+                                             Requirements, Resource,
+                                             TensorFusionWorkload)
+    from tensor_fusion_amd.webhook import PodMutator
+
+    store = Store()
+    wl = TensorFusionWorkload()
+    wl.meta.name = "app-wl"
+    wl.meta.namespace = "default"
+    wl.status.recommendation = Recommendation(
+        resources=Requirements(requests=Resource(400.0, 32 << 30, 0),
+                               limits=Resource(400.0, 32 << 30, 0)),
+        reason="percentile")
+    store.create(wl)
+    mut = PodMutator(store)
+
+    pod = Pod()
+    pod.meta.name = "app-1"
+    pod.meta.namespace = "default"
+    pod.meta.labels = {C.LabelEnabled: "true", C.LabelWorkload: "app-wl"}
+    prof = mut.parse(pod)
+    assert prof.resources.requests.tflops == 400.0
+    assert prof.resources.requests.vram == 32 << 30
+
+    # explicit annotations pin the size — recommendation must not win
+    pod2 = Pod()
+    pod2.meta.name = "app-2"
+    pod2.meta.namespace = "default"
+    pod2.meta.labels = {C.LabelEnabled: "true", C.LabelWorkload: "app-wl"}
+    pod2.meta.annotations = {C.AnnoTflopsRequest: "100",
+                             C.AnnoVramRequest: str(8 << 30)}
+    prof2 = mut.parse(pod2)
+    assert prof2.resources.requests.tflops == 100.0
+    assert prof2.resources.requests.vram == 8 << 30
