@@ -244,17 +244,41 @@ class GPUNodeReconciler(Reconciler):
         hyp_name = f"hypervisor-{req.name}"
         hyp = self.store.try_get("Pod", hyp_name, "tensor-fusion-sys")
         if hyp is None:
+            pool = self.store.try_get("GPUPool", gn.pool) if gn.pool                 else None
             pod = Pod()
             pod.meta.name = hyp_name
             pod.meta.namespace = "tensor-fusion-sys"
             pod.meta.labels[C.LabelComponent] = C.ComponentHypervisor
             pod.meta.labels[C.LabelNode] = req.name
+            if pool is not None:
+                pod.meta.labels[C.LabelPool] = pool.meta.name
             pod.meta.owner = f"GPUNode//{req.name}"
+            # provider env + isolation args from the pool's component
+            # config (reference gpunode_controller.go:1029-1143)
+            iso = (pool.node_manager.isolation_default
+                   if pool is not None else C.IsolationSoft)
+            env = {
+                "TF_NODE_NAME": req.name,
+                "TF_ACCELERATOR_LIB": f"/opt/tensor-fusion/"
+                                      f"{C.AcceleratorLibName}",
+                "TF_ISOLATION_DEFAULT": iso,
+                "HSA_ENABLE_IPC_MODE_LEGACY": "0",
+            }
+            tmpl = (pool.components.hypervisor_template
+                    if pool is not None else {}) or {}
+            env.update({k: str(v) for k, v in
+                        (tmpl.get("env") or {}).items()})
+            image = (pool.components.hypervisor_image
+                     if pool is not None else "")
             pod.containers = [Container(
                 name="hypervisor",
-                command=["python", "-m", "tensor_fusion_amd.hypervisor.main",
-                         "--node", req.name],
-                env={"TF_NODE_NAME": req.name})]
+                image=image or "tensor-fusion/hypervisor:latest",
+                command=["python", "-m",
+                         "tensor_fusion_amd.hypervisor.main",
+                         "--node", req.name,
+                         "--backend", "kubernetes",
+                         "--device-plugin"],
+                env=env)]
             pod.status.node = req.name  # host pod, not scheduled by us
             pod.status.phase = "Running"
             try:
