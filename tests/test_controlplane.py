@@ -158,7 +158,7 @@ class TestControllers:
         assert gn.status.total.vram == 2 * C.MI355X_VRAM_BYTES
         pool = store.get("GPUPool", "pool-a")
         assert pool.status.gpu_count == 2
-        assert not mgr.errors(), mgr.errors()[:1]  # no silent reconcile
+        assert not mgr.errors, mgr.errors[:1]  # no silent reconcile
         # failures (this exact assert caught a bad field name once)
         # oversell: 500% tflops, +50% vram
         assert pool.status.virtual_total.tflops == pytest.approx(2 * 2500 * 5)
