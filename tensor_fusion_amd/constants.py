@@ -11,6 +11,7 @@ brand-new MI355X-native implementation.
 Domain = "tensor-fusion.ai"
 NodeUsedByTaintKey = f"{Domain}/used-by"  # legacy NoSchedule taint
 TensorFusionSystemName = "tensor-fusion"
+Finalizer = f"{Domain}/finalizer"  # worker pods: dealloc-before-delete
 Version = "0.2.0"  # round-2 build
 
 # ---------------------------------------------------------------- labels

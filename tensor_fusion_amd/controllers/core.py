@@ -308,6 +308,9 @@ def generate_worker_pod(wl: TensorFusionWorkload, index: int) -> Pod:
 
     prof = wl.profile
     pod = Pod()
+    # dealloc-before-delete guard (reference pod_controller.go:223
+    # handleWorkerPodFinalizer): PodReconciler strips it after dealloc
+    pod.meta.finalizers = [C.Finalizer]
     pod.meta.name = f"{wl.meta.name}-worker-{index}"
     pod.meta.namespace = wl.meta.namespace
     pod.meta.labels = {
@@ -495,6 +498,20 @@ class PodReconciler(Reconciler):
 
     def reconcile(self, req: Request):
         pod = self.store.get(self.kind, req.name, req.namespace)
+        if pod.meta.deletion_ts and C.Finalizer in pod.meta.finalizers:
+            # finalizer path: release the allocation BEFORE the object
+            # can disappear, then let the apiserver finish the delete
+            if self.allocator is not None:
+                try:
+                    self.allocator.dealloc(pod.meta.key)
+                except Exception:
+                    pass
+
+            def _strip(obj):
+                obj.meta.finalizers = [f for f in obj.meta.finalizers
+                                       if f != C.Finalizer]
+            self.store.patch(self.kind, req.name, req.namespace, _strip)
+            return 0.0
         if pod.meta.labels.get(C.LabelComponent) == C.ComponentClient:
             conn_name = pod.containers[0].env.get(
                 C.EnvConnectionName, f"{req.name}-conn") if pod.containers \

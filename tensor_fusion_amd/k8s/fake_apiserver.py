@@ -136,6 +136,14 @@ class FakeApiServer:
         tab = self._table(group, resource)
         meta = obj.setdefault("metadata", {})
         key = (meta.get("namespace", ""), meta["name"])
+        # deletion completes once a deleting object's finalizers empty
+        # (real-apiserver garbage collection semantics)
+        if typ == "MODIFIED" and meta.get("deletionTimestamp") \
+                and not meta.get("finalizers"):
+            tab.objs.pop(key, None)
+            meta["resourceVersion"] = self._next_rv()
+            self._emit(tab, "DELETED", obj)
+            return copy.deepcopy(obj)
         meta["resourceVersion"] = self._next_rv()
         if typ == "ADDED":
             self._uid += 1
@@ -353,6 +361,16 @@ class FakeApiServer:
                 if cur is None:
                     return JSONResponse({"message": "not found"},
                                         status_code=404)
+                if cur.get("metadata", {}).get("finalizers"):
+                    # real-apiserver semantics: finalizers defer the
+                    # delete — set deletionTimestamp, wait for a
+                    # controller to strip them
+                    nxt = copy.deepcopy(cur)
+                    nxt["metadata"].setdefault(
+                        "deletionTimestamp",
+                        "1970-01-01T00:00:01Z")
+                    out = self._store(group, resource, nxt, "MODIFIED")
+                    return JSONResponse(out)
                 del tab.objs[key]
                 cur = copy.deepcopy(cur)
                 cur["metadata"]["resourceVersion"] = self._next_rv()

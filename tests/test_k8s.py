@@ -474,3 +474,38 @@ class TestInformer410Recovery:
         assert len(relists) >= 2, "410 must force a relist"
         with mu:
             assert seen.get("b") == "8", seen
+
+
+class TestFinalizerFlow:
+    def test_worker_finalizer_defers_delete_until_dealloc(self):
+        """Real-apiserver finalizer semantics on the wire: DELETE of a
+        pod carrying tensor-fusion.ai/finalizer only sets
+        deletionTimestamp; once the controller strips the finalizer
+        (after dealloc) the object is actually deleted and a DELETED
+        watch event fires (reference pod_controller.go:223)."""
+
+        import tensor_fusion_amd.constants as C
+        from tensor_fusion_amd.k8s.fake_apiserver import serve_in_thread
+
+        srv, base, us = serve_in_thread()
+        try:
+            cli = K8sClient(base)
+            pod = {"apiVersion": "v1", "kind": "Pod",
+                   "metadata": {"name": "w0", "namespace": "default",
+                                "finalizers": [C.Finalizer],
+                                "labels": {
+                                    C.LabelComponent: C.ComponentWorker}},
+                   "spec": {"containers": [{"name": "w",
+                                            "image": "i"}]}}
+            cli.create(pod)
+            cli.delete("Pod", "w0", "default")
+            cur = cli.get("Pod", "w0", "default")  # still there
+            assert cur["metadata"]["deletionTimestamp"]
+            assert cur["metadata"]["finalizers"] == [C.Finalizer]
+
+            # controller strips the finalizer -> delete completes
+            cur["metadata"]["finalizers"] = []
+            cli.update(cur)
+            assert cli.try_get("Pod", "w0", "default") is None
+        finally:
+            us.should_exit = True
