@@ -509,3 +509,74 @@ class TestFinalizerFlow:
             assert cli.try_get("Pod", "w0", "default") is None
         finally:
             us.should_exit = True
+
+
+class TestWebhookTLS:
+    def test_certs_generate_and_serve_https(self, tmp_path):
+        """certs.generate produces a CA-signed server pair with the
+        right SANs; the webhook serves real HTTPS with it and a client
+        trusting only the caBundle verifies the connection — exactly
+        what a kube-apiserver does with MutatingWebhookConfiguration's
+        caBundle."""
+
+        import base64
+        import json
+        import socket
+        import ssl
+        import threading
+
+        import requests
+        import uvicorn
+
+        from tensor_fusion_amd.api.store import Store
+        from tensor_fusion_amd.k8s.certs import generate
+        from tensor_fusion_amd.server.webhook_server import \
+            create_webhook_app
+        from tensor_fusion_amd.webhook import PodMutator
+
+        d = str(tmp_path)
+        bundle = generate(d, "localhost")
+        ca = base64.b64decode(bundle)
+        assert b"BEGIN CERTIFICATE" in ca
+
+        app = create_webhook_app(PodMutator(Store()))
+        with socket.socket() as sk:
+            sk.bind(("127.0.0.1", 0))
+            port = sk.getsockname()[1]
+        srv = uvicorn.Server(uvicorn.Config(
+            app, host="127.0.0.1", port=port, log_level="error",
+            ssl_certfile=f"{d}/tls.crt", ssl_keyfile=f"{d}/tls.key"))
+        threading.Thread(target=srv.run, daemon=True).start()
+        import time
+        body = json.dumps({
+            "apiVersion": "admission.k8s.io/v1",
+            "kind": "AdmissionReview",
+            "request": {"uid": "u", "namespace": "default",
+                        "object": {"metadata": {"name": "p",
+                                                "namespace": "default"},
+                                   "spec": {"containers": [
+                                       {"name": "m", "image": "i"}]}}}})
+        deadline = time.time() + 15
+        r = None
+        while time.time() < deadline:
+            try:
+                # verify against ONLY the generated CA ("caBundle")
+                r = requests.post(
+                    f"https://localhost:{port}/mutate-v1-pod",
+                    data=body,
+                    headers={"content-type": "application/json"},
+                    verify=f"{d}/ca.crt", timeout=2)
+                break
+            except requests.ConnectionError:
+                time.sleep(0.2)
+        assert r is not None and r.status_code == 200
+        assert r.json()["response"]["allowed"] is True
+        # an untrusting client must fail verification
+        try:
+            requests.post(f"https://localhost:{port}/mutate-v1-pod",
+                          data=body, timeout=2)
+            raised = False
+        except requests.exceptions.SSLError:
+            raised = True
+        assert raised
+        srv.should_exit = True
