@@ -223,7 +223,19 @@ class K8sStore(Store):
             if e.not_found:
                 raise NotFound(f"{kind} {namespace}/{name}")
             raise
+        # finalizers defer deletion on a real apiserver: the object then
+        # still exists with deletionTimestamp set and MUST stay in the
+        # local cache (controllers reconcile it to strip finalizers) —
+        # dropping it locally would orphan the wire object forever
+        try:
+            remote = self.client.get(kind, name, namespace)
+        except ApiError:
+            remote = None
         cur = self.try_get(kind, name, namespace)
+        if remote is not None:
+            from . import serde as _serde
+            self._apply_remote("MODIFIED", _serde.from_k8s(remote))
+            return
         if cur is not None:
             self._apply_remote("DELETED", cur)
         # owner-reference GC: the embedded store cascades in-process; on a
