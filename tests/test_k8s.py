@@ -580,3 +580,72 @@ class TestWebhookTLS:
             raised = True
         assert raised
         srv.should_exit = True
+
+
+class TestFinalizerScaleDownRace:
+    def test_scale_down_never_orphans_deleting_workers(self):
+        """Regression: K8sStore.delete must keep a finalizer-deferred
+        pod in the local cache so PodReconciler can strip the finalizer
+        — dropping it locally orphaned the wire object ~25% of runs
+        (race between the delete write-through and the watch event)."""
+
+        import time as _t
+
+        import tensor_fusion_amd.constants as C
+        from tensor_fusion_amd.k8s.bridge import K8sStore
+        from tensor_fusion_amd.k8s.fake_apiserver import serve_in_thread
+        from tensor_fusion_amd.operator import build_operator
+
+        for trial in range(3):
+            srv, base, us = serve_in_thread()
+            try:
+                cli = K8sClient(base)
+                for crd in all_crds().values():
+                    cli.create(crd)
+                node = {"apiVersion": "v1", "kind": "Node",
+                        "metadata": {"name": "n0"},
+                        "status": {"conditions": [
+                            {"type": "Ready", "status": "True"}]}}
+                cli.create(node)
+                g = T.GPU()
+                g.meta.name = "n0-g0"
+                g.status.uuid = "u0"
+                g.status.node = "n0"
+                g.status.capacity = T.Resource(
+                    2500.0, 288 << 30, 100.0)
+                g.status.available = T.Resource(
+                    2500.0, 288 << 30, 100.0)
+                cli.create(serde.to_k8s(g))
+                store = K8sStore(K8sClient(base)).start()
+                op = build_operator(store=store)
+                wl = T.TensorFusionWorkload()
+                wl.meta.name = "w"
+                wl.meta.namespace = "default"
+                wl.replicas = 1
+                wl.profile.resources.requests = T.Resource(
+                    100.0, 8 << 30, 5.0)
+                wl.profile.resources.limits = T.Resource(
+                    100.0, 8 << 30, 5.0)
+                cli.create(serde.to_k8s(wl))
+                deadline = _t.time() + 10
+                while _t.time() < deadline:
+                    op.tick()
+                    if cli.list_items("Pod", "default"):
+                        break
+                    _t.sleep(0.05)
+                cli.patch("TensorFusionWorkload", "w",
+                          {"spec": {"replicas": 0}}, namespace="default")
+                deadline = _t.time() + 10
+                ok = False
+                while _t.time() < deadline:
+                    op.tick()
+                    if not cli.list_items("Pod", "default"):
+                        ok = True
+                        break
+                    _t.sleep(0.05)
+                assert ok, (trial,
+                            cli.list_items("Pod", "default"))
+                op.stop()
+                store.stop()
+            finally:
+                us.should_exit = True
