@@ -193,6 +193,11 @@ class Store:
             return copy.deepcopy(obj)
 
     def delete(self, kind: str, name: str, namespace: str = "") -> None:
+        # Embedded-mode note: deletion is immediate (DELETED events carry
+        # the final object state, so dealloc-on-event is exact). Real
+        # finalizer deferral — deletionTimestamp until controllers strip
+        # metadata.finalizers — lives in the k8s plane (K8sStore +
+        # fake_apiserver), where an apiserver owns object lifetime.
         key = f"{namespace}/{name}" if namespace else name
         with self._Mutate(self):
             bucket = self._objs.get(kind, {})
