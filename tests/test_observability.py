@@ -390,3 +390,38 @@ def test_recommendation_applied_at_admission():
     prof2 = mut.parse(pod2)
     assert prof2.resources.requests.tflops == 100.0
     assert prof2.resources.requests.vram == 8 << 30
+
+
+def test_assign_port_and_index_routes():
+    """Operator HTTP routes /assign-host-port and /assign-index
+    (reference router assign_host_port.go / assign_index.go):
+    authorized callers get monotonically distinct assignments; missing
+    auth is 401."""
+
+    from fastapi.testclient import TestClient
+
+    from tensor_fusion_amd.operator import build_operator
+    from tensor_fusion_amd.server import create_operator_app
+    from tensor_fusion_amd.server.operator_server import make_token
+
+    op = build_operator()
+    app = create_operator_app(op.store, allocator=op.allocator,
+                              port_allocator=op.port_allocator,
+                              index_allocator=op.index_allocator)
+    c = TestClient(app)
+    assert c.post("/assign-host-port?pod_name=p1").status_code == 401
+
+    hdr = {"Authorization": "Bearer " +
+           make_token("tf-dev-secret", "default", "p1")}
+    p1 = c.post("/assign-host-port?pod_name=p1", headers=hdr).json()
+    p2 = c.post("/assign-host-port?pod_name=p2", headers=hdr).json()
+    assert p1["hostPort"] != p2["hostPort"]
+    assert 42000 <= p1["hostPort"] < 62000
+
+    i1 = c.post("/assign-index?pod_name=p1", headers=hdr).json()
+    i2 = c.post("/assign-index?pod_name=p2", headers=hdr).json()
+    assert i1["index"] != i2["index"]
+    assert 1 <= i1["index"] <= 32
+    # idempotent per pod (re-admission must not leak a second slot)
+    assert c.post("/assign-index?pod_name=p1",
+                  headers=hdr).json()["index"] == i1["index"]
