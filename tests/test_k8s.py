@@ -649,3 +649,42 @@ class TestFinalizerScaleDownRace:
                 store.stop()
             finally:
                 us.should_exit = True
+
+
+class TestMergePatchProperties:
+    def test_rfc7386_properties_randomized(self):
+        """Property-fuzz RFC-7386 semantics of the fake apiserver's
+        merge patch: applying a doc's own diff reproduces it, null
+        deletes, scalars/lists replace wholesale, dicts merge deep."""
+
+        import random
+
+        from tensor_fusion_amd.k8s.bridge import _diff_merge
+        from tensor_fusion_amd.k8s.fake_apiserver import _merge_patch
+
+        rng = random.Random(11)
+
+        def rand_doc(depth=0):
+            if depth > 2 or (depth > 0 and rng.random() < 0.3):
+                return rng.choice([1, "s", True, [1, 2],
+                                   ["a"], 3.5, "x"])
+            # top level is always a dict (k8s objects are)
+            return {f"k{i}": rand_doc(depth + 1)
+                    for i in range(rng.randint(1, 4))}
+
+        for _ in range(200):
+            a, b = rand_doc(), rand_doc()
+            # patching a with diff(a->b) must yield b exactly;
+            # a None diff asserts the docs were already equal
+            d = _diff_merge(a, b)
+            if d is None:
+                assert a == b
+                continue
+            got = _merge_patch(a, d)
+            assert got == b, (a, b, d, got)
+        # explicit RFC cases
+        assert _merge_patch({"a": 1, "b": 2}, {"b": None}) == {"a": 1}
+        assert _merge_patch({"a": {"x": 1}}, {"a": {"y": 2}}) == \
+            {"a": {"x": 1, "y": 2}}
+        assert _merge_patch({"a": [1, 2]}, {"a": [3]}) == {"a": [3]}
+        assert _merge_patch("scalar", {"a": 1}) == {"a": 1}
