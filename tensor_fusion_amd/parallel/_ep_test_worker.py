@@ -42,4 +42,21 @@ if rank == 0:
     used = {int(i) // per for i in ids}
     assert len(used) == world, used
     print("EP_OK", err)
+
+# edge case: EVERY token routes to expert 0 — the non-owning rank's
+# experts receive ZERO tokens and the all-to-all count handshake must
+# move empty payloads without deadlock or shape errors
+with torch.no_grad():
+    # zero router weights: every logit ties at 0 and argmax picks
+    # expert 0 deterministically for every token
+    ref.router.weight.zero_()
+    ep.router.weight.copy_(ref.router.weight)
+    want2 = ref(x)
+    got2 = ep(x)
+err2 = (got2 - want2).abs().max().item()
+assert err2 < 1e-5, f"rank {rank}: starved-expert err {err2}"
+ids2 = ref.router(x).argmax(-1)
+assert set(ids2.tolist()) == {0}
+if rank == 0:
+    print("EP_EMPTY_OK", err2)
 dist.destroy_process_group()

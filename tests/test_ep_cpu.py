@@ -28,3 +28,4 @@ def test_ep_matches_reference_2rank():
     assert all(p.returncode == 0 for p in procs), \
         "\n".join(o + e for o, e in outs)
     assert "EP_OK" in outs[0][0]
+    assert "EP_EMPTY_OK" in outs[0][0]  # zero-token expert path
