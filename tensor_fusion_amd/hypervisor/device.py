@@ -232,19 +232,28 @@ class Accelerator:
             raise RuntimeError("tf_accel_device_count failed")
         return n.value
 
-    def register_log_callback(self, fn) -> None:
-        """fn(level:int, msg:str); keeps the ctypes thunk alive on self
-        (the C side stores the raw pointer)."""
+    # The C side stores the RAW function pointer: thunks must outlive
+    # every library that might call them, so they are pinned process-
+    # wide (a dropped thunk after GC is a guaranteed segfault on the
+    # next log emission — found by test-order shuffling).
+    _LOG_THUNKS: list = []
+    _LOG_CB_T = ctypes.CFUNCTYPE(None, ctypes.c_int, ctypes.c_char_p)
 
-        cb_t = ctypes.CFUNCTYPE(None, ctypes.c_int, ctypes.c_char_p)
+    def register_log_callback(self, fn) -> None:
+        """fn(level:int, msg:str)."""
 
         def thunk(level, msg):
             try:
                 fn(int(level), (msg or b"").decode(errors="replace"))
             except Exception:
                 pass
-        self._log_cb = cb_t(thunk)
-        self._lib.tf_accel_register_log_callback(self._log_cb)
+        cb = self._LOG_CB_T(thunk)
+        Accelerator._LOG_THUNKS.append(cb)
+        self._lib.tf_accel_register_log_callback(cb)
+
+    def unregister_log_callback(self) -> None:
+        self._lib.tf_accel_register_log_callback(
+            ctypes.cast(None, self._LOG_CB_T))
 
 
 class DeviceController:

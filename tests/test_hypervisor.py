@@ -314,6 +314,15 @@ def test_accelerator_full_abi_sweep():
     acc = Accelerator(mock_devices=2)
     logs = []
     acc.register_log_callback(lambda lvl, msg: logs.append((lvl, msg)))
+    try:
+        _abi_sweep_body(acc)
+    finally:
+        # the C side keeps the raw pointer — clear it so later tests
+        # sharing the dlopen'd library never call a dead thunk
+        acc.unregister_log_callback()
+
+
+def _abi_sweep_body(acc):
     assert acc.device_count() == 2
     devs = acc.devices()
     assert len(devs) == 2 and devs[0].compute_units == 256
